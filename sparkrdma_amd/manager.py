@@ -1,0 +1,325 @@
+"""ShuffleManager — the plugin entry point, executor side.
+
+MI355X re-design of the reference's ``RdmaShuffleManager``
+(RdmaShuffleManager.scala:38-42,143,234,263,293,301). Same public surface:
+
+* ``register_shuffle(num_maps, num_partitions)`` → ShuffleHandle
+* ``get_writer(handle, map_id)`` / ``get_reader(handle, start, end)``
+* ``unregister_shuffle(shuffle_id)`` / ``stop()``
+
+Differences forced by the platform, not the design: there is no Spark DAG
+scheduler above us, so ``register_shuffle`` may be called from any
+executor (it RPCs the driver, which allocates the table) and a ``barrier``
+is exposed for stage boundaries.
+
+The reference smuggles the driver table's (addr, len, rkey) through the
+serialized shuffle handle (RdmaUtils.scala:145-159); here the handle
+carries the table's /dev/shm path — same trick, no extra RPC on the hot
+path. Publishing a map output is a one-sided 12-byte write at
+``map_id * 12`` (reference RdmaShuffleManager.scala:384-418); fetching the
+table is a one-sided read, cached per shuffle id (:341-376).
+"""
+
+from __future__ import annotations
+
+import logging
+import mmap
+import os
+import socket
+import threading
+import time
+from dataclasses import dataclass
+from typing import Dict, List, Optional
+
+import numpy as np
+
+from .block_pool import BlockPool
+from .conf import ShuffleConf
+from .map_output import (DriverTable, MAP_ENTRY_SIZE, MapTaskOutput,
+                         make_key, split_key)
+from . import rpc
+from .segments import (FIRST_DATA_SEGMENT_ID, HostSegment, META_SEGMENT_ID,
+                       META_TABLE_REGION_OFF, SegmentRegistry, segment_path)
+from .stats import ShuffleReaderStats
+
+log = logging.getLogger(__name__)
+
+META_SEGMENT_SIZE = 256 << 20   # sparse; map-output tables live here
+
+
+@dataclass(frozen=True)
+class ShuffleHandle:
+    """Carries everything a task needs — reference RdmaBaseShuffleHandle."""
+    shuffle_id: int
+    num_maps: int
+    num_partitions: int
+    driver_table_path: str
+
+
+class ShuffleManager:
+    def __init__(self, conf: ShuffleConf, executor_id: int,
+                 is_executor: bool = True, app_id: Optional[str] = None,
+                 driver_port: Optional[int] = None):
+        self.conf = conf
+        self.executor_id = executor_id
+        self.app_id = app_id          # filled by announce if None
+        self._driver_port = driver_port if driver_port is not None else conf.driver_port
+        self._conn: Optional[rpc.MsgConnection] = None
+        self._members: Dict[int, rpc.ExecutorInfo] = {}
+        self._announce_evt = threading.Event()
+        self._reply_lock = threading.Lock()
+        self._replies: List[tuple] = []
+        self._reply_evt = threading.Event()
+        self._meta_segment: Optional[HostSegment] = None
+        self._meta_bump = META_TABLE_REGION_OFF
+        self._meta_lock = threading.Lock()
+        self._pool: Optional[BlockPool] = None
+        self._data_segments: Dict[int, HostSegment] = {}
+        self._next_segment_id = FIRST_DATA_SEGMENT_ID
+        self._registry: Optional[SegmentRegistry] = None
+        self._driver_tables: Dict[int, mmap.mmap] = {}   # shuffle_id -> rw mmap
+        self._cached_tables: Dict[int, list] = {}        # shuffle_id -> [(addr,key)]
+        self._shuffle_outputs: Dict[int, Dict[int, list]] = {}  # sid -> map_id -> blocks
+        self.reader_stats = (ShuffleReaderStats(conf)
+                             if conf.collect_shuffle_reader_stats else None)
+        self._stopped = False
+        if is_executor:
+            self._connect_and_hello()
+
+    # ------------------------------------------------------------------
+    # control plane
+
+    def _connect_and_hello(self) -> None:
+        """Lazy-node-start analog (reference :186-232): connect to driver,
+        send Hello, wait for the Announce that names the app id."""
+        deadline = time.monotonic() + self.conf.rdma_cm_event_timeout_ms / 1000
+        last_err = None
+        for _ in range(self.conf.max_connection_attempts):
+            try:
+                sock = socket.create_connection(
+                    (self.conf.driver_host, self._driver_port),
+                    timeout=max(0.1, deadline - time.monotonic()))
+                break
+            except OSError as e:
+                last_err = e
+                time.sleep(0.1)
+        else:
+            raise ConnectionError(
+                f"cannot reach driver at {self.conf.driver_host}:{self._driver_port}") \
+                from last_err
+        self._conn = rpc.MsgConnection(sock, self.conf.recv_wr_size)
+        self._recv_thread = threading.Thread(
+            target=self._recv_loop, name=f"sparkrdma-exec{self.executor_id}-recv",
+            daemon=True)
+        self._recv_thread.start()
+        meta_path = ""  # path known only after app_id arrives; re-hello below
+        info = rpc.ExecutorInfo(self.executor_id, self.conf.driver_host, 0,
+                                self.conf.resolved_gpu_id(), meta_path)
+        self._conn.send(rpc.MSG_HELLO, rpc.pack_hello(info))
+        if not self._announce_evt.wait(self.conf.rdma_cm_event_timeout_ms / 1000):
+            raise TimeoutError("no announce from driver")
+        self._init_segments()
+
+    def _recv_loop(self) -> None:
+        try:
+            while True:
+                msg = self._conn.recv_any()
+                if msg is None:
+                    return
+                mtype, body = msg
+                if mtype == rpc.MSG_ANNOUNCE:
+                    app_id, members = rpc.unpack_announce(body)
+                    self.app_id = app_id
+                    self._members = {m.executor_id: m for m in members}
+                    self._announce_evt.set()
+                else:
+                    with self._reply_lock:
+                        self._replies.append((mtype, body))
+                        self._reply_evt.set()
+        except (OSError, ValueError):
+            return
+
+    def _rpc_call(self, mtype: int, body: bytes, timeout: float = 30.0) -> tuple:
+        self._conn.send(mtype, body)
+        deadline = time.monotonic() + timeout
+        while True:
+            if not self._reply_evt.wait(max(0.0, deadline - time.monotonic())):
+                raise TimeoutError(f"driver RPC {mtype} timed out")
+            with self._reply_lock:
+                if self._replies:
+                    reply = self._replies.pop(0)
+                    if not self._replies:
+                        self._reply_evt.clear()
+                    return reply
+                self._reply_evt.clear()
+
+    def _init_segments(self) -> None:
+        path = segment_path(self.conf.shm_dir, self.app_id, self.executor_id,
+                            META_SEGMENT_ID)
+        self._meta_segment = HostSegment(path, META_SEGMENT_SIZE)
+        self._registry = SegmentRegistry(self.conf.shm_dir, self.app_id)
+        self._pool = BlockPool(
+            slab_size=self._host_slab_size(),
+            max_bytes=self.conf.max_buffer_allocation_size,
+            alloc_slab=self._alloc_host_slab,
+            free_slab=self._free_host_slab)
+        for size, count in self.conf.pre_allocate_buffers.items():
+            self._pool.preallocate(size, count)
+
+    def _host_slab_size(self) -> int:
+        return min(1 << 30, self.conf.max_buffer_allocation_size)
+
+    def _alloc_host_slab(self, size: int) -> int:
+        seg_id = self._next_segment_id
+        self._next_segment_id += 1
+        path = segment_path(self.conf.shm_dir, self.app_id, self.executor_id, seg_id)
+        self._data_segments[seg_id] = HostSegment(path, size)
+        return seg_id
+
+    def _free_host_slab(self, seg_id: int) -> None:
+        seg = self._data_segments.pop(seg_id)
+        seg.close()
+        seg.unlink()
+
+    # ------------------------------------------------------------------
+    # ShuffleManager public surface
+
+    def register_shuffle(self, num_maps: int, num_partitions: int) -> ShuffleHandle:
+        mtype, body = self._rpc_call(
+            rpc.MSG_REGISTER, rpc.pack_register(-1, num_maps, num_partitions))
+        assert mtype == rpc.MSG_HANDLE
+        sid, nm, np_, uri, _ = rpc.unpack_handle(body)
+        return ShuffleHandle(sid, nm, np_, uri)
+
+    def get_writer(self, handle: ShuffleHandle, map_id: int):
+        from .writer import ShuffleWriter
+        return ShuffleWriter(self, handle, map_id)
+
+    def get_reader(self, handle: ShuffleHandle, start_partition: int,
+                   end_partition: int):
+        from .reader import ShuffleReader
+        return ShuffleReader(self, handle, start_partition, end_partition)
+
+    def unregister_shuffle(self, shuffle_id: int) -> None:
+        # release served blocks (liveness discipline: blocks stay alive until
+        # shuffle removal — reference RdmaShuffleManager.scala:293-299)
+        outputs = self._shuffle_outputs.pop(shuffle_id, {})
+        for blocks in outputs.values():
+            for b in blocks:
+                b.release()
+        mm = self._driver_tables.pop(shuffle_id, None)
+        if mm is not None:
+            mm.close()
+        self._cached_tables.pop(shuffle_id, None)
+        if self._conn is not None:
+            try:
+                self._rpc_call(rpc.MSG_UNREGISTER, rpc.pack_unregister(shuffle_id))
+            except (TimeoutError, OSError):
+                pass
+
+    def barrier(self, timeout: float = 600.0) -> None:
+        mtype, _ = self._rpc_call(rpc.MSG_BARRIER, b"", timeout=timeout)
+        assert mtype == rpc.MSG_BARRIER_OK
+
+    def stop(self) -> None:
+        if self._stopped:
+            return
+        self._stopped = True
+        if self.reader_stats is not None:
+            self.reader_stats.print_histograms(log)
+        if self._pool is not None:
+            log.info("%s", self._pool.format_stats())
+        if self._conn is not None:
+            try:
+                self._conn.send(rpc.MSG_BYE, b"")
+            except OSError:
+                pass
+            self._conn.close()
+        for mm in self._driver_tables.values():
+            mm.close()
+        self._driver_tables.clear()
+        if self._registry is not None:
+            self._registry.close()
+        for seg in self._data_segments.values():
+            seg.close()
+            seg.unlink()
+        self._data_segments.clear()
+        if self._meta_segment is not None:
+            self._meta_segment.close()
+            self._meta_segment.unlink()
+
+    # ------------------------------------------------------------------
+    # map-output plumbing (used by writer/reader)
+
+    def alloc_table(self, num_partitions: int) -> tuple:
+        """Allocate a MapTaskOutput table in the metadata segment; returns
+        (MapTaskOutput, table_addr)."""
+        nbytes = num_partitions * 16
+        with self._meta_lock:
+            addr = self._meta_bump
+            self._meta_bump += (nbytes + 63) & ~63
+            if self._meta_bump > META_SEGMENT_SIZE:
+                raise MemoryError("metadata segment exhausted")
+        table = MapTaskOutput(num_partitions,
+                              self._meta_segment.view(addr, nbytes))
+        return table, addr
+
+    def _driver_table_mm(self, handle: ShuffleHandle) -> mmap.mmap:
+        mm = self._driver_tables.get(handle.shuffle_id)
+        if mm is None:
+            fd = os.open(handle.driver_table_path, os.O_RDWR)
+            try:
+                mm = mmap.mmap(fd, max(handle.num_maps * MAP_ENTRY_SIZE, 4096))
+            finally:
+                os.close(fd)
+            self._driver_tables[handle.shuffle_id] = mm
+        return mm
+
+    def publish_map_output(self, handle: ShuffleHandle, map_id: int,
+                           table_addr: int) -> None:
+        """One-sided 12-byte write into the driver table at map_id*12
+        (reference RdmaShuffleManager.scala:410-412)."""
+        import struct
+        mm = self._driver_table_mm(handle)
+        key = make_key(self.executor_id, META_SEGMENT_ID)
+        struct.pack_into("<QI", mm, map_id * MAP_ENTRY_SIZE, table_addr, key)
+
+    def get_map_task_output_table(self, handle: ShuffleHandle) -> list:
+        """Hop 1: read the whole driver table one-sidedly, poll until every
+        map has published (key != 0), cache per shuffle id (reference
+        :341-376 + partitionLocationFetchTimeout)."""
+        cached = self._cached_tables.get(handle.shuffle_id)
+        if cached is not None:
+            return cached
+        mm = self._driver_table_mm(handle)
+        deadline = time.monotonic() + \
+            self.conf.partition_location_fetch_timeout_ms / 1000
+        delay = 0.0005
+        while True:
+            raw = bytes(mm[:handle.num_maps * MAP_ENTRY_SIZE])
+            entries = DriverTable.parse(raw)
+            if all(key != 0 for _, key in entries):
+                self._cached_tables[handle.shuffle_id] = entries
+                return entries
+            if time.monotonic() > deadline:
+                missing = sum(1 for _, key in entries if key == 0)
+                raise TimeoutError(
+                    f"shuffle {handle.shuffle_id}: {missing}/{handle.num_maps} "
+                    f"map outputs unpublished after timeout")
+            time.sleep(delay)
+            delay = min(delay * 2, 0.05)
+
+    def keep_alive(self, handle: ShuffleHandle, map_id: int, blocks: list) -> None:
+        """Pin a map task's data blocks until unregister_shuffle."""
+        self._shuffle_outputs.setdefault(handle.shuffle_id, {})[map_id] = blocks
+
+    # one-sided remote read used by the fetcher (hops 2 and 3)
+    def remote_read(self, key: int, addr: int, length: int) -> bytes:
+        return self._registry.read(key, addr, length)
+
+    def data_segment(self, seg_id: int) -> HostSegment:
+        return self._data_segments[seg_id]
+
+    @property
+    def pool(self) -> BlockPool:
+        return self._pool

@@ -1,0 +1,245 @@
+"""Shuffle read path — the 3-hop one-sided fetch pipeline.
+
+Re-design of RdmaShuffleReader.scala + RdmaShuffleFetcherIterator.scala
+(the reference's hot path). Same structure:
+
+* HOP 1: read the whole driver table once, cached per shuffle
+  (RdmaShuffleManager.scala:341-376) — here a polled shm read.
+* HOP 2: per map task, one-sided read of its MapTaskOutput entries for
+  [start_partition, end_partition] (RdmaShuffleFetcherIterator.scala:
+  297-311) — here a pread of the owner's metadata segment.
+* HOP 3: coalesce consecutive block locations up to shuffle_read_block_size
+  (:240-263), gate on max_bytes_in_flight and a per-source request cap
+  (:264-273, :82-83), randomize the pending queue to spread load across
+  source executors (:74-79), then issue one-sided data reads; results are
+  surfaced through a blocking queue as they land (:340-382).
+
+Transport-agnostic: ``manager.remote_read`` is shm pread in host mode and
+an xGMI peer copy in GPU mode — the flow-control logic is identical, as it
+was in the reference (it is pure byte-budget accounting).
+"""
+
+from __future__ import annotations
+
+import queue
+import random
+import threading
+import time
+from concurrent.futures import ThreadPoolExecutor
+from dataclasses import dataclass
+from typing import Iterator, List, Optional
+
+from .manager import ShuffleHandle, ShuffleManager
+from .map_output import ENTRY_SIZE, MapTaskOutput, split_key
+from .stats import TaskMetrics
+
+
+@dataclass(frozen=True)
+class BlockRef:
+    """One shuffle block to fetch: (map_id, partition) at (key, addr, len)."""
+    map_id: int
+    partition: int
+    key: int
+    addr: int
+    length: int
+
+
+@dataclass
+class CoalescedFetch:
+    """One one-sided read covering >=1 consecutive blocks (same segment,
+    contiguous addresses) — the reference's scatter-list READ collapses to
+    a single contiguous read here because consecutive partitions of one map
+    task are laid out back-to-back by the writer."""
+    key: int
+    addr: int
+    length: int
+    blocks: List[BlockRef]
+
+
+@dataclass
+class FetchResult:
+    fetch: CoalescedFetch
+    data: Optional[bytes]
+    error: Optional[BaseException] = None
+    latency_ms: float = 0.0
+
+
+class FetchFailedError(RuntimeError):
+    """Analog of Spark's FetchFailedException — one failure fails the task
+    (reference RdmaShuffleFetcherIterator.scala:167)."""
+
+
+def coalesce_blocks(blocks: List[BlockRef], max_bytes: int,
+                    max_blocks: int = 1 << 30) -> List[CoalescedFetch]:
+    """Group blocks into minimal one-sided reads.
+
+    Blocks merge while they are in the same segment AND contiguous AND the
+    running size stays under max_bytes (reference :240-263; a block larger
+    than max_bytes still fetches as one read). Zero-length blocks are
+    dropped.
+    """
+    out: List[CoalescedFetch] = []
+    cur: Optional[CoalescedFetch] = None
+    for b in blocks:
+        if b.length == 0:
+            continue
+        if (cur is not None and b.key == cur.key
+                and b.addr == cur.addr + cur.length
+                and cur.length + b.length <= max_bytes
+                and len(cur.blocks) < max_blocks):
+            cur.blocks.append(b)
+            cur.length += b.length
+        else:
+            cur = CoalescedFetch(b.key, b.addr, b.length, [b])
+            out.append(cur)
+    return out
+
+
+class FetcherIterator:
+    """Async fetch pipeline with byte-budget flow control."""
+
+    def __init__(self, manager: ShuffleManager, handle: ShuffleHandle,
+                 start_partition: int, end_partition: int,
+                 num_workers: int = 8, seed: Optional[int] = None):
+        self.manager = manager
+        self.handle = handle
+        self.start_partition = start_partition
+        self.end_partition = end_partition  # inclusive
+        self.metrics = TaskMetrics()
+        conf = manager.conf
+        self._max_bytes_in_flight = conf.max_bytes_in_flight
+        self._read_block = conf.shuffle_read_block_size
+        self._reqs_limit = conf.resolved_read_requests_limit()
+        self._results: "queue.Queue[FetchResult]" = queue.Queue()
+        self._pending: List[CoalescedFetch] = []
+        self._lock = threading.Lock()
+        self._bytes_in_flight = 0
+        self._reqs_in_flight = 0
+        self._outstanding = 0     # fetches not yet surfaced to the consumer
+        self._pool = ThreadPoolExecutor(max_workers=num_workers,
+                                        thread_name_prefix="sparkrdma-fetch")
+        self._rng = random.Random(seed)
+        self._failed: Optional[BaseException] = None
+        self._start()
+
+    # ------------------------------------------------------------------
+
+    def _start(self) -> None:
+        mgr = self.manager
+        entries = mgr.get_map_task_output_table(self.handle)  # HOP 1
+        blocks: List[BlockRef] = []
+        span = self.end_partition - self.start_partition + 1
+        for map_id, (table_addr, table_key) in enumerate(entries):
+            # HOP 2: one-sided read of this map's location entries
+            raw = mgr.remote_read(
+                table_key, table_addr + self.start_partition * ENTRY_SIZE,
+                span * ENTRY_SIZE)
+            for i, loc in enumerate(MapTaskOutput.parse_locations(raw)):
+                blocks.append(BlockRef(map_id, self.start_partition + i,
+                                       loc.key, loc.addr, loc.length))
+        fetches = coalesce_blocks(blocks, self._read_block)
+        # randomize to spread load over source executors (reference :74-79)
+        self._rng.shuffle(fetches)
+        with self._lock:
+            self._pending = fetches
+            self._outstanding = len(fetches)
+        self._pump()
+
+    def _pump(self) -> None:
+        """Issue pending fetches while under the byte budget (reference
+        :264-273, re-pumped from next() :365-374)."""
+        to_issue = []
+        with self._lock:
+            while self._pending:
+                f = self._pending[-1]
+                if (self._bytes_in_flight + f.length > self._max_bytes_in_flight
+                        and self._bytes_in_flight > 0):
+                    break
+                if self._reqs_in_flight >= self._reqs_limit:
+                    break
+                self._pending.pop()
+                self._bytes_in_flight += f.length
+                self._reqs_in_flight += 1
+                to_issue.append(f)
+        for f in to_issue:
+            self._pool.submit(self._do_fetch, f)
+
+    def _do_fetch(self, f: CoalescedFetch) -> None:
+        t0 = time.perf_counter()
+        try:
+            data = self.manager.remote_read(f.key, f.addr, f.length)
+            if len(data) != f.length:
+                raise FetchFailedError(
+                    f"short read: {len(data)}/{f.length} at key={f.key:#x}")
+            self._results.put(FetchResult(
+                f, data, latency_ms=(time.perf_counter() - t0) * 1e3))
+        except BaseException as e:  # surfaced to consumer, fails the task
+            self._results.put(FetchResult(f, None, error=e))
+
+    # ------------------------------------------------------------------
+
+    def __iter__(self) -> Iterator[tuple]:
+        """Yields (BlockRef, memoryview) per block, in arrival order."""
+        mgr = self.manager
+        my_exec = mgr.executor_id
+        while True:
+            with self._lock:
+                if self._outstanding == 0:
+                    break
+            t0 = time.perf_counter_ns()
+            res = self._results.get()
+            self.metrics.fetch_wait_ns += time.perf_counter_ns() - t0
+            with self._lock:
+                self._bytes_in_flight -= res.fetch.length
+                self._reqs_in_flight -= 1
+                self._outstanding -= 1
+            self._pump()
+            if res.error is not None:
+                self._failed = res.error
+                self._pool.shutdown(wait=False)
+                raise FetchFailedError(
+                    f"fetch of {len(res.fetch.blocks)} blocks at "
+                    f"key={res.fetch.key:#x} failed") from res.error
+            owner = split_key(res.fetch.key)[0]
+            remote = owner != my_exec
+            if remote:
+                self.metrics.remote_bytes_read += res.fetch.length
+                self.metrics.remote_blocks_fetched += len(res.fetch.blocks)
+                if mgr.reader_stats is not None:
+                    mgr.reader_stats.update(owner, res.latency_ms)
+            else:
+                self.metrics.local_bytes_read += res.fetch.length
+                self.metrics.local_blocks_fetched += len(res.fetch.blocks)
+            view = memoryview(res.data)
+            off = 0
+            for b in res.fetch.blocks:
+                yield b, view[off:off + b.length]
+                off += b.length
+        self._pool.shutdown(wait=False)
+
+
+class ShuffleReader:
+    """Public reader: iterate raw blocks or aggregate per partition."""
+
+    def __init__(self, manager: ShuffleManager, handle: ShuffleHandle,
+                 start_partition: int, end_partition: int):
+        self.manager = manager
+        self.handle = handle
+        self.start_partition = start_partition
+        self.end_partition = end_partition
+        self.fetcher = FetcherIterator(manager, handle,
+                                       start_partition, end_partition)
+
+    def __iter__(self):
+        return iter(self.fetcher)
+
+    def collect_partitions(self) -> dict:
+        """partition -> list of data views (one per map task, unordered)."""
+        out = {p: [] for p in range(self.start_partition, self.end_partition + 1)}
+        for ref, data in self.fetcher:
+            out[ref.partition].append(data)
+        return out
+
+    @property
+    def metrics(self) -> TaskMetrics:
+        return self.fetcher.metrics

@@ -1,0 +1,135 @@
+"""One-sided memory segments — the MI355X replacement for ibverbs MRs.
+
+In the reference, serving executors register memory with the NIC
+(RdmaBuffer.java:118-126, RdmaMappedFile.java:163-189) and remote peers
+read it with one-sided RDMA READ, never involving the server's CPU.
+
+The MI355X-native equivalent on one node:
+
+* **host segments** — files in /dev/shm, mmapped by the owner; a fetcher
+  opens the same file and ``pread``s at (addr, len). The owner is never
+  involved in a read: genuinely one-sided.
+* **HBM segments** — hipMalloc slabs exported once via
+  ``hipIpcGetMemHandle``; a fetcher opens the handle and issues
+  ``hipMemcpyPeerAsync`` over xGMI (see ops/hipshuffle). The 64-byte IPC
+  handle of each slab is published in the owner's *metadata* host segment
+  so resolution itself is one-sided.
+
+A segment is addressed by ``key = (executor_id << 16) | segment_id``
+(map_output.make_key); paths are deterministic in (app_id, executor_id,
+segment_id) so key→path resolution needs no lookup service.
+"""
+
+from __future__ import annotations
+
+import mmap
+import os
+import threading
+from typing import Dict, Optional
+
+from .map_output import make_key, split_key
+
+# well-known segment ids (segment_id 0 is reserved == "unpublished")
+META_SEGMENT_ID = 1      # metadata: map-output tables + HBM slab handle table
+FIRST_DATA_SEGMENT_ID = 2
+
+# metadata segment layout: a slab-handle table header then a bump region for
+# map-output tables. Each slab-handle slot: 8B size + 64B hipIpcMemHandle.
+SLAB_TABLE_MAX = 256
+SLAB_SLOT_SIZE = 8 + 64
+META_TABLE_REGION_OFF = 16 + SLAB_TABLE_MAX * SLAB_SLOT_SIZE
+
+
+def segment_path(shm_dir: str, app_id: str, executor_id: int, segment_id: int) -> str:
+    return os.path.join(shm_dir, f"sparkrdma_{app_id}_e{executor_id}_s{segment_id}")
+
+
+def driver_table_path(shm_dir: str, app_id: str, shuffle_id: int) -> str:
+    return os.path.join(shm_dir, f"sparkrdma_{app_id}_driver_sh{shuffle_id}")
+
+
+class HostSegment:
+    """Owner-side mmapped shm file (sparse; pages materialize on touch)."""
+
+    def __init__(self, path: str, size: int, create: bool = True):
+        self.path = path
+        self.size = size
+        flags = os.O_RDWR | (os.O_CREAT if create else 0)
+        fd = os.open(path, flags, 0o600)
+        try:
+            if create:
+                os.ftruncate(fd, size)
+            self.mm = mmap.mmap(fd, size)
+        finally:
+            os.close(fd)
+
+    def write(self, addr: int, data: bytes) -> None:
+        self.mm[addr:addr + len(data)] = data
+
+    def read(self, addr: int, length: int) -> bytes:
+        return bytes(self.mm[addr:addr + length])
+
+    def view(self, addr: int, length: int) -> memoryview:
+        return memoryview(self.mm)[addr:addr + length]
+
+    def close(self) -> None:
+        self.mm.close()
+
+    def unlink(self) -> None:
+        try:
+            os.unlink(self.path)
+        except FileNotFoundError:
+            pass
+
+
+class HostSegmentReader:
+    """Fetcher-side one-sided reader of a peer's segment (pread, no mmap —
+    avoids faulting the whole file into this process)."""
+
+    def __init__(self, path: str):
+        self.path = path
+        self.fd = os.open(path, os.O_RDONLY)
+
+    def read(self, addr: int, length: int) -> bytes:
+        return os.pread(self.fd, length, addr)
+
+    def read_into(self, buf, addr: int) -> int:
+        return os.preadv(self.fd, [buf], addr)
+
+    def close(self) -> None:
+        os.close(self.fd)
+
+
+class SegmentRegistry:
+    """Fetcher-side cache of opened peer segments, keyed by the 32-bit
+    location key — the analog of the reference's channel cache
+    (RdmaNode.java:283-353): open once per (importer, segment), reuse.
+    """
+
+    def __init__(self, shm_dir: str, app_id: str):
+        self.shm_dir = shm_dir
+        self.app_id = app_id
+        self._readers: Dict[int, HostSegmentReader] = {}
+        self._lock = threading.Lock()
+
+    def reader(self, key: int) -> HostSegmentReader:
+        r = self._readers.get(key)
+        if r is not None:
+            return r
+        with self._lock:
+            r = self._readers.get(key)
+            if r is None:
+                exec_id, seg_id = split_key(key)
+                path = segment_path(self.shm_dir, self.app_id, exec_id, seg_id)
+                r = HostSegmentReader(path)
+                self._readers[key] = r
+        return r
+
+    def read(self, key: int, addr: int, length: int) -> bytes:
+        return self.reader(key).read(addr, length)
+
+    def close(self) -> None:
+        with self._lock:
+            for r in self._readers.values():
+                r.close()
+            self._readers.clear()

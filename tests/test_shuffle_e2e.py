@@ -1,0 +1,169 @@
+"""End-to-end shuffle over the host one-sided data plane (no GPU).
+
+The analog of SURVEY §4 config 1 (local-cluster groupByKey): multiple
+executors (in-process here; multi-process in test_multiprocess.py) write a
+shuffle, every executor reads its partitions one-sidedly, and the merged
+result equals the input.
+"""
+
+import numpy as np
+import pytest
+
+from sparkrdma_amd.conf import ShuffleConf
+from sparkrdma_amd.driver import Driver
+from sparkrdma_amd.manager import ShuffleManager
+from sparkrdma_amd.partitioner import HashPartitioner
+from sparkrdma_amd.reader import FetchFailedError
+from sparkrdma_amd.writer import unpack_partition_segment
+
+
+@pytest.fixture
+def cluster(tmp_path):
+    conf = ShuffleConf(shm_dir=str(tmp_path), max_buffer_allocation_size=1 << 30)
+    driver = Driver(conf)
+    conf.driver_port = driver.port
+    managers = [ShuffleManager(conf, executor_id=i, driver_port=driver.port)
+                for i in range(3)]
+    yield driver, managers
+    for m in managers:
+        m.stop()
+    driver.stop()
+
+
+def test_membership_announce(cluster):
+    driver, managers = cluster
+    import time
+    deadline = time.monotonic() + 5
+    while time.monotonic() < deadline:
+        if all(len(m._members) == 3 for m in managers):
+            break
+        time.sleep(0.01)
+    for m in managers:
+        assert set(m._members) == {0, 1, 2}
+        assert m.app_id == driver.app_id
+
+
+def test_fixed_width_shuffle_roundtrip(cluster):
+    _, managers = cluster
+    R = 8
+    VW = 8  # value width
+    part = HashPartitioner(R)
+    handle = managers[0].register_shuffle(num_maps=3, num_partitions=R)
+    rng = np.random.default_rng(7)
+    all_keys = []
+    for map_id, mgr in enumerate(managers):
+        keys = rng.integers(0, 2 ** 63, 5000, dtype=np.uint64)
+        values = keys.view(np.uint8).reshape(-1, 8).copy()  # value = key bytes
+        all_keys.append(keys)
+        w = mgr.get_writer(handle, map_id)
+        w.write_batch(keys, values)
+        w.stop(True, partitioner=part)
+
+    # each executor reads a disjoint partition range
+    got_keys = []
+    for i, mgr in enumerate(managers):
+        lo = i * R // 3
+        hi = (i + 1) * R // 3 - 1
+        reader = mgr.get_reader(handle, lo, hi)
+        parts = reader.collect_partitions()
+        for p, chunks in parts.items():
+            for chunk in chunks:
+                k, v = unpack_partition_segment(chunk, VW)
+                # value integrity: value bytes == key bytes
+                assert np.array_equal(k.view(np.uint8), v.reshape(-1))
+                # partition correctness
+                assert np.all(part.partition_ids(np.asarray(k)) == p)
+                got_keys.append(np.array(k))
+        # metrics: remote + local bytes observed
+        assert reader.metrics.remote_blocks_fetched + \
+            reader.metrics.local_blocks_fetched > 0
+    got = np.sort(np.concatenate(got_keys))
+    want = np.sort(np.concatenate(all_keys))
+    assert np.array_equal(got, want)
+    managers[0].unregister_shuffle(handle.shuffle_id)
+
+
+def test_bytes_record_shuffle_groupby(cluster):
+    """groupByKey over pickled python records — the plumbing config."""
+    import pickle
+    _, managers = cluster
+    R = 4
+    handle = managers[0].register_shuffle(num_maps=3, num_partitions=R)
+    n_per_map = 1000
+    for map_id, mgr in enumerate(managers):
+        w = mgr.get_writer(handle, map_id)
+        records = [(f"k{i % 50}", (map_id, i)) for i in range(n_per_map)]
+        w.write_records(records, None)
+        w.stop(True)
+    # single executor reads all partitions and groups
+    reader = managers[0].get_reader(handle, 0, R - 1)
+    groups = {}
+    for ref, data in reader:
+        buf = bytes(data)
+        off = 0
+        while off < len(buf):
+            obj, off = _unpickle_one(buf, off)
+            groups.setdefault(obj[0], []).append(obj[1])
+    assert len(groups) == 50
+    assert sum(len(v) for v in groups.values()) == 3 * n_per_map
+    for k, vals in groups.items():
+        assert len(vals) == 3 * n_per_map // 50
+
+
+def _unpickle_one(buf, off):
+    import pickletools
+    import pickle
+    import io
+    bio = io.BytesIO(buf[off:])
+    obj = pickle.load(bio)
+    return obj, off + bio.tell()
+
+
+def test_empty_partitions(cluster):
+    _, managers = cluster
+    R = 64  # more partitions than records -> many empty
+    part = HashPartitioner(R)
+    handle = managers[0].register_shuffle(num_maps=1, num_partitions=R)
+    keys = np.arange(10, dtype=np.uint64)
+    w = managers[0].get_writer(handle, 0)
+    w.write_batch(keys)
+    w.stop(True, partitioner=part)
+    reader = managers[1].get_reader(handle, 0, R - 1)
+    total = []
+    for ref, data in reader:
+        k, _ = unpack_partition_segment(data, 0)
+        total.append(np.array(k))
+    assert np.array_equal(np.sort(np.concatenate(total)), keys)
+
+
+def test_unpublished_map_times_out(cluster):
+    _, managers = cluster
+    conf_backup = managers[1].conf.partition_location_fetch_timeout_ms
+    managers[1].conf.partition_location_fetch_timeout_ms = 200
+    try:
+        handle = managers[0].register_shuffle(num_maps=2, num_partitions=2)
+        w = managers[0].get_writer(handle, 0)
+        w.write_batch(np.arange(5, dtype=np.uint64))
+        w.stop(True, partitioner=HashPartitioner(2))
+        # map 1 never publishes
+        with pytest.raises(TimeoutError):
+            managers[1].get_reader(handle, 0, 1)
+    finally:
+        managers[1].conf.partition_location_fetch_timeout_ms = conf_backup
+
+
+def test_barrier(cluster):
+    _, managers = cluster
+    import threading
+    results = []
+
+    def go(m):
+        m.barrier()
+        results.append(m.executor_id)
+
+    threads = [threading.Thread(target=go, args=(m,)) for m in managers]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=10)
+    assert sorted(results) == [0, 1, 2]
