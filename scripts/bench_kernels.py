@@ -1,0 +1,55 @@
+"""Micro-bench of the radix kernels on one MI355X.
+
+Run on a GPU box:  python scripts/bench_kernels.py [n_million]
+Prints partition + sort throughput (records/s and effective GB/s).
+"""
+
+import sys
+import time
+
+import torch
+
+from sparkrdma_amd.ops import load
+from sparkrdma_amd.ops.radix import radix_partition, sort_pairs
+
+
+def bench(fn, iters=5, warmup=2):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+def main():
+    n = int(float(sys.argv[1]) * 1e6) if len(sys.argv) > 1 else 64_000_000
+    load()
+    torch.manual_seed(0)
+    keys = torch.randint(-2**63, 2**63 - 1, (n,), dtype=torch.int64,
+                         device="cuda")
+    vals = torch.randint(-2**63, 2**63 - 1, (n,), dtype=torch.int64,
+                         device="cuda")
+    rec_bytes = 16
+
+    for nbits in (8, 10, 11):
+        t = bench(lambda: radix_partition(keys, vals, nbits))
+        print(f"partition nbits={nbits}: {t*1e3:8.2f} ms  "
+              f"{n/t/1e9:6.2f} Grec/s  {n*rec_bytes/t/1e9:7.1f} GB/s payload")
+
+    for passes, end_bit in ((4, 32), (7, 56), (8, 64)):
+        kk = keys.clone()
+        vv = vals.clone()
+        t = bench(lambda: sort_pairs(kk, vv, 0, end_bit))
+        print(f"sort {passes} passes (bits 0..{end_bit}): {t*1e3:8.2f} ms  "
+              f"{n/t/1e9:6.2f} Grec/s  {n*rec_bytes/t/1e9:7.1f} GB/s payload")
+
+    # torch baseline for context
+    t = bench(lambda: torch.sort(keys)[0])
+    print(f"torch.sort keys-only baseline: {t*1e3:8.2f} ms  {n/t/1e9:6.2f} Grec/s")
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,63 @@
+// pybind11 bindings for libhipshuffle — torch-free; Python passes raw
+// device pointers (tensor.data_ptr()) and stream handles
+// (torch.cuda.current_stream().cuda_stream).
+
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "common.h"
+#include "hipshuffle.h"
+
+namespace py = pybind11;
+namespace hs = hipshuffle;
+
+PYBIND11_MODULE(_hipshuffle, m) {
+  m.doc() = "MI355X-native shuffle data plane: HBM slabs, ROCm IPC, xGMI "
+            "peer copies, CDNA4 radix partition/sort kernels";
+
+  m.def("device_count", &hs::device_count);
+  m.def("set_device", [](int d) { HIP_CHECK(hipSetDevice(d)); });
+  m.def("device_synchronize", [] { HIP_CHECK(hipDeviceSynchronize()); });
+
+  // slabs + IPC
+  m.def("slab_alloc", &hs::slab_alloc_id, py::arg("bytes"));
+  m.def("slab_free", &hs::slab_free_id);
+  m.def("slab_base", &hs::slab_base_id);
+  m.def("slab_handle",
+        [](int id) { return py::bytes(hs::slab_handle_id(id)); });
+  m.def("ipc_open", [](py::bytes handle) {
+    return hs::ipc_open(std::string(handle));
+  });
+  m.def("ipc_close", &hs::ipc_close);
+  m.def("enable_peer_access", &hs::enable_peer_access);
+
+  // copy engine
+  m.def("read_batch", &hs::read_batch_ids, py::arg("peer"), py::arg("dsts"),
+        py::arg("srcs"), py::arg("sizes"),
+        py::call_guard<py::gil_scoped_release>());
+  m.def("poll_event", &hs::poll_event,
+        py::call_guard<py::gil_scoped_release>());
+  m.def("wait_event", &hs::wait_event,
+        py::call_guard<py::gil_scoped_release>());
+  m.def("memcpy_h2d", &hs::memcpy_h2d,
+        py::call_guard<py::gil_scoped_release>());
+  m.def("memcpy_d2h", &hs::memcpy_d2h,
+        py::call_guard<py::gil_scoped_release>());
+
+  // kernels
+  m.def("radix_hist_bytes", &hs::radix_hist_bytes);
+  m.def("radix_hist", &hs::radix_hist, py::arg("keys"), py::arg("n"),
+        py::arg("shift"), py::arg("nbits"), py::arg("hist"),
+        py::arg("stream") = 0);
+  m.def("radix_scan", &hs::radix_scan, py::arg("hist"), py::arg("n"),
+        py::arg("nbits"), py::arg("totals"), py::arg("stream") = 0);
+  m.def("radix_scatter", &hs::radix_scatter, py::arg("keys"), py::arg("vals"),
+        py::arg("n"), py::arg("shift"), py::arg("nbits"), py::arg("hist"),
+        py::arg("key_dst"), py::arg("val_dst"), py::arg("stream") = 0);
+  m.def("sort_workspace_bytes", &hs::sort_workspace_bytes);
+  m.def("sort_pairs_u64", &hs::sort_pairs_u64, py::arg("keys"),
+        py::arg("vals"), py::arg("tmp_keys"), py::arg("tmp_vals"),
+        py::arg("n"), py::arg("start_bit"), py::arg("end_bit"), py::arg("ws"),
+        py::arg("stream") = 0,
+        py::call_guard<py::gil_scoped_release>());
+}

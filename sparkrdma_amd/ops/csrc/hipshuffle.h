@@ -1,0 +1,41 @@
+// Public surface of the hipshuffle native library.
+#pragma once
+
+#include <cstddef>
+#include <cstdint>
+#include <string>
+#include <vector>
+
+namespace hipshuffle {
+
+// pool.cpp
+int slab_alloc_id(size_t bytes);
+void slab_free_id(int id);
+uintptr_t slab_base_id(int id);
+std::string slab_handle_id(int id);
+uintptr_t ipc_open(const std::string& handle_bytes);
+void ipc_close(uintptr_t ptr);
+void enable_peer_access(int peer_device);
+uint64_t read_batch_ids(int peer, const std::vector<uintptr_t>& dsts,
+                        const std::vector<uintptr_t>& srcs,
+                        const std::vector<size_t>& sizes);
+bool poll_event(uint64_t id);
+void wait_event(uint64_t id);
+void memcpy_h2d(uintptr_t dst, uintptr_t src, size_t n);
+void memcpy_d2h(uintptr_t dst, uintptr_t src, size_t n);
+
+// kernels.hip
+size_t radix_hist_bytes(uint32_t n, int nbits);
+void radix_hist(uintptr_t keys, uint32_t n, int shift, int nbits,
+                uintptr_t hist, uintptr_t stream);
+void radix_scan(uintptr_t hist, uint32_t n, int nbits, uintptr_t totals,
+                uintptr_t stream);
+void radix_scatter(uintptr_t keys, uintptr_t vals, uint32_t n, int shift,
+                   int nbits, uintptr_t hist, uintptr_t key_dst,
+                   uintptr_t val_dst, uintptr_t stream);
+size_t sort_workspace_bytes(uint32_t n);
+int sort_pairs_u64(uintptr_t keys, uintptr_t vals, uintptr_t tmp_keys,
+                   uintptr_t tmp_vals, uint32_t n, int start_bit, int end_bit,
+                   uintptr_t ws, uintptr_t stream);
+
+}  // namespace hipshuffle

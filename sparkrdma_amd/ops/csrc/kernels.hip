@@ -1,0 +1,432 @@
+// CDNA4 (gfx950) shuffle kernels: radix bucket-scatter machinery.
+//
+// One primitive serves both hot paths (SURVEY.md §2.3 "new GPU kernels"):
+//   * map-side partition+serialize: digit = top log2(R) key bits; the
+//     scatter writes keys/values STRAIGHT into their final positions in
+//     HBM block buffers (per-digit destination base pointers) — no
+//     intermediate file, replacing Spark's CPU sort-shuffle writer
+//     (reference RdmaWrapperShuffleWriter.scala:83-102).
+//   * reduce-side LSD radix sort: 8-bit digits, ping-pong buffers.
+//
+// Design (MI355X-first, not a CUDA port):
+//   * wave = 64: rank computation is a 64-lane ballot multi-split
+//     (8..12 __ballot()s of 64 bits), per-wave LDS counters — no atomics.
+//   * stability: wave w owns the contiguous element chunk
+//     [tile + w*64*ITEMS, ...), iterated lane-major, so block order ==
+//     memory order and ranks are stable.
+//   * scattered global writes are coalesced through an LDS exchange
+//     (elements digit-sorted in LDS, then written out linearly — each
+//     digit's run is a contiguous global write burst).
+//   * 3-kernel pass: hist -> column scan -> scatter (classic upsweep/
+//     downsweep; digit-major [ND][nb] hist layout so the scan kernel
+//     reads contiguously).
+//
+// Limits: n < 2^32 elements per call; buckets = pow2, 2^8..2^12.
+
+#include "common.h"
+
+#include <hip/hip_runtime.h>
+
+namespace hipshuffle {
+
+constexpr int BLOCK = 256;        // 4 waves
+constexpr int ITEMS = 16;         // elements per thread
+constexpr int TILE = BLOCK * ITEMS;  // 4096 elements per workgroup
+constexpr int NW = BLOCK / kWave;    // waves per block
+
+__device__ __forceinline__ uint32_t wave_inclusive_scan(uint32_t v) {
+#pragma unroll
+  for (int d = 1; d < kWave; d <<= 1) {
+    uint32_t t = __shfl_up(v, d);
+    if ((threadIdx.x & (kWave - 1)) >= (unsigned)d) v += t;
+  }
+  return v;
+}
+
+// Exclusive scan of arr[nd] in LDS (nd pow2, >= blockDim); sums: >=256 u32.
+__device__ void block_exscan(uint32_t* arr, uint32_t* sums, int nd) {
+  const int tid = threadIdx.x;
+  const int per = nd / BLOCK;
+  uint32_t run = 0;
+#pragma unroll 1
+  for (int k = 0; k < per; ++k) {
+    uint32_t v = arr[tid * per + k];
+    arr[tid * per + k] = run;
+    run += v;
+  }
+  sums[tid] = run;
+  __syncthreads();
+  if (tid < kWave) {
+    uint32_t off = 0;
+#pragma unroll 1
+    for (int c = 0; c < BLOCK / kWave; ++c) {
+      uint32_t v = sums[c * kWave + tid];
+      uint32_t inc = wave_inclusive_scan(v);
+      sums[c * kWave + tid] = off + inc - v;
+      off += __shfl(inc, kWave - 1);
+    }
+  }
+  __syncthreads();
+  uint32_t base = sums[tid];
+#pragma unroll 1
+  for (int k = 0; k < per; ++k) arr[tid * per + k] += base;
+  __syncthreads();
+}
+
+// 64-lane multi-split: lanes with equal `digit` (among `validmask` lanes).
+template <int NBITS>
+__device__ __forceinline__ uint64_t match_lanes(uint32_t digit,
+                                                uint64_t validmask) {
+  uint64_t match = validmask;
+#pragma unroll
+  for (int b = 0; b < NBITS; ++b) {
+    uint64_t vote = __ballot((digit >> b) & 1);
+    match &= ((digit >> b) & 1) ? vote : ~vote;
+  }
+  return match;
+}
+
+// ---------------------------------------------------------------------------
+// Kernel 1: per-block digit histogram. hist layout: [ND][nb] (digit-major).
+
+template <int NBITS>
+__global__ __launch_bounds__(BLOCK) void radix_hist_kernel(
+    const uint64_t* __restrict__ keys, uint32_t n, int shift,
+    uint32_t* __restrict__ hist, uint32_t nb) {
+  constexpr int ND = 1 << NBITS;
+  extern __shared__ char smem_raw[];
+  uint32_t* counters = reinterpret_cast<uint32_t*>(smem_raw);  // [NW][ND]
+  const int tid = threadIdx.x;
+  const int lane = tid & (kWave - 1);
+  const int wave = tid >> 6;
+  const uint32_t b = blockIdx.x;
+  const uint64_t tile_start = (uint64_t)b * TILE;
+  uint32_t* my = counters + wave * ND;
+  for (int d = tid; d < NW * ND; d += BLOCK) counters[d] = 0;
+  __syncthreads();
+
+  const uint64_t chunk = tile_start + (uint64_t)wave * (ITEMS * kWave);
+#pragma unroll 1
+  for (int i = 0; i < ITEMS; ++i) {
+    uint64_t e = chunk + (uint64_t)i * kWave + lane;
+    bool valid = e < n;
+    uint64_t k = valid ? keys[e] : 0;
+    uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
+    uint64_t vm = __ballot(valid);
+    if (valid) {
+      uint64_t match = match_lanes<NBITS>(d, vm);
+      uint64_t lt = (1ull << lane) - 1;
+      if ((match & lt) == 0) my[d] += (uint32_t)__popcll(match);
+    }
+  }
+  __syncthreads();
+  for (int d = tid; d < ND; d += BLOCK) {
+    uint32_t s = 0;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) s += counters[w * ND + d];
+    hist[(uint64_t)d * nb + b] = s;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Kernel 2: column scan. Block d converts hist[d][0..nb) to its exclusive
+// prefix (counts of digit d in blocks < b) and writes totals[d].
+
+__global__ __launch_bounds__(BLOCK) void radix_scan_kernel(
+    uint32_t* __restrict__ hist, uint32_t nb,
+    uint32_t* __restrict__ totals) {
+  __shared__ uint32_t vals[BLOCK];
+  __shared__ uint32_t sums[BLOCK];
+  const int tid = threadIdx.x;
+  const uint32_t d = blockIdx.x;
+  uint32_t* col = hist + (uint64_t)d * nb;
+  uint32_t running = 0;
+  for (uint32_t base = 0; base < nb; base += BLOCK) {
+    uint32_t idx = base + tid;
+    uint32_t v = idx < nb ? col[idx] : 0;
+    vals[tid] = v;
+    __syncthreads();
+    block_exscan(vals, sums, BLOCK);
+    if (idx < nb) col[idx] = running + vals[tid];
+    // chunk total = last exclusive + last value
+    uint32_t chunk_total = vals[BLOCK - 1];
+    __syncthreads();  // vals reuse barrier (block_exscan already synced)
+    if (tid == BLOCK - 1) sums[0] = chunk_total + v;
+    __syncthreads();
+    running += sums[0];
+    __syncthreads();
+  }
+  if (tid == 0) totals[d] = running;
+}
+
+// ---------------------------------------------------------------------------
+// Kernel 2b (sort path): turn totals into per-digit destination pointers
+// inside the ping-pong output buffers — keeps the whole sort device-side
+// (no host round trip between passes).
+
+template <int NBITS>
+__global__ void radix_digit_bases_kernel(const uint32_t* __restrict__ totals,
+                                         uint64_t out_keys, uint64_t out_vals,
+                                         uint64_t* __restrict__ key_dst,
+                                         uint64_t* __restrict__ val_dst) {
+  constexpr int ND = 1 << NBITS;
+  __shared__ uint32_t arr[ND];
+  __shared__ uint32_t sums[BLOCK];
+  const int tid = threadIdx.x;
+  for (int d = tid; d < ND; d += BLOCK) arr[d] = totals[d];
+  __syncthreads();
+  block_exscan(arr, sums, ND);
+  for (int d = tid; d < ND; d += BLOCK) {
+    key_dst[d] = out_keys + (uint64_t)arr[d] * 8;
+    val_dst[d] = out_vals ? out_vals + (uint64_t)arr[d] * 8 : 0;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Kernel 3: rank + LDS exchange + scatter to per-digit destinations.
+
+template <int NBITS, bool HAS_VAL>
+__global__ __launch_bounds__(BLOCK) void radix_scatter_kernel(
+    const uint64_t* __restrict__ keys, const uint64_t* __restrict__ vals,
+    uint32_t n, int shift, const uint32_t* __restrict__ hist, uint32_t nb,
+    const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst) {
+  constexpr int ND = 1 << NBITS;
+  extern __shared__ char smem_raw[];
+  // layout: exchange u64[TILE] | counters u32[NW][ND] | start u32[ND]
+  //         | pref u32[ND] | sums u32[BLOCK]
+  uint64_t* exch = reinterpret_cast<uint64_t*>(smem_raw);
+  uint32_t* counters = reinterpret_cast<uint32_t*>(exch + TILE);
+  uint32_t* start = counters + NW * ND;
+  uint32_t* pref = start + ND;
+  uint32_t* sums = pref + ND;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (kWave - 1);
+  const int wave = tid >> 6;
+  const uint32_t b = blockIdx.x;
+  const uint64_t tile_start = (uint64_t)b * TILE;
+  const uint32_t tile_n =
+      (uint32_t)min((uint64_t)TILE, (uint64_t)n - tile_start);
+  uint32_t* my = counters + wave * ND;
+
+  for (int d = tid; d < NW * ND; d += BLOCK) counters[d] = 0;
+  // preload this block's cross-block digit prefixes
+  for (int d = tid; d < ND; d += BLOCK) pref[d] = hist[(uint64_t)d * nb + b];
+  __syncthreads();
+
+  // phase A: per-element (digit, rank within wave&digit), stable
+  uint64_t key_reg[ITEMS];
+  uint32_t rank_reg[ITEMS];
+  uint32_t dig_reg[ITEMS];
+  const uint64_t chunk = tile_start + (uint64_t)wave * (ITEMS * kWave);
+#pragma unroll 1
+  for (int i = 0; i < ITEMS; ++i) {
+    uint64_t e = chunk + (uint64_t)i * kWave + lane;
+    bool valid = e < n;
+    uint64_t k = valid ? keys[e] : 0;
+    key_reg[i] = k;
+    uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
+    dig_reg[i] = d;
+    uint64_t vm = __ballot(valid);
+    uint32_t r = 0;
+    if (valid) {
+      uint64_t match = match_lanes<NBITS>(d, vm);
+      uint64_t lt = (1ull << lane) - 1;
+      uint32_t rank_in_iter = (uint32_t)__popcll(match & lt);
+      uint32_t c = my[d];
+      r = c + rank_in_iter;
+      if (rank_in_iter == 0) my[d] = c + (uint32_t)__popcll(match);
+    }
+    rank_reg[i] = r;
+  }
+  __syncthreads();
+
+  // scan waves per digit; block totals into start[]
+  for (int d = tid; d < ND; d += BLOCK) {
+    uint32_t run = 0;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) {
+      uint32_t t = counters[w * ND + d];
+      counters[w * ND + d] = run;
+      run += t;
+    }
+    start[d] = run;
+  }
+  __syncthreads();
+  block_exscan(start, sums, ND);  // start[d] = block-local digit start
+
+  // LDS exchange: keys to digit-sorted local order
+#pragma unroll 1
+  for (int i = 0; i < ITEMS; ++i) {
+    uint64_t e = chunk + (uint64_t)i * kWave + lane;
+    if (e < n) {
+      uint32_t d = dig_reg[i];
+      uint32_t j = start[d] + my[d] + rank_reg[i];
+      rank_reg[i] = j;  // reuse: now holds local sorted position
+      exch[j] = key_reg[i];
+    }
+  }
+  __syncthreads();
+
+  // linear write-out of keys; remember (digit, global offset) per slot
+  uint32_t out_d[ITEMS];
+  uint32_t out_off[ITEMS];
+#pragma unroll 1
+  for (int i = 0; i < ITEMS; ++i) {
+    uint32_t j = i * BLOCK + tid;
+    if (j < tile_n) {
+      uint64_t k = exch[j];
+      uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
+      uint32_t off = pref[d] + (j - start[d]);
+      out_d[i] = d;
+      out_off[i] = off;
+      reinterpret_cast<uint64_t*>(key_dst[d])[off] = k;
+    }
+  }
+  if (HAS_VAL) {
+    __syncthreads();
+#pragma unroll 1
+    for (int i = 0; i < ITEMS; ++i) {
+      uint64_t e = chunk + (uint64_t)i * kWave + lane;
+      if (e < n) exch[rank_reg[i]] = vals[e];
+    }
+    __syncthreads();
+#pragma unroll 1
+    for (int i = 0; i < ITEMS; ++i) {
+      uint32_t j = i * BLOCK + tid;
+      if (j < tile_n)
+        reinterpret_cast<uint64_t*>(val_dst[out_d[i]])[out_off[i]] = exch[j];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+
+static inline uint32_t num_tiles(uint32_t n) {
+  return (uint32_t)(((uint64_t)n + TILE - 1) / TILE);
+}
+
+size_t radix_hist_bytes(uint32_t n, int nbits) {
+  return (size_t)(1u << nbits) * num_tiles(n) * sizeof(uint32_t);
+}
+
+template <int NBITS>
+static void hist_launch(const uint64_t* keys, uint32_t n, int shift,
+                        uint32_t* hist, hipStream_t s) {
+  uint32_t nb = num_tiles(n);
+  size_t lds = (size_t)NW * (1 << NBITS) * 4;
+  hipLaunchKernelGGL(radix_hist_kernel<NBITS>, dim3(nb), dim3(BLOCK), lds, s,
+                     keys, n, shift, hist, nb);
+  HIP_CHECK(hipGetLastError());
+}
+
+template <int NBITS, bool HAS_VAL>
+static void scatter_launch(const uint64_t* keys, const uint64_t* vals,
+                           uint32_t n, int shift, const uint32_t* hist,
+                           const uint64_t* key_dst, const uint64_t* val_dst,
+                           hipStream_t s) {
+  constexpr int ND = 1 << NBITS;
+  uint32_t nb = num_tiles(n);
+  size_t lds = (size_t)TILE * 8 + (size_t)NW * ND * 4 + (size_t)ND * 4 * 2 +
+               BLOCK * 4;
+  auto kfn = radix_scatter_kernel<NBITS, HAS_VAL>;
+  static bool attr_set[13] = {};
+  if (lds > 64 * 1024 && !attr_set[NBITS]) {
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(kfn),
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+    attr_set[NBITS] = true;
+  }
+  hipLaunchKernelGGL(kfn, dim3(nb), dim3(BLOCK), lds, s, keys, vals, n, shift,
+                     hist, nb, key_dst, val_dst);
+  HIP_CHECK(hipGetLastError());
+}
+
+#define DISPATCH_NBITS(nbits, FN, ...)                        \
+  switch (nbits) {                                            \
+    case 8: FN<8>(__VA_ARGS__); break;                        \
+    case 9: FN<9>(__VA_ARGS__); break;                        \
+    case 10: FN<10>(__VA_ARGS__); break;                      \
+    case 11: FN<11>(__VA_ARGS__); break;                      \
+    case 12: FN<12>(__VA_ARGS__); break;                      \
+    default: throw std::runtime_error("nbits must be 8..12"); \
+  }
+
+void radix_hist(uintptr_t keys, uint32_t n, int shift, int nbits,
+                uintptr_t hist, uintptr_t stream) {
+  auto s = reinterpret_cast<hipStream_t>(stream);
+  DISPATCH_NBITS(nbits, hist_launch, reinterpret_cast<const uint64_t*>(keys),
+                 n, shift, reinterpret_cast<uint32_t*>(hist), s);
+}
+
+void radix_scan(uintptr_t hist, uint32_t n, int nbits, uintptr_t totals,
+                uintptr_t stream) {
+  auto s = reinterpret_cast<hipStream_t>(stream);
+  uint32_t nb = num_tiles(n);
+  hipLaunchKernelGGL(radix_scan_kernel, dim3(1 << nbits), dim3(BLOCK), 0, s,
+                     reinterpret_cast<uint32_t*>(hist), nb,
+                     reinterpret_cast<uint32_t*>(totals));
+  HIP_CHECK(hipGetLastError());
+}
+
+template <int NBITS>
+static void scatter_hv(const uint64_t* keys, const uint64_t* vals, uint32_t n,
+                       int shift, const uint32_t* hist, const uint64_t* kd,
+                       const uint64_t* vd, hipStream_t s) {
+  if (vals)
+    scatter_launch<NBITS, true>(keys, vals, n, shift, hist, kd, vd, s);
+  else
+    scatter_launch<NBITS, false>(keys, nullptr, n, shift, hist, kd, nullptr, s);
+}
+
+void radix_scatter(uintptr_t keys, uintptr_t vals, uint32_t n, int shift,
+                   int nbits, uintptr_t hist, uintptr_t key_dst,
+                   uintptr_t val_dst, uintptr_t stream) {
+  auto s = reinterpret_cast<hipStream_t>(stream);
+  DISPATCH_NBITS(nbits, scatter_hv, reinterpret_cast<const uint64_t*>(keys),
+                 reinterpret_cast<const uint64_t*>(vals), n, shift,
+                 reinterpret_cast<const uint32_t*>(hist),
+                 reinterpret_cast<const uint64_t*>(key_dst),
+                 reinterpret_cast<const uint64_t*>(val_dst), s);
+}
+
+// Full LSD sort of (keys[, vals]) by bits [start_bit, end_bit).
+// ws layout: hist u32[ND*nb] | totals u32[256] | key_dst u64[256] |
+//            val_dst u64[256].  Returns which buffer holds the result:
+// 0 = keys/vals, 1 = tmp_keys/tmp_vals.
+size_t sort_workspace_bytes(uint32_t n) {
+  return radix_hist_bytes(n, 8) + 256 * 4 + 256 * 8 * 2;
+}
+
+int sort_pairs_u64(uintptr_t keys, uintptr_t vals, uintptr_t tmp_keys,
+                   uintptr_t tmp_vals, uint32_t n, int start_bit, int end_bit,
+                   uintptr_t ws, uintptr_t stream) {
+  auto s = reinterpret_cast<hipStream_t>(stream);
+  uint32_t nb = num_tiles(n);
+  uint32_t* hist = reinterpret_cast<uint32_t*>(ws);
+  uint32_t* totals = hist + (size_t)256 * nb;
+  uint64_t* key_dst = reinterpret_cast<uint64_t*>(totals + 256);
+  uint64_t* val_dst = key_dst + 256;
+  uintptr_t src_k = keys, src_v = vals, dst_k = tmp_keys, dst_v = tmp_vals;
+  int cur = 0;
+  for (int bit = start_bit; bit < end_bit; bit += 8) {
+    hist_launch<8>(reinterpret_cast<const uint64_t*>(src_k), n, bit, hist, s);
+    hipLaunchKernelGGL(radix_scan_kernel, dim3(256), dim3(BLOCK), 0, s, hist,
+                       nb, totals);
+    HIP_CHECK(hipGetLastError());
+    hipLaunchKernelGGL(radix_digit_bases_kernel<8>, dim3(1), dim3(BLOCK), 0, s,
+                       totals, (uint64_t)dst_k, (uint64_t)dst_v, key_dst,
+                       val_dst);
+    HIP_CHECK(hipGetLastError());
+    scatter_hv<8>(reinterpret_cast<const uint64_t*>(src_k),
+                  reinterpret_cast<const uint64_t*>(src_v), n, bit, hist,
+                  key_dst, val_dst, s);
+    std::swap(src_k, dst_k);
+    std::swap(src_v, dst_v);
+    cur ^= 1;
+  }
+  return cur;
+}
+
+}  // namespace hipshuffle

@@ -1,0 +1,218 @@
+// HBM slab backend + xGMI peer-copy engine.
+//
+// MI355X-native replacement for the reference's verbs primitives
+// (SURVEY.md §2.3 mapping table):
+//   ibv_reg_mr                -> hipMalloc slab + hipIpcGetMemHandle
+//   rdma_cm connect           -> hipIpcOpenMemHandle + enable peer access
+//   IBV_WR_RDMA_READ scatter  -> hipMemcpyAsync D2D over xGMI on a per-peer
+//                                stream; the "signaled last WR" is a
+//                                hipEvent recorded after the batch
+//   CQ poller                 -> event queries from the Python fetch pool
+//
+// Allocation POLICY (buddy classes, trim) stays in tested Python
+// (block_pool.py); this layer only provides slabs and copies.
+
+#include "common.h"
+
+#include <atomic>
+#include <cstring>
+#include <mutex>
+#include <unordered_map>
+#include <vector>
+
+namespace hipshuffle {
+
+// ---------------------------------------------------------------------------
+// slabs
+
+struct Slab {
+  void* ptr = nullptr;
+  size_t size = 0;
+  hipIpcMemHandle_t handle{};
+};
+
+class SlabPool {
+ public:
+  // Returns (slab_id). Handle bytes retrievable via handle_bytes().
+  int alloc(size_t bytes) {
+    Slab s;
+    HIP_CHECK(hipMalloc(&s.ptr, bytes));
+    s.size = bytes;
+    HIP_CHECK(hipIpcGetMemHandle(&s.handle, s.ptr));
+    std::lock_guard<std::mutex> g(mu_);
+    int id = next_id_++;
+    slabs_[id] = s;
+    return id;
+  }
+
+  void free(int id) {
+    std::lock_guard<std::mutex> g(mu_);
+    auto it = slabs_.find(id);
+    if (it == slabs_.end()) throw std::runtime_error("bad slab id");
+    HIP_CHECK(hipFree(it->second.ptr));
+    slabs_.erase(it);
+  }
+
+  const Slab& get(int id) {
+    std::lock_guard<std::mutex> g(mu_);
+    auto it = slabs_.find(id);
+    if (it == slabs_.end()) throw std::runtime_error("bad slab id");
+    return it->second;
+  }
+
+  uintptr_t base(int id) { return reinterpret_cast<uintptr_t>(get(id).ptr); }
+
+  std::string handle_bytes(int id) {
+    const Slab& s = get(id);
+    return std::string(reinterpret_cast<const char*>(&s.handle),
+                       sizeof(hipIpcMemHandle_t));
+  }
+
+ private:
+  std::mutex mu_;
+  std::unordered_map<int, Slab> slabs_;
+  int next_id_ = 0;
+};
+
+SlabPool& slab_pool() {
+  static SlabPool p;
+  return p;
+}
+
+// ---------------------------------------------------------------------------
+// imported (peer) slabs — opened IPC handles, cached by the Python side
+
+uintptr_t ipc_open(const std::string& handle_bytes) {
+  if (handle_bytes.size() != sizeof(hipIpcMemHandle_t))
+    throw std::runtime_error("bad ipc handle size");
+  hipIpcMemHandle_t h;
+  std::memcpy(&h, handle_bytes.data(), sizeof(h));
+  void* ptr = nullptr;
+  HIP_CHECK(hipIpcOpenMemHandle(&ptr, h, hipIpcMemLazyEnablePeerAccess));
+  return reinterpret_cast<uintptr_t>(ptr);
+}
+
+void ipc_close(uintptr_t ptr) {
+  HIP_CHECK(hipIpcCloseMemHandle(reinterpret_cast<void*>(ptr)));
+}
+
+void enable_peer_access(int peer_device) {
+  hipError_t e = hipDeviceEnablePeerAccess(peer_device, 0);
+  if (e != hipSuccess && e != hipErrorPeerAccessAlreadyEnabled) HIP_CHECK(e);
+}
+
+// ---------------------------------------------------------------------------
+// per-peer copy engine: streams + event ring
+//
+// The reference posts a scatter list of READ WRs and signals only the last
+// (RdmaChannel.java:484-517). Here: N hipMemcpyAsync on the peer's stream
+// then one recorded event = the signaled WR. Completion is polled
+// (hipEventQuery) from Python's fetch threads — the CQ-poller analog.
+
+struct PeerEngine {
+  hipStream_t stream = nullptr;
+  std::mutex mu;
+};
+
+class CopyEngine {
+ public:
+  static constexpr int kMaxPeers = 64;
+
+  CopyEngine() : peers_(kMaxPeers) {}
+
+  hipStream_t stream_for(int peer) {
+    PeerEngine& p = peers_.at(peer);
+    std::lock_guard<std::mutex> g(p.mu);
+    if (!p.stream)
+      HIP_CHECK(hipStreamCreateWithFlags(&p.stream, hipStreamNonBlocking));
+    return p.stream;
+  }
+
+  // Issue a batch of copies; returns an event id to poll.
+  uint64_t read_batch(int peer, const std::vector<uintptr_t>& dsts,
+                      const std::vector<uintptr_t>& srcs,
+                      const std::vector<size_t>& sizes) {
+    hipStream_t s = stream_for(peer);
+    for (size_t i = 0; i < dsts.size(); ++i) {
+      HIP_CHECK(hipMemcpyAsync(reinterpret_cast<void*>(dsts[i]),
+                               reinterpret_cast<void*>(srcs[i]), sizes[i],
+                               hipMemcpyDeviceToDevice, s));
+    }
+    hipEvent_t ev;
+    HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+    HIP_CHECK(hipEventRecord(ev, s));
+    uint64_t id = next_event_.fetch_add(1);
+    std::lock_guard<std::mutex> g(ev_mu_);
+    events_[id] = ev;
+    return id;
+  }
+
+  // true = complete (event destroyed), false = still in flight
+  bool poll(uint64_t id) {
+    hipEvent_t ev;
+    {
+      std::lock_guard<std::mutex> g(ev_mu_);
+      auto it = events_.find(id);
+      if (it == events_.end()) return true;  // already reaped
+      ev = it->second;
+    }
+    hipError_t e = hipEventQuery(ev);
+    if (e == hipErrorNotReady) return false;
+    HIP_CHECK(e);
+    std::lock_guard<std::mutex> g(ev_mu_);
+    events_.erase(id);
+    HIP_CHECK(hipEventDestroy(ev));
+    return true;
+  }
+
+  void wait(uint64_t id) {
+    hipEvent_t ev;
+    {
+      std::lock_guard<std::mutex> g(ev_mu_);
+      auto it = events_.find(id);
+      if (it == events_.end()) return;
+      ev = it->second;
+    }
+    HIP_CHECK(hipEventSynchronize(ev));
+    std::lock_guard<std::mutex> g(ev_mu_);
+    events_.erase(id);
+    HIP_CHECK(hipEventDestroy(ev));
+  }
+
+ private:
+  std::vector<PeerEngine> peers_;
+  std::atomic<uint64_t> next_event_{1};
+  std::mutex ev_mu_;
+  std::unordered_map<uint64_t, hipEvent_t> events_;
+};
+
+CopyEngine& copy_engine() {
+  static CopyEngine e;
+  return e;
+}
+
+// flat C++ surface (hipshuffle.h)
+int slab_alloc_id(size_t bytes) { return slab_pool().alloc(bytes); }
+void slab_free_id(int id) { slab_pool().free(id); }
+uintptr_t slab_base_id(int id) { return slab_pool().base(id); }
+std::string slab_handle_id(int id) { return slab_pool().handle_bytes(id); }
+uint64_t read_batch_ids(int peer, const std::vector<uintptr_t>& dsts,
+                        const std::vector<uintptr_t>& srcs,
+                        const std::vector<size_t>& sizes) {
+  return copy_engine().read_batch(peer, dsts, srcs, sizes);
+}
+bool poll_event(uint64_t id) { return copy_engine().poll(id); }
+void wait_event(uint64_t id) { return copy_engine().wait(id); }
+
+// host<->device staging helpers (bytes path / spill)
+void memcpy_h2d(uintptr_t dst, uintptr_t src, size_t n) {
+  HIP_CHECK(hipMemcpy(reinterpret_cast<void*>(dst),
+                      reinterpret_cast<void*>(src), n, hipMemcpyHostToDevice));
+}
+
+void memcpy_d2h(uintptr_t dst, uintptr_t src, size_t n) {
+  HIP_CHECK(hipMemcpy(reinterpret_cast<void*>(dst),
+                      reinterpret_cast<void*>(src), n, hipMemcpyDeviceToHost));
+}
+
+}  // namespace hipshuffle

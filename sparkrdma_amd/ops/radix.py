@@ -1,0 +1,91 @@
+"""Torch-facing wrappers over the CDNA4 radix kernels.
+
+Fail-loud policy: on a CUDA(ROCm)-visible machine these functions REQUIRE
+the native extension; there is no silent eager fallback (CPU oracles live
+in tests, not here).
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from . import load
+
+
+def _stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+def radix_partition(keys: torch.Tensor, vals: Optional[torch.Tensor],
+                    nbits: int, shift: Optional[int] = None,
+                    key_dst: Optional[torch.Tensor] = None,
+                    val_dst: Optional[torch.Tensor] = None
+                    ) -> Tuple[torch.Tensor, torch.Tensor, Optional[torch.Tensor]]:
+    """One bucket-scatter pass: digit = (key >> shift) & (2^nbits - 1).
+
+    If key_dst/val_dst (u64 device-pointer tables, one per bucket) are
+    given, elements scatter straight to those addresses (the map-side
+    "write into HBM block buffers" path) and only ``counts`` is returned
+    meaningfully. Otherwise contiguous outputs are allocated and per-digit
+    bases derived from the counts.
+
+    Returns (counts u32[2^nbits] on device, keys_out, vals_out).
+    """
+    m = load()
+    n = keys.numel()
+    assert keys.dtype in (torch.int64, torch.uint64), "keys must be 64-bit"
+    nd = 1 << nbits
+    if shift is None:
+        shift = 64 - nbits
+    dev = keys.device
+    hist = torch.empty(m.radix_hist_bytes(n, nbits) // 4,
+                       dtype=torch.int32, device=dev)
+    totals = torch.empty(nd, dtype=torch.int32, device=dev)
+    s = _stream()
+    m.radix_hist(keys.data_ptr(), n, shift, nbits, hist.data_ptr(), s)
+    m.radix_scan(hist.data_ptr(), n, nbits, totals.data_ptr(), s)
+    keys_out = vals_out = None
+    if key_dst is None:
+        counts64 = totals.to(torch.int64)
+        bases = torch.cumsum(counts64, 0) - counts64
+        keys_out = torch.empty_like(keys)
+        key_dst = keys_out.data_ptr() + bases * 8   # int64 device addresses
+        if vals is not None:
+            vals_out = torch.empty_like(vals)
+            val_dst = vals_out.data_ptr() + bases * 8
+        else:
+            val_dst = torch.zeros(nd, dtype=torch.int64, device=dev)
+    m.radix_scatter(keys.data_ptr(),
+                    vals.data_ptr() if vals is not None else 0,
+                    n, shift, nbits, hist.data_ptr(),
+                    key_dst.data_ptr(), val_dst.data_ptr(), s)
+    return totals, keys_out, vals_out
+
+
+def sort_pairs(keys: torch.Tensor, vals: Optional[torch.Tensor] = None,
+               start_bit: int = 0, end_bit: int = 64
+               ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+    """LSD radix sort of u64 keys (and optional u64 payload) on device.
+
+    Sorts by bits [start_bit, end_bit) — callers that range-partitioned by
+    the top bits pass end_bit = 64 - log2(R) and save whole passes.
+    """
+    m = load()
+    n = keys.numel()
+    if n == 0:
+        return keys, vals
+    dev = keys.device
+    tmp_k = torch.empty_like(keys)
+    tmp_v = torch.empty_like(vals) if vals is not None else None
+    ws = torch.empty(m.sort_workspace_bytes(n), dtype=torch.uint8, device=dev)
+    end_bit = start_bit + ((end_bit - start_bit + 7) // 8) * 8  # whole digits
+    end_bit = min(end_bit, 64)
+    res = m.sort_pairs_u64(
+        keys.data_ptr(), vals.data_ptr() if vals is not None else 0,
+        tmp_k.data_ptr(), tmp_v.data_ptr() if tmp_v is not None else 0,
+        n, start_bit, end_bit, ws.data_ptr(), _stream())
+    if res == 0:
+        return keys, vals
+    return tmp_k, tmp_v
