@@ -1,0 +1,132 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: TeraSort over the MI355X one-sided shuffle.
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W` (launched
+under torch.distributed.run for N>1, one rank per GPU). Rank 0 prints ONE
+JSON line; `value` is the whole-job aggregate sorted throughput in GB/s
+(BASELINE.json metric: 320 GB TeraSort wall-clock + shuffle-read GB/s —
+ms_per_step is the wall-clock of one complete sort job of the configured
+dataset; value = dataset_bytes / wall_clock aggregated over all ranks).
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--gb-per-gpu", type=float,
+                    default=float(os.environ.get("TERASORT_GB_PER_GPU", 40)))
+    ap.add_argument("--mode", choices=["framework", "rccl"],
+                    default=os.environ.get("TERASORT_MODE", "framework"))
+    ap.add_argument("--partitions-per-executor", type=int, default=0,
+                    help="0 = auto (pow2, ~128 per GPU)")
+    ap.add_argument("--validate", action="store_true")
+    ap.add_argument("--cpu", action="store_true",
+                    help="force CPU path (plumbing debug)")
+    args = ap.parse_args()
+
+    import torch
+    use_cuda = torch.cuda.is_available() and not args.cpu
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", args.gpus))
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        dist = tdist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group("nccl" if use_cuda else "gloo")
+        if use_cuda:
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+
+    from sparkrdma_amd.conf import ShuffleConf
+    from sparkrdma_amd.engine import Engine
+    from sparkrdma_amd.workloads.terasort import TeraSort
+
+    n_rec = int(args.gb_per_gpu * (1 << 30) / TeraSort.RECORD_BYTES)
+    ppe = args.partitions_per_executor
+    if ppe == 0:
+        ppe = 128  # world sizes are pow2 in the driver's sweep
+    conf = ShuffleConf(transport="ipc" if use_cuda else "shm")
+    if use_cuda:
+        # input + served blocks + fetch dst + sort ping-pong, with headroom
+        conf.hbm_pool_size = int(args.gb_per_gpu * 1.25 * (1 << 30))
+        conf.shuffle_write_block_size = 64 << 20
+        conf.max_bytes_in_flight = 4 << 30
+    eng = Engine(conf, rank=rank, world_size=world)
+
+    ts = TeraSort(eng, n_rec, partitions_per_executor=ppe,
+                  device="cuda" if use_cuda else "cpu",
+                  mode=args.mode if use_cuda else "framework",
+                  validate=args.validate)
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        else:
+            eng.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        ts.run_step()
+    barrier_sync()
+    t0 = time.perf_counter()
+    results = [ts.run_step() for _ in range(args.steps)]
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if dist is not None:
+        t = torch.tensor([elapsed])
+        if use_cuda:
+            t = t.cuda()
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1e3
+    total_bytes = n_rec * TeraSort.RECORD_BYTES * world
+    value = total_bytes / (elapsed / args.steps) / 1e9  # GB/s whole job
+    remote_gb = sum(r.remote_bytes for r in results) / 1e9
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "terasort_sorted_gb_per_s",
+            "value": round(value, 3),
+            "unit": "GB/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "u64key+u64payload",
+            "data": "synthetic",
+            "config": {
+                "model": "terasort",
+                "global_batch": n_rec * world,
+                "seq_len": TeraSort.RECORD_BYTES,
+                "dataset_gb": round(total_bytes / (1 << 30), 1),
+                "partitions": world * ppe,
+                "mode": ts.mode,
+                "parallelism": f"shuffle{world}",
+                "wall_clock_s_per_job": round(elapsed / args.steps, 3),
+                "remote_gb_per_step": round(remote_gb / max(1, args.steps), 2),
+            },
+        }))
+    eng.shutdown()
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

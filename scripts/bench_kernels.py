@@ -4,8 +4,11 @@ Run on a GPU box:  python scripts/bench_kernels.py [n_million]
 Prints partition + sort throughput (records/s and effective GB/s).
 """
 
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

@@ -47,6 +47,9 @@ class Engine:
             dconf = self.conf
             dconf.driver_port = driver_port
             self.driver = Driver(dconf)
+            driver_port = self.driver.port  # resolves port-0 binds
+        elif driver_port == 0:
+            raise ValueError("non-zero driver_port required on ranks > 0")
         self.manager = ShuffleManager(self.conf, executor_id=self.rank,
                                       driver_port=driver_port)
         # wait until every executor is announced before any stage runs
@@ -62,7 +65,29 @@ class Engine:
     # convenience passthroughs -----------------------------------------
 
     def register_shuffle(self, num_maps: int, num_partitions: int):
-        return self.manager.register_shuffle(num_maps, num_partitions)
+        """SPMD-collective: rank 0 registers with the driver; every rank
+        derives the same handle (ids are sequential, the table path is
+        deterministic in (app_id, shuffle_id)) — the analog of Spark
+        broadcasting the serialized handle to executors."""
+        from .manager import ShuffleHandle
+        from .segments import driver_table_path
+        sid = getattr(self, "_shuffle_seq", 0)
+        self._shuffle_seq = sid + 1
+        if self.rank == 0:
+            h = self.manager.register_shuffle(num_maps, num_partitions)
+            if h.shuffle_id != sid:
+                raise RuntimeError(
+                    f"shuffle id drift: driver={h.shuffle_id} local={sid} "
+                    "(register_shuffle must go through Engine)")
+        self.barrier()  # driver table exists before anyone writes it
+        return ShuffleHandle(sid, num_maps, num_partitions,
+                             driver_table_path(self.conf.shm_dir,
+                                               self.manager.app_id, sid))
+
+    def unregister_shuffle(self, handle) -> None:
+        self.barrier()  # every reader done with served blocks
+        self.manager.unregister_shuffle(handle.shuffle_id,
+                                        notify_driver=self.rank == 0)
 
     def barrier(self) -> None:
         self.manager.barrier()

@@ -1,0 +1,149 @@
+"""GPU data plane: HBM block pool served over ROCm IPC / xGMI.
+
+The MI355X realization of the reference's registered-memory + one-sided
+READ machinery:
+
+* served map outputs live in hipMalloc **slabs** owned by the native lib,
+  sub-allocated by the same buddy BlockPool as host mode. Each slab is
+  exported once via hipIpcGetMemHandle; its 64-byte handle + size are
+  published in a fixed slot of the owner's host *metadata segment*, so a
+  fetcher resolves (key -> handle) one-sidedly — no RPC, mirroring how the
+  reference smuggles rkeys through one-sided table reads.
+* a fetch is a batch of hipMemcpyAsync D2D on the per-peer stream with one
+  completion event — the scatter-list RDMA READ with a single signaled WR
+  (RdmaChannel.java:484-517).
+* GPU segment ids carry the high bit (0x8000); the low 15 bits index the
+  owner's slab-table slot. Addresses in BlockLocations are slab OFFSETS,
+  valid in any importing process.
+
+Fetch destinations are transient torch tensors (consumed by the reduce
+task); only *served* memory needs the IPC-exported slabs.
+"""
+
+from __future__ import annotations
+
+import struct
+import threading
+from typing import Dict, Optional
+
+import torch
+
+from .block_pool import BlockPool
+from .map_output import make_key, split_key
+from .segments import META_SEGMENT_ID, SLAB_SLOT_SIZE, SLAB_TABLE_MAX
+
+GPU_SEG_FLAG = 0x8000
+
+
+def is_gpu_key(key: int) -> bool:
+    return bool((key & 0xFFFF) & GPU_SEG_FLAG)
+
+
+class GpuDataPlane:
+    def __init__(self, conf, executor_id: int, meta_segment, registry,
+                 device: Optional[int] = None):
+        from .ops import load
+        self.hs = load()
+        self.conf = conf
+        self.executor_id = executor_id
+        self.meta_segment = meta_segment     # our own (HostSegment)
+        self.registry = registry             # peers' segments (SegmentRegistry)
+        self.device = conf.resolved_gpu_id() if device is None else device
+        if torch.cuda.is_available():
+            torch.cuda.set_device(self.device)
+        self.hs.set_device(self.device)
+        self._slab_ids: Dict[int, int] = {}   # slot -> native slab id
+        self._slab_bases: Dict[int, int] = {}  # slot -> local base ptr
+        self._next_slot = 0
+        self._peer_bases: Dict[int, int] = {}  # remote key -> imported base
+        self._lock = threading.Lock()
+        pool_max = conf.hbm_pool_size or self._auto_pool_bytes()
+        self.pool = BlockPool(
+            slab_size=conf.hbm_slab_size, max_bytes=pool_max,
+            alloc_slab=self._alloc_slab, free_slab=self._free_slab)
+
+    def _auto_pool_bytes(self) -> int:
+        if torch.cuda.is_available():
+            free, _total = torch.cuda.mem_get_info(self.device)
+            return max(self.conf.hbm_slab_size, int(free * 0.5))
+        return 4 << 30
+
+    # ------------------------------------------------------------------
+    # slab backend for BlockPool
+
+    def _alloc_slab(self, size: int) -> int:
+        with self._lock:
+            slot = self._next_slot
+            if slot >= SLAB_TABLE_MAX:
+                raise MemoryError("slab table full")
+            self._next_slot += 1
+        sid = self.hs.slab_alloc(size)
+        self._slab_ids[slot] = sid
+        self._slab_bases[slot] = self.hs.slab_base(sid)
+        handle = self.hs.slab_handle(sid)
+        # publish (size, handle) one-sidedly in the metadata segment
+        self.meta_segment.write(16 + slot * SLAB_SLOT_SIZE,
+                                struct.pack("<Q", size) + handle)
+        return GPU_SEG_FLAG | slot
+
+    def _free_slab(self, seg_id: int) -> None:
+        slot = seg_id & 0x7FFF
+        self.meta_segment.write(16 + slot * SLAB_SLOT_SIZE, b"\0" * 8)
+        sid = self._slab_ids.pop(slot)
+        self._slab_bases.pop(slot)
+        self.hs.slab_free(sid)
+
+    def local_base(self, seg_id: int) -> int:
+        return self._slab_bases[seg_id & 0x7FFF]
+
+    # ------------------------------------------------------------------
+    # one-sided fetch (hop 3)
+
+    def _resolve_base(self, key: int) -> int:
+        """key -> device pointer of the owning slab in THIS process."""
+        base = self._peer_bases.get(key)
+        if base is not None:
+            return base
+        with self._lock:
+            base = self._peer_bases.get(key)
+            if base is not None:
+                return base
+            exec_id, seg_id = split_key(key)
+            slot = seg_id & 0x7FFF
+            if exec_id == self.executor_id:
+                base = self._slab_bases[slot]
+            else:
+                raw = self.registry.read(
+                    make_key(exec_id, META_SEGMENT_ID),
+                    16 + slot * SLAB_SLOT_SIZE, SLAB_SLOT_SIZE)
+                (size,) = struct.unpack_from("<Q", raw, 0)
+                if size == 0:
+                    raise RuntimeError(
+                        f"peer {exec_id} slab slot {slot} not published")
+                base = self.hs.ipc_open(bytes(raw[8:8 + 64]))
+            self._peer_bases[key] = base
+            return base
+
+    def read_device(self, key: int, addr: int, length: int) -> torch.Tensor:
+        """One-sided read of (key, addr, length) into a fresh device tensor.
+
+        Blocking form (the fetcher's thread pool provides the async rim,
+        like the reference's CQ threads)."""
+        exec_id, _ = split_key(key)
+        base = self._resolve_base(key)
+        dst = torch.empty(length, dtype=torch.uint8,
+                          device=f"cuda:{self.device}")
+        ev = self.hs.read_batch(exec_id % 64, [dst.data_ptr()],
+                                [base + addr], [length])
+        self.hs.wait_event(ev)
+        return dst
+
+    def stop(self) -> None:
+        for key, base in list(self._peer_bases.items()):
+            exec_id, _ = split_key(key)
+            if exec_id != self.executor_id:
+                try:
+                    self.hs.ipc_close(base)
+                except Exception:
+                    pass
+        self._peer_bases.clear()

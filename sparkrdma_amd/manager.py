@@ -47,6 +47,14 @@ log = logging.getLogger(__name__)
 META_SEGMENT_SIZE = 256 << 20   # sparse; map-output tables live here
 
 
+def _cuda_available() -> bool:
+    try:
+        import torch
+        return torch.cuda.is_available()
+    except Exception:
+        return False
+
+
 @dataclass(frozen=True)
 class ShuffleHandle:
     """Carries everything a task needs — reference RdmaBaseShuffleHandle."""
@@ -82,6 +90,7 @@ class ShuffleManager:
         self._shuffle_outputs: Dict[int, Dict[int, list]] = {}  # sid -> map_id -> blocks
         self.reader_stats = (ShuffleReaderStats(conf)
                              if conf.collect_shuffle_reader_stats else None)
+        self.gpu = None
         self._stopped = False
         if is_executor:
             self._connect_and_hello()
@@ -165,6 +174,12 @@ class ShuffleManager:
             free_slab=self._free_host_slab)
         for size, count in self.conf.pre_allocate_buffers.items():
             self._pool.preallocate(size, count)
+        self.gpu = None
+        if self.conf.transport in ("ipc", "rccl") or (
+                self.conf.transport == "auto" and _cuda_available()):
+            from .gpu_plane import GpuDataPlane
+            self.gpu = GpuDataPlane(self.conf, self.executor_id,
+                                    self._meta_segment, self._registry)
 
     def _host_slab_size(self) -> int:
         return min(1 << 30, self.conf.max_buffer_allocation_size)
@@ -200,7 +215,7 @@ class ShuffleManager:
         from .reader import ShuffleReader
         return ShuffleReader(self, handle, start_partition, end_partition)
 
-    def unregister_shuffle(self, shuffle_id: int) -> None:
+    def unregister_shuffle(self, shuffle_id: int, notify_driver: bool = True) -> None:
         # release served blocks (liveness discipline: blocks stay alive until
         # shuffle removal — reference RdmaShuffleManager.scala:293-299)
         outputs = self._shuffle_outputs.pop(shuffle_id, {})
@@ -211,7 +226,7 @@ class ShuffleManager:
         if mm is not None:
             mm.close()
         self._cached_tables.pop(shuffle_id, None)
-        if self._conn is not None:
+        if notify_driver and self._conn is not None:
             try:
                 self._rpc_call(rpc.MSG_UNREGISTER, rpc.pack_unregister(shuffle_id))
             except (TimeoutError, OSError):
@@ -229,6 +244,8 @@ class ShuffleManager:
             self.reader_stats.print_histograms(log)
         if self._pool is not None:
             log.info("%s", self._pool.format_stats())
+        if self.gpu is not None:
+            self.gpu.stop()
         if self._conn is not None:
             try:
                 self._conn.send(rpc.MSG_BYE, b"")
@@ -316,6 +333,12 @@ class ShuffleManager:
     # one-sided remote read used by the fetcher (hops 2 and 3)
     def remote_read(self, key: int, addr: int, length: int) -> bytes:
         return self._registry.read(key, addr, length)
+
+    def remote_read_device(self, key: int, addr: int, length: int):
+        """Hop-3 one-sided read of a GPU-resident block (xGMI peer copy)."""
+        if self.gpu is None:
+            raise RuntimeError("GPU data plane not initialized")
+        return self.gpu.read_device(key, addr, length)
 
     def data_segment(self, seg_id: int) -> HostSegment:
         return self._data_segments[seg_id]

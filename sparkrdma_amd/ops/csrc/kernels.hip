@@ -43,15 +43,17 @@ __device__ __forceinline__ uint32_t wave_inclusive_scan(uint32_t v) {
   return v;
 }
 
-// Exclusive scan of arr[nd] in LDS (nd pow2, >= blockDim); sums: >=256 u32.
+// Exclusive scan of arr[nd] in LDS (nd pow2); sums: >=256 u32.
 __device__ void block_exscan(uint32_t* arr, uint32_t* sums, int nd) {
   const int tid = threadIdx.x;
-  const int per = nd / BLOCK;
+  const int per = (nd + BLOCK - 1) / BLOCK;  // >= 1
+  const int lo = tid * per;
   uint32_t run = 0;
 #pragma unroll 1
   for (int k = 0; k < per; ++k) {
-    uint32_t v = arr[tid * per + k];
-    arr[tid * per + k] = run;
+    int idx = lo + k;
+    uint32_t v = idx < nd ? arr[idx] : 0;
+    if (idx < nd) arr[idx] = run;
     run += v;
   }
   sums[tid] = run;
@@ -69,7 +71,10 @@ __device__ void block_exscan(uint32_t* arr, uint32_t* sums, int nd) {
   __syncthreads();
   uint32_t base = sums[tid];
 #pragma unroll 1
-  for (int k = 0; k < per; ++k) arr[tid * per + k] += base;
+  for (int k = 0; k < per; ++k) {
+    int idx = lo + k;
+    if (idx < nd) arr[idx] += base;
+  }
   __syncthreads();
 }
 
@@ -345,12 +350,16 @@ static void scatter_launch(const uint64_t* keys, const uint64_t* vals,
 
 #define DISPATCH_NBITS(nbits, FN, ...)                        \
   switch (nbits) {                                            \
+    case 4: FN<4>(__VA_ARGS__); break;                        \
+    case 5: FN<5>(__VA_ARGS__); break;                        \
+    case 6: FN<6>(__VA_ARGS__); break;                        \
+    case 7: FN<7>(__VA_ARGS__); break;                        \
     case 8: FN<8>(__VA_ARGS__); break;                        \
     case 9: FN<9>(__VA_ARGS__); break;                        \
     case 10: FN<10>(__VA_ARGS__); break;                      \
     case 11: FN<11>(__VA_ARGS__); break;                      \
     case 12: FN<12>(__VA_ARGS__); break;                      \
-    default: throw std::runtime_error("nbits must be 8..12"); \
+    default: throw std::runtime_error("nbits must be 4..12"); \
   }
 
 void radix_hist(uintptr_t keys, uint32_t n, int shift, int nbits,

@@ -36,16 +36,17 @@ def radix_partition(keys: torch.Tensor, vals: Optional[torch.Tensor],
     m = load()
     n = keys.numel()
     assert keys.dtype in (torch.int64, torch.uint64), "keys must be 64-bit"
-    nd = 1 << nbits
     if shift is None:
         shift = 64 - nbits
+    nbits_eff = max(nbits, 4)  # kernel instantiations start at 4 bits; the
+    nd = 1 << nbits_eff        # extra mask bits read zeros above the digit
     dev = keys.device
-    hist = torch.empty(m.radix_hist_bytes(n, nbits) // 4,
+    hist = torch.empty(m.radix_hist_bytes(n, nbits_eff) // 4,
                        dtype=torch.int32, device=dev)
     totals = torch.empty(nd, dtype=torch.int32, device=dev)
     s = _stream()
-    m.radix_hist(keys.data_ptr(), n, shift, nbits, hist.data_ptr(), s)
-    m.radix_scan(hist.data_ptr(), n, nbits, totals.data_ptr(), s)
+    m.radix_hist(keys.data_ptr(), n, shift, nbits_eff, hist.data_ptr(), s)
+    m.radix_scan(hist.data_ptr(), n, nbits_eff, totals.data_ptr(), s)
     keys_out = vals_out = None
     if key_dst is None:
         counts64 = totals.to(torch.int64)
@@ -59,9 +60,9 @@ def radix_partition(keys: torch.Tensor, vals: Optional[torch.Tensor],
             val_dst = torch.zeros(nd, dtype=torch.int64, device=dev)
     m.radix_scatter(keys.data_ptr(),
                     vals.data_ptr() if vals is not None else 0,
-                    n, shift, nbits, hist.data_ptr(),
+                    n, shift, nbits_eff, hist.data_ptr(),
                     key_dst.data_ptr(), val_dst.data_ptr(), s)
-    return totals, keys_out, vals_out
+    return totals[:1 << nbits], keys_out, vals_out
 
 
 def sort_pairs(keys: torch.Tensor, vals: Optional[torch.Tensor] = None,

@@ -49,7 +49,14 @@ class RangePartitioner:
         span = (key_max - key_min + 1) if key_max < 2 ** 64 - 1 else 2 ** 64
         bounds = [key_min + (span * (i + 1)) // num_partitions
                   for i in range(num_partitions - 1)]
-        return cls(np.array(bounds, dtype=np.uint64))
+        p = cls(np.array(bounds, dtype=np.uint64))
+        # GPU fast path: pow2 partitions over a pow2 key span starting at 0
+        # => partition id is a plain top-bit shift the radix kernel can use.
+        nbits = (num_partitions - 1).bit_length()
+        if (key_min == 0 and (1 << nbits) == num_partitions
+                and span & (span - 1) == 0):
+            p.gpu_shift = span.bit_length() - 1 - nbits
+        return p
 
     def partition_ids(self, keys: np.ndarray) -> np.ndarray:
         k = keys.astype(np.uint64, copy=False)

@@ -165,9 +165,13 @@ class FetcherIterator:
             self._pool.submit(self._do_fetch, f)
 
     def _do_fetch(self, f: CoalescedFetch) -> None:
+        from .gpu_plane import is_gpu_key
         t0 = time.perf_counter()
         try:
-            data = self.manager.remote_read(f.key, f.addr, f.length)
+            if is_gpu_key(f.key):
+                data = self.manager.remote_read_device(f.key, f.addr, f.length)
+            else:
+                data = self.manager.remote_read(f.key, f.addr, f.length)
             if len(data) != f.length:
                 raise FetchFailedError(
                     f"short read: {len(data)}/{f.length} at key={f.key:#x}")
@@ -210,7 +214,10 @@ class FetcherIterator:
             else:
                 self.metrics.local_bytes_read += res.fetch.length
                 self.metrics.local_blocks_fetched += len(res.fetch.blocks)
-            view = memoryview(res.data)
+            data = res.data
+            view = (memoryview(data)
+                    if isinstance(data, (bytes, bytearray, memoryview))
+                    else data)  # device tensor: torch slicing below
             off = 0
             for b in res.fetch.blocks:
                 yield b, view[off:off + b.length]

@@ -1,0 +1,216 @@
+"""TeraSort — the reference's headline workload (README.md:7-17).
+
+Synthetic records: u64 key (uniform random) + u64 payload. One *step* is a
+complete sort job over the fixed per-executor dataset:
+
+  map:    radix-partition the local shard by the top log2(R) key bits into
+          R global partitions, serialized straight into HBM blocks
+          (GPU) or shm blocks (CPU)
+  shuffle: every executor one-sidedly fetches its partition range
+           (xGMI peer copies / shm preads), or RCCL all-to-all in
+           stage mode
+  reduce: LSD radix sort of each partition over the REMAINING low bits
+          (the top bits are constant within a partition — the global
+          partitioning is the MSD pass)
+
+Output invariant (validated): concatenated partitions are globally sorted
+and are a permutation of the input.
+"""
+
+from __future__ import annotations
+
+import logging
+import time
+from dataclasses import dataclass
+from typing import Optional
+
+import numpy as np
+
+from ..engine import Engine
+from ..partitioner import RangePartitioner
+from ..writer import unpack_partition_segment
+
+log = logging.getLogger(__name__)
+
+
+@dataclass
+class TeraSortResult:
+    seconds: float
+    records: int
+    bytes_sorted: int
+    write_s: float
+    fetch_s: float
+    sort_s: float
+    remote_bytes: int
+
+
+class TeraSort:
+    RECORD_BYTES = 16
+
+    def __init__(self, engine: Engine, records_per_executor: int,
+                 partitions_per_executor: int = 64,
+                 device: str = "cpu", mode: str = "framework",
+                 validate: bool = False, seed: int = 0):
+        self.engine = engine
+        self.n = records_per_executor
+        self.device = device
+        self.mode = mode
+        self.validate = validate
+        W = engine.world_size
+        # R must be pow2 for the GPU top-bits partitioner
+        R = W * partitions_per_executor
+        if R & (R - 1):
+            raise ValueError("world_size * partitions_per_executor must be pow2")
+        self.R = R
+        self.ppe = partitions_per_executor
+        self.part = RangePartitioner.uniform(R)
+        self.low_bits = 64 - (R - 1).bit_length()
+        rank = engine.rank
+        if device == "cuda":
+            import torch
+            g = torch.Generator(device="cuda").manual_seed(seed * 1000 + rank)
+            self.keys = torch.randint(-2**63, 2**63 - 1, (self.n,),
+                                      dtype=torch.int64, device="cuda",
+                                      generator=g)
+            self.vals = self.keys.clone()  # payload := key bytes (integrity)
+        else:
+            rng = np.random.default_rng(seed * 1000 + rank)
+            self.keys = rng.integers(0, 2**64, self.n, dtype=np.uint64)
+            self.vals = self.keys.view(np.uint8).reshape(-1, 8).copy()
+
+    # ------------------------------------------------------------------
+
+    def run_step(self) -> TeraSortResult:
+        if self.mode == "rccl":
+            return self._step_rccl()
+        return self._step_framework()
+
+    def _step_framework(self) -> TeraSortResult:
+        eng = self.engine
+        rank = eng.rank
+        t0 = time.perf_counter()
+        handle = eng.register_shuffle(eng.world_size, self.R)
+        w = eng.manager.get_writer(handle, rank)
+        if self.device == "cuda":
+            w.write_device_batch(self.keys, self.vals)
+        else:
+            w.write_batch(self.keys, self.vals)
+        w.stop(True, partitioner=self.part)
+        t_write = time.perf_counter()
+        eng.barrier()
+        lo, hi = rank * self.ppe, (rank + 1) * self.ppe - 1
+        reader = eng.manager.get_reader(handle, lo, hi)
+        parts = reader.collect_partitions()
+        t_fetch = time.perf_counter()
+        sorted_parts = self._reduce(parts)
+        t_sort = time.perf_counter()
+        if self.validate:
+            self._validate(sorted_parts, lo)
+        eng.unregister_shuffle(handle)
+        dt = time.perf_counter() - t0
+        return TeraSortResult(
+            seconds=dt, records=self.n,
+            bytes_sorted=self.n * self.RECORD_BYTES,
+            write_s=t_write - t0, fetch_s=t_fetch - t_write,
+            sort_s=t_sort - t_fetch,
+            remote_bytes=reader.metrics.remote_bytes_read)
+
+    def _reduce(self, parts: dict) -> dict:
+        """Sort each partition's fetched chunks over the low bits."""
+        out = {}
+        if self.device == "cuda":
+            import torch
+            from ..ops.radix import sort_pairs
+            for p, chunks in parts.items():
+                if not chunks:
+                    out[p] = (None, None)
+                    continue
+                ks, vs = [], []
+                for c in chunks:
+                    t = c.view(torch.int64)
+                    nrec = t.numel() // 2
+                    ks.append(t[:nrec])
+                    vs.append(t[nrec:])
+                k = torch.cat(ks) if len(ks) > 1 else ks[0].contiguous()
+                v = torch.cat(vs) if len(vs) > 1 else vs[0].contiguous()
+                out[p] = sort_pairs(k, v, 0, self.low_bits)
+            torch.cuda.synchronize()
+        else:
+            for p, chunks in parts.items():
+                ks, vs = [], []
+                for c in chunks:
+                    k, v = unpack_partition_segment(c, 8)
+                    ks.append(np.array(k))
+                    vs.append(np.array(v))
+                if not ks:
+                    out[p] = (None, None)
+                    continue
+                k = np.concatenate(ks)
+                v = np.concatenate(vs)
+                order = np.argsort(k, kind="stable")
+                out[p] = (k[order], v[order])
+        return out
+
+    def _validate(self, sorted_parts: dict, lo: int) -> None:
+        prev_max = None
+        total = 0
+        for p in sorted(sorted_parts):
+            k, v = sorted_parts[p]
+            if k is None:
+                continue
+            if self.device == "cuda":
+                import torch
+                ku = k.cpu().numpy().view(np.uint64)
+                vu = v.cpu().numpy().view(np.uint64)
+            else:
+                ku, vu = k, v.view(np.uint64).reshape(-1)
+            assert np.all(ku[1:] >= ku[:-1]), f"partition {p} not sorted"
+            assert np.array_equal(ku, vu.reshape(-1)), "payload corrupted"
+            pids = self.part.partition_ids(ku)
+            assert np.all(pids == p), f"foreign keys in partition {p}"
+            if prev_max is not None and len(ku):
+                assert ku[0] >= prev_max
+            if len(ku):
+                prev_max = ku[-1]
+            total += len(ku)
+        log.info("validated %d records in partitions >= %d", total, lo)
+
+    # ------------------------------------------------------------------
+
+    def _step_rccl(self) -> TeraSortResult:
+        """Stage-mode shuffle: RCCL all_to_all_single over xGMI
+        (SURVEY §7.1 'collective-form option')."""
+        import torch
+        import torch.distributed as dist
+        from ..ops.radix import radix_partition, sort_pairs
+        eng = self.engine
+        W = eng.world_size
+        t0 = time.perf_counter()
+        wbits = (W - 1).bit_length()
+        if W == 1:
+            k, v = sort_pairs(self.keys.clone(), self.vals.clone(), 0, 64)
+            torch.cuda.synchronize()
+            dt = time.perf_counter() - t0
+            return TeraSortResult(dt, self.n, self.n * 16, 0, 0, dt, 0)
+        counts, k_part, v_part = radix_partition(
+            self.keys, self.vals, wbits, shift=64 - wbits)
+        in_splits = counts.to(torch.int64)
+        out_splits = torch.empty_like(in_splits)
+        dist.all_to_all_single(out_splits, in_splits)
+        in_l = in_splits.cpu().tolist()
+        out_l = out_splits.cpu().tolist()
+        recv_k = torch.empty(sum(out_l), dtype=torch.int64, device="cuda")
+        recv_v = torch.empty_like(recv_k)
+        dist.all_to_all_single(recv_k, k_part, out_l, in_l)
+        dist.all_to_all_single(recv_v, v_part, out_l, in_l)
+        t_fetch = time.perf_counter()
+        k_sorted, v_sorted = sort_pairs(recv_k, recv_v, 0, 64 - wbits)
+        torch.cuda.synchronize()
+        t_sort = time.perf_counter()
+        if self.validate:
+            ku = k_sorted.cpu().numpy().view(np.uint64)
+            assert np.all(ku[1:] >= ku[:-1])
+        dt = time.perf_counter() - t0
+        return TeraSortResult(dt, self.n, self.n * 16,
+                              t_fetch - t0, 0, t_sort - t_fetch,
+                              int(sum(out_l)) * 16)

@@ -73,6 +73,14 @@ class ShuffleWriter:
         self._value_batches.append(
             None if values is None else np.ascontiguousarray(values, dtype=np.uint8))
 
+    # -- GPU tensor path: CDNA4 partition kernel into HBM blocks --------
+
+    def write_device_batch(self, keys, values=None) -> None:
+        """keys/values: int64 CUDA tensors. Partitioned and serialized by
+        the radix kernel straight into HBM pool blocks at stop()."""
+        self._gpu_batches = getattr(self, "_gpu_batches", [])
+        self._gpu_batches.append((keys, values))
+
     # -- arbitrary-record path -----------------------------------------
 
     def write_records(self, records, partitioner) -> None:
@@ -96,11 +104,14 @@ class ShuffleWriter:
         if not success:
             return
         t0 = time.perf_counter_ns()
-        if self._byte_records is not None:
-            segments = [b"".join(recs) for recs in self._byte_records]
+        if getattr(self, "_gpu_batches", None):
+            self._commit_gpu(partitioner)
         else:
-            segments = self._partition_fixed(partitioner)
-        self._commit_segments(segments)
+            if self._byte_records is not None:
+                segments = [b"".join(recs) for recs in self._byte_records]
+            else:
+                segments = self._partition_fixed(partitioner)
+            self._commit_segments(segments)
         self.metrics.write_ns += time.perf_counter_ns() - t0
 
     def _partition_fixed(self, partitioner) -> List[bytes]:
@@ -125,6 +136,92 @@ class ShuffleWriter:
                     keys_sorted[starts[p]:ends[p]],
                     vals_sorted[starts[p]:ends[p]] if vals_sorted is not None else None)
                 for p in range(R)]
+
+    def _commit_gpu(self, partitioner) -> None:
+        """Map-side GPU write: one radix pass whose scatter writes each
+        partition's [keys|vals] segment STRAIGHT into its final position in
+        HBM pool blocks (the north-star replacement for the reference's
+        CPU writer + mmap + register, RdmaMappedFile.java:113-189)."""
+        import torch
+        from .ops import load as ops_load
+        mgr = self.manager
+        if mgr.gpu is None:
+            raise RuntimeError("GPU writer requires the GPU data plane")
+        hs = ops_load()
+        R = self.handle.num_partitions
+        nbits = max((R - 1).bit_length(), 1)
+        if (1 << nbits) != R or nbits > 12:
+            raise ValueError(
+                f"GPU partitioner requires pow2 partitions <= 4096, got {R}")
+        shift = getattr(partitioner, "gpu_shift", 64 - nbits)
+        nbits_eff = max(nbits, 4)  # kernel instantiations start at 4 bits
+        batches = self._gpu_batches
+        keys = (batches[0][0] if len(batches) == 1
+                else torch.cat([b[0] for b in batches]))
+        vals = (batches[0][1] if len(batches) == 1
+                else torch.cat([b[1] for b in batches]))
+        has_val = vals is not None
+        n = keys.numel()
+        dev = keys.device
+        stream = torch.cuda.current_stream().cuda_stream
+        nd = 1 << nbits_eff
+        hist = torch.empty(hs.radix_hist_bytes(n, nbits_eff) // 4,
+                           dtype=torch.int32, device=dev)
+        totals = torch.empty(nd, dtype=torch.int32, device=dev)
+        hs.radix_hist(keys.data_ptr(), n, shift, nbits_eff, hist.data_ptr(),
+                      stream)
+        hs.radix_scan(hist.data_ptr(), n, nbits_eff, totals.data_ptr(), stream)
+        counts = totals.cpu().numpy().astype(np.int64)[:R]  # syncs the stream
+        rec_w = HEADER_W + (8 if has_val else 0)
+        seg_bytes = counts * rec_w
+
+        # greedy chunking of partitions into HBM blocks (same policy as host)
+        table, table_addr = mgr.alloc_table(R)
+        write_block = mgr.conf.shuffle_write_block_size
+        pool = mgr.gpu.pool
+        blocks = []
+        key_dst = np.zeros(nd, dtype=np.int64)
+        val_dst = np.zeros(nd, dtype=np.int64)
+        group, group_bytes = [], 0
+        flushes = []
+        for p in range(R):
+            if group and group_bytes + seg_bytes[p] > write_block:
+                flushes.append(group)
+                group, group_bytes = [], 0
+            group.append(p)
+            group_bytes += int(seg_bytes[p])
+        if group:
+            flushes.append(group)
+        meta_key = make_key(mgr.executor_id, 1)
+        for parts in flushes:
+            total = int(sum(seg_bytes[p] for p in parts))
+            if total == 0:
+                for p in parts:
+                    table.put(p, 0, 0, meta_key)
+                continue
+            blk = pool.get(total)
+            base = mgr.gpu.local_base(blk.segment_id)
+            key = make_key(mgr.executor_id, blk.segment_id)
+            off = blk.offset
+            for p in parts:
+                nb = int(seg_bytes[p])
+                table.put(p, off, nb, key)
+                if nb:
+                    key_dst[p] = base + off
+                    val_dst[p] = base + off + counts[p] * 8
+                off += nb
+                self.metrics.bytes_written += nb
+            blocks.append(blk)
+        kd = torch.from_numpy(key_dst).to(dev)
+        vd = torch.from_numpy(val_dst).to(dev)
+        hs.radix_scatter(keys.data_ptr(),
+                         vals.data_ptr() if has_val else 0,
+                         n, shift, nbits_eff, hist.data_ptr(),
+                         kd.data_ptr(), vd.data_ptr(), stream)
+        torch.cuda.synchronize()
+        self.metrics.records_written += n
+        mgr.keep_alive(self.handle, self.map_id, blocks)
+        mgr.publish_map_output(self.handle, self.map_id, table_addr)
 
     def _commit_segments(self, segments: List[bytes]) -> None:
         """Greedy chunking at partition boundaries + table fill + publish."""
