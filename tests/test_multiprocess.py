@@ -24,19 +24,8 @@ def _worker(rank, world_size, driver_port, shm_dir, q):
                      driver_port=driver_port)
         R = world_size * 2
         part = RangePartitioner.uniform(R, key_max=2 ** 32 - 1)
-        if rank == 0:
-            handle = eng.register_shuffle(num_maps=world_size, num_partitions=R)
-            hbytes = pickle.dumps(handle)
-        else:
-            handle = None
-        eng.barrier()
-        if rank != 0:
-            # every rank can ask the driver; ids are deterministic, but use
-            # register-on-rank0 + re-read: shuffle_id 0 was just created.
-            from sparkrdma_amd.manager import ShuffleHandle
-            from sparkrdma_amd.segments import driver_table_path
-            handle = ShuffleHandle(0, world_size, R, driver_table_path(
-                shm_dir, eng.manager.app_id, 0))
+        # SPMD-collective registration: all ranks call, handles agree
+        handle = eng.register_shuffle(num_maps=world_size, num_partitions=R)
         rng = np.random.default_rng(rank)
         keys = rng.integers(0, 2 ** 32, 20000, dtype=np.uint64)
         w = eng.manager.get_writer(handle, rank)
