@@ -57,10 +57,10 @@ def main():
         ppe = 128  # world sizes are pow2 in the driver's sweep
     conf = ShuffleConf(transport="ipc" if use_cuda else "shm")
     if use_cuda:
-        # input + served blocks + fetch dst + sort ping-pong, with headroom
-        conf.hbm_pool_size = int(args.gb_per_gpu * 1.25 * (1 << 30))
+        # served blocks are pow2-rounded (buddy) -> up to 2x data, + slack
+        conf.hbm_pool_size = int((args.gb_per_gpu * 2.5 + 2) * (1 << 30))
         conf.shuffle_write_block_size = 64 << 20
-        conf.max_bytes_in_flight = 4 << 30
+        conf.max_bytes_in_flight = 8 << 30
     eng = Engine(conf, rank=rank, world_size=world)
 
     ts = TeraSort(eng, n_rec, partitions_per_executor=ppe,

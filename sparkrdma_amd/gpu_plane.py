@@ -77,7 +77,14 @@ class GpuDataPlane:
             if slot >= SLAB_TABLE_MAX:
                 raise MemoryError("slab table full")
             self._next_slot += 1
-        sid = self.hs.slab_alloc(size)
+        try:
+            sid = self.hs.slab_alloc(size)
+        except RuntimeError:
+            # torch's caching allocator may be hoarding freed blocks;
+            # release them and retry once before giving up
+            if torch.cuda.is_available():
+                torch.cuda.empty_cache()
+            sid = self.hs.slab_alloc(size)
         self._slab_ids[slot] = sid
         self._slab_bases[slot] = self.hs.slab_base(sid)
         handle = self.hs.slab_handle(sid)

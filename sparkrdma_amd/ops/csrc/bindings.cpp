@@ -46,11 +46,13 @@ PYBIND11_MODULE(_hipshuffle, m) {
 
   // kernels
   m.def("radix_hist_bytes", &hs::radix_hist_bytes);
+  m.def("radix_scan_ws_bytes", &hs::radix_scan_ws_bytes);
   m.def("radix_hist", &hs::radix_hist, py::arg("keys"), py::arg("n"),
         py::arg("shift"), py::arg("nbits"), py::arg("hist"),
         py::arg("stream") = 0);
   m.def("radix_scan", &hs::radix_scan, py::arg("hist"), py::arg("n"),
-        py::arg("nbits"), py::arg("totals"), py::arg("stream") = 0);
+        py::arg("nbits"), py::arg("totals"), py::arg("scan_ws"),
+        py::arg("stream") = 0);
   m.def("radix_scatter", &hs::radix_scatter, py::arg("keys"), py::arg("vals"),
         py::arg("n"), py::arg("shift"), py::arg("nbits"), py::arg("hist"),
         py::arg("key_dst"), py::arg("val_dst"), py::arg("stream") = 0);

@@ -26,10 +26,11 @@ void memcpy_d2h(uintptr_t dst, uintptr_t src, size_t n);
 
 // kernels.hip
 size_t radix_hist_bytes(uint32_t n, int nbits);
+size_t radix_scan_ws_bytes(uint32_t n, int nbits);
 void radix_hist(uintptr_t keys, uint32_t n, int shift, int nbits,
                 uintptr_t hist, uintptr_t stream);
 void radix_scan(uintptr_t hist, uint32_t n, int nbits, uintptr_t totals,
-                uintptr_t stream);
+                uintptr_t scan_ws, uintptr_t stream);
 void radix_scatter(uintptr_t keys, uintptr_t vals, uint32_t n, int shift,
                    int nbits, uintptr_t hist, uintptr_t key_dst,
                    uintptr_t val_dst, uintptr_t stream);

@@ -92,7 +92,9 @@ __device__ __forceinline__ uint64_t match_lanes(uint32_t digit,
 }
 
 // ---------------------------------------------------------------------------
-// Kernel 1: per-block digit histogram. hist layout: [ND][nb] (digit-major).
+// Kernel 1: per-block digit histogram. hist layout: [nb][ND] (row-major) —
+// hist writes, scan passes and the scatter's prefix preload are all
+// coalesced across the 256 threads (thread==digit lane).
 
 template <int NBITS>
 __global__ __launch_bounds__(BLOCK) void radix_hist_kernel(
@@ -129,39 +131,68 @@ __global__ __launch_bounds__(BLOCK) void radix_hist_kernel(
     uint32_t s = 0;
 #pragma unroll
     for (int w = 0; w < NW; ++w) s += counters[w * ND + d];
-    hist[(uint64_t)d * nb + b] = s;
+    hist[(uint64_t)b * ND + d] = s;
   }
 }
 
 // ---------------------------------------------------------------------------
-// Kernel 2: column scan. Block d converts hist[d][0..nb) to its exclusive
-// prefix (counts of digit d in blocks < b) and writes totals[d].
+// Kernel 2: hierarchical scan of hist[nb][ND] along the block axis per
+// digit. Three coalesced passes (thread == digit lane, rows iterated
+// serially): K2a per-chunk digit sums -> K2b single-block scan over
+// chunks (+ digit totals) -> K2c rewrite rows as running prefixes.
+// Cross-chunk sums travel through global memory between kernels — kernel
+// boundaries are the release/acquire, no in-launch cross-XCD protocol
+// needed. The matrix is ~2% of a pass's traffic; the naive per-column
+// chained scan was O(nb) *latency* (4.5 ms at nb=655k).
 
-__global__ __launch_bounds__(BLOCK) void radix_scan_kernel(
-    uint32_t* __restrict__ hist, uint32_t nb,
-    uint32_t* __restrict__ totals) {
-  __shared__ uint32_t vals[BLOCK];
-  __shared__ uint32_t sums[BLOCK];
-  const int tid = threadIdx.x;
-  const uint32_t d = blockIdx.x;
-  uint32_t* col = hist + (uint64_t)d * nb;
-  uint32_t running = 0;
-  for (uint32_t base = 0; base < nb; base += BLOCK) {
-    uint32_t idx = base + tid;
-    uint32_t v = idx < nb ? col[idx] : 0;
-    vals[tid] = v;
-    __syncthreads();
-    block_exscan(vals, sums, BLOCK);
-    if (idx < nb) col[idx] = running + vals[tid];
-    // chunk total = last exclusive + last value
-    uint32_t chunk_total = vals[BLOCK - 1];
-    __syncthreads();  // vals reuse barrier (block_exscan already synced)
-    if (tid == BLOCK - 1) sums[0] = chunk_total + v;
-    __syncthreads();
-    running += sums[0];
-    __syncthreads();
+template <int ND>
+__global__ __launch_bounds__(BLOCK) void scan_chunk_sums_kernel(
+    const uint32_t* __restrict__ hist, uint32_t nb, uint32_t chunk_rows,
+    uint32_t* __restrict__ partial) {
+  const uint32_t c = blockIdx.x;
+  const uint32_t lo = c * chunk_rows, hi = min(lo + chunk_rows, nb);
+  for (int d = threadIdx.x; d < ND; d += BLOCK) {
+    uint32_t run = 0;
+    for (uint32_t r = lo; r < hi; ++r) run += hist[(uint64_t)r * ND + d];
+    partial[(uint64_t)c * ND + d] = run;
   }
-  if (tid == 0) totals[d] = running;
+}
+
+template <int ND>
+__global__ __launch_bounds__(BLOCK) void scan_chunks_kernel(
+    uint32_t* __restrict__ partial, uint32_t chunks,
+    uint32_t* __restrict__ totals) {
+  for (int d = threadIdx.x; d < ND; d += BLOCK) {
+    uint32_t run = 0;
+    for (uint32_t c = 0; c < chunks; ++c) {
+      uint32_t v = partial[(uint64_t)c * ND + d];
+      partial[(uint64_t)c * ND + d] = run;
+      run += v;
+    }
+    totals[d] = run;
+  }
+}
+
+template <int ND>
+__global__ __launch_bounds__(BLOCK) void scan_rewrite_kernel(
+    uint32_t* __restrict__ hist, uint32_t nb, uint32_t chunk_rows,
+    const uint32_t* __restrict__ partial) {
+  const uint32_t c = blockIdx.x;
+  const uint32_t lo = c * chunk_rows, hi = min(lo + chunk_rows, nb);
+  for (int d = threadIdx.x; d < ND; d += BLOCK) {
+    uint32_t run = partial[(uint64_t)c * ND + d];
+    for (uint32_t r = lo; r < hi; ++r) {
+      uint32_t v = hist[(uint64_t)r * ND + d];
+      hist[(uint64_t)r * ND + d] = run;
+      run += v;
+    }
+  }
+}
+
+static inline uint32_t scan_chunk_rows(uint32_t nb) {
+  // target ~2048 blocks; at least 64 rows per chunk to amortize
+  uint32_t c = (nb + 2047) / 2048;
+  return c < 64 ? 64 : c;
 }
 
 // ---------------------------------------------------------------------------
@@ -215,8 +246,8 @@ __global__ __launch_bounds__(BLOCK) void radix_scatter_kernel(
   uint32_t* my = counters + wave * ND;
 
   for (int d = tid; d < NW * ND; d += BLOCK) counters[d] = 0;
-  // preload this block's cross-block digit prefixes
-  for (int d = tid; d < ND; d += BLOCK) pref[d] = hist[(uint64_t)d * nb + b];
+  // preload this block's cross-block digit prefixes (coalesced row read)
+  for (int d = tid; d < ND; d += BLOCK) pref[d] = hist[(uint64_t)b * ND + d];
   __syncthreads();
 
   // phase A: per-element (digit, rank within wave&digit), stable
@@ -369,14 +400,37 @@ void radix_hist(uintptr_t keys, uint32_t n, int shift, int nbits,
                  n, shift, reinterpret_cast<uint32_t*>(hist), s);
 }
 
+size_t radix_scan_ws_bytes(uint32_t n, int nbits) {
+  uint32_t nb = num_tiles(n);
+  uint32_t chunk = scan_chunk_rows(nb);
+  uint32_t chunks = (nb + chunk - 1) / chunk;
+  return (size_t)chunks * (1u << nbits) * sizeof(uint32_t);
+}
+
+template <int NBITS>
+static void scan_launch(uint32_t* hist, uint32_t nb, uint32_t* partial,
+                        uint32_t* totals, hipStream_t s) {
+  constexpr int ND = 1 << NBITS;
+  uint32_t chunk = scan_chunk_rows(nb);
+  uint32_t chunks = (nb + chunk - 1) / chunk;
+  hipLaunchKernelGGL(scan_chunk_sums_kernel<ND>, dim3(chunks), dim3(BLOCK), 0,
+                     s, hist, nb, chunk, partial);
+  HIP_CHECK(hipGetLastError());
+  hipLaunchKernelGGL(scan_chunks_kernel<ND>, dim3(1), dim3(BLOCK), 0, s,
+                     partial, chunks, totals);
+  HIP_CHECK(hipGetLastError());
+  hipLaunchKernelGGL(scan_rewrite_kernel<ND>, dim3(chunks), dim3(BLOCK), 0, s,
+                     hist, nb, chunk, partial);
+  HIP_CHECK(hipGetLastError());
+}
+
 void radix_scan(uintptr_t hist, uint32_t n, int nbits, uintptr_t totals,
-                uintptr_t stream) {
+                uintptr_t scan_ws, uintptr_t stream) {
   auto s = reinterpret_cast<hipStream_t>(stream);
   uint32_t nb = num_tiles(n);
-  hipLaunchKernelGGL(radix_scan_kernel, dim3(1 << nbits), dim3(BLOCK), 0, s,
-                     reinterpret_cast<uint32_t*>(hist), nb,
-                     reinterpret_cast<uint32_t*>(totals));
-  HIP_CHECK(hipGetLastError());
+  DISPATCH_NBITS(nbits, scan_launch, reinterpret_cast<uint32_t*>(hist), nb,
+                 reinterpret_cast<uint32_t*>(scan_ws),
+                 reinterpret_cast<uint32_t*>(totals), s);
 }
 
 template <int NBITS>
@@ -405,7 +459,8 @@ void radix_scatter(uintptr_t keys, uintptr_t vals, uint32_t n, int shift,
 //            val_dst u64[256].  Returns which buffer holds the result:
 // 0 = keys/vals, 1 = tmp_keys/tmp_vals.
 size_t sort_workspace_bytes(uint32_t n) {
-  return radix_hist_bytes(n, 8) + 256 * 4 + 256 * 8 * 2;
+  return radix_hist_bytes(n, 8) + radix_scan_ws_bytes(n, 8) + 256 * 4 +
+         256 * 8 * 2;
 }
 
 int sort_pairs_u64(uintptr_t keys, uintptr_t vals, uintptr_t tmp_keys,
@@ -414,16 +469,15 @@ int sort_pairs_u64(uintptr_t keys, uintptr_t vals, uintptr_t tmp_keys,
   auto s = reinterpret_cast<hipStream_t>(stream);
   uint32_t nb = num_tiles(n);
   uint32_t* hist = reinterpret_cast<uint32_t*>(ws);
-  uint32_t* totals = hist + (size_t)256 * nb;
+  uint32_t* scan_ws = hist + (size_t)256 * nb;
+  uint32_t* totals = scan_ws + radix_scan_ws_bytes(n, 8) / 4;
   uint64_t* key_dst = reinterpret_cast<uint64_t*>(totals + 256);
   uint64_t* val_dst = key_dst + 256;
   uintptr_t src_k = keys, src_v = vals, dst_k = tmp_keys, dst_v = tmp_vals;
   int cur = 0;
   for (int bit = start_bit; bit < end_bit; bit += 8) {
     hist_launch<8>(reinterpret_cast<const uint64_t*>(src_k), n, bit, hist, s);
-    hipLaunchKernelGGL(radix_scan_kernel, dim3(256), dim3(BLOCK), 0, s, hist,
-                       nb, totals);
-    HIP_CHECK(hipGetLastError());
+    scan_launch<8>(hist, nb, scan_ws, totals, s);
     hipLaunchKernelGGL(radix_digit_bases_kernel<8>, dim3(1), dim3(BLOCK), 0, s,
                        totals, (uint64_t)dst_k, (uint64_t)dst_v, key_dst,
                        val_dst);

@@ -43,10 +43,13 @@ def radix_partition(keys: torch.Tensor, vals: Optional[torch.Tensor],
     dev = keys.device
     hist = torch.empty(m.radix_hist_bytes(n, nbits_eff) // 4,
                        dtype=torch.int32, device=dev)
+    scan_ws = torch.empty(m.radix_scan_ws_bytes(n, nbits_eff) // 4,
+                          dtype=torch.int32, device=dev)
     totals = torch.empty(nd, dtype=torch.int32, device=dev)
     s = _stream()
     m.radix_hist(keys.data_ptr(), n, shift, nbits_eff, hist.data_ptr(), s)
-    m.radix_scan(hist.data_ptr(), n, nbits_eff, totals.data_ptr(), s)
+    m.radix_scan(hist.data_ptr(), n, nbits_eff, totals.data_ptr(),
+                 scan_ws.data_ptr(), s)
     keys_out = vals_out = None
     if key_dst is None:
         counts64 = totals.to(torch.int64)

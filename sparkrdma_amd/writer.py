@@ -167,10 +167,13 @@ class ShuffleWriter:
         nd = 1 << nbits_eff
         hist = torch.empty(hs.radix_hist_bytes(n, nbits_eff) // 4,
                            dtype=torch.int32, device=dev)
+        scan_ws = torch.empty(hs.radix_scan_ws_bytes(n, nbits_eff) // 4,
+                              dtype=torch.int32, device=dev)
         totals = torch.empty(nd, dtype=torch.int32, device=dev)
         hs.radix_hist(keys.data_ptr(), n, shift, nbits_eff, hist.data_ptr(),
                       stream)
-        hs.radix_scan(hist.data_ptr(), n, nbits_eff, totals.data_ptr(), stream)
+        hs.radix_scan(hist.data_ptr(), n, nbits_eff, totals.data_ptr(),
+                      scan_ws.data_ptr(), stream)
         counts = totals.cpu().numpy().astype(np.int64)[:R]  # syncs the stream
         rec_w = HEADER_W + (8 if has_val else 0)
         seg_bytes = counts * rec_w

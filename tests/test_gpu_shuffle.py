@@ -1,0 +1,92 @@
+"""End-to-end GPU shuffle: HBM blocks, IPC publication, one-sided device
+fetch, radix reduce sort — single process and cross-process on one GPU."""
+
+import multiprocessing as mp
+import os
+import socket
+import sys
+
+import numpy as np
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+def test_smoke_entry():
+    import __graft_entry__ as ge
+    ge.smoke()
+
+
+def test_terasort_framework_single_gpu(tmp_path):
+    import torch
+    from sparkrdma_amd.conf import ShuffleConf
+    from sparkrdma_amd.engine import Engine
+    from sparkrdma_amd.workloads.terasort import TeraSort
+
+    conf = ShuffleConf(transport="ipc", hbm_pool_size=2 << 30,
+                       shm_dir=str(tmp_path))
+    eng = Engine(conf, rank=0, world_size=1, driver_port=0)
+    try:
+        ts = TeraSort(eng, records_per_executor=2_000_000,
+                      partitions_per_executor=512, device="cuda",
+                      validate=True)
+        r1 = ts.run_step()
+        r2 = ts.run_step()  # steps are re-runnable (pool reuse, new shuffle)
+        assert r1.records == 2_000_000
+        assert eng.manager.gpu.pool.stats.used_bytes == 0  # all freed
+    finally:
+        eng.shutdown()
+
+
+def _ipc_worker(rank, world, driver_port, shm_dir, q):
+    try:
+        sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+        import torch
+        from sparkrdma_amd.conf import ShuffleConf
+        from sparkrdma_amd.engine import Engine
+        from sparkrdma_amd.partitioner import RangePartitioner
+        from sparkrdma_amd.workloads.terasort import TeraSort
+
+        conf = ShuffleConf(transport="ipc", hbm_pool_size=1 << 30,
+                           shm_dir=shm_dir, gpu_id=0)  # both ranks share GPU 0
+        eng = Engine(conf, rank=rank, world_size=world,
+                     driver_port=driver_port)
+        ts = TeraSort(eng, records_per_executor=500_000,
+                      partitions_per_executor=128, device="cuda",
+                      validate=True)
+        res = ts.run_step()
+        q.put((rank, res.records, res.remote_bytes))
+        eng.barrier()
+        eng.shutdown()
+    except BaseException as e:
+        import traceback
+        q.put((rank, f"ERROR: {e}\n{traceback.format_exc()}", 0))
+        raise
+
+
+def test_cross_process_ipc_shuffle(tmp_path):
+    """Two executor processes sharing one MI355X exchange HBM blocks via
+    hipIpc handles — validates the whole one-sided GPU path across a real
+    process boundary (the 8-GPU topology collapsed onto one device)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    procs = [ctx.Process(target=_ipc_worker,
+                         args=(r, 2, port, str(tmp_path), q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    remote_total = 0
+    for _ in range(2):
+        rank, records, remote = q.get(timeout=300)
+        assert not isinstance(records, str), f"rank {rank}: {records}"
+        assert records == 500_000
+        remote_total += remote
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    # each rank must have fetched ~half its data from the peer process
+    assert remote_total > 0
