@@ -43,11 +43,13 @@ def main():
               f"{n/t/1e9:6.2f} Grec/s  {n*rec_bytes/t/1e9:7.1f} GB/s payload")
 
     for passes, end_bit in ((4, 32), (7, 56), (8, 64)):
-        kk = keys.clone()
-        vv = vals.clone()
-        t = bench(lambda: sort_pairs(kk, vv, 0, end_bit))
-        print(f"sort {passes} passes (bits 0..{end_bit}): {t*1e3:8.2f} ms  "
-              f"{n/t/1e9:6.2f} Grec/s  {n*rec_bytes/t/1e9:7.1f} GB/s payload")
+        for osweep in (False, True):
+            kk = keys.clone()
+            vv = vals.clone()
+            t = bench(lambda: sort_pairs(kk, vv, 0, end_bit, onesweep=osweep))
+            tag = "onesweep" if osweep else "3-kernel"
+            print(f"sort {passes}p {tag} (bits 0..{end_bit}): {t*1e3:8.2f} ms  "
+                  f"{n/t/1e9:6.2f} Grec/s  {n*rec_bytes/t/1e9:7.1f} GB/s payload")
 
     # torch baseline for context
     t = bench(lambda: torch.sort(keys)[0])

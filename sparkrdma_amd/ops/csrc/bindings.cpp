@@ -62,4 +62,10 @@ PYBIND11_MODULE(_hipshuffle, m) {
         py::arg("n"), py::arg("start_bit"), py::arg("end_bit"), py::arg("ws"),
         py::arg("stream") = 0,
         py::call_guard<py::gil_scoped_release>());
+  m.def("onesweep_workspace_bytes", &hs::onesweep_workspace_bytes);
+  m.def("onesweep_sort_pairs_u64", &hs::onesweep_sort_pairs_u64,
+        py::arg("keys"), py::arg("vals"), py::arg("tmp_keys"),
+        py::arg("tmp_vals"), py::arg("n"), py::arg("start_bit"),
+        py::arg("end_bit"), py::arg("ws"), py::arg("stream") = 0,
+        py::call_guard<py::gil_scoped_release>());
 }

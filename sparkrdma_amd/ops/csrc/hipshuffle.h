@@ -38,5 +38,10 @@ size_t sort_workspace_bytes(uint32_t n);
 int sort_pairs_u64(uintptr_t keys, uintptr_t vals, uintptr_t tmp_keys,
                    uintptr_t tmp_vals, uint32_t n, int start_bit, int end_bit,
                    uintptr_t ws, uintptr_t stream);
+size_t onesweep_workspace_bytes(uint32_t n, int passes);
+int onesweep_sort_pairs_u64(uintptr_t keys, uintptr_t vals,
+                            uintptr_t tmp_keys, uintptr_t tmp_vals,
+                            uint32_t n, int start_bit, int end_bit,
+                            uintptr_t ws, uintptr_t stream);
 
 }  // namespace hipshuffle

@@ -337,6 +337,214 @@ __global__ __launch_bounds__(BLOCK) void radix_scatter_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Onesweep path (8-bit digits): one kernel per pass with decoupled
+// lookback — removes the per-pass hist read and the scan kernels.
+//
+// Status word per (block, digit): 2-bit flag | 30-bit count/prefix packed
+// in ONE u32, so a single relaxed agent-scope atomic is a complete
+// publish (no multi-location ordering; the guide's G16 concerns collapse
+// to single-word atomicity). Agent-scope atomics are cache-coherent
+// across XCDs. Blocks take an atomic ticket at start, so a block's
+// predecessors in lookback order are already launched -> no deadlock,
+// and the ticket determines the tile, so results are placement- and
+// timing-independent. Requires n < 2^30 (reduce-side sorts are
+// per-partition, well under; larger calls fall back to the 3-kernel path).
+
+constexpr uint32_t FLAG_AGG = 1u << 30;
+constexpr uint32_t FLAG_INC = 2u << 30;
+constexpr uint32_t FLAG_MASK = 3u << 30;
+constexpr uint32_t VAL_MASK = (1u << 30) - 1;
+
+// Global digit totals of EVERY pass in one read (digit counts are
+// order-independent, so pass k's totals can be computed from pass 0's
+// input).
+template <int MAX_PASSES>
+__global__ __launch_bounds__(BLOCK) void onesweep_hist_all_kernel(
+    const uint64_t* __restrict__ keys, uint32_t n, int start_bit, int passes,
+    uint32_t* __restrict__ totals /* [passes][256] */) {
+  __shared__ uint32_t cnt[NW][MAX_PASSES * 256];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  for (int i = tid; i < NW * MAX_PASSES * 256; i += BLOCK)
+    cnt[0][i] = 0;  // flat zero (cnt rows contiguous)
+  __syncthreads();
+  const uint64_t stride = (uint64_t)gridDim.x * BLOCK;
+  for (uint64_t e = (uint64_t)blockIdx.x * BLOCK + tid; e < n; e += stride) {
+    uint64_t k = keys[e];
+    for (int p = 0; p < passes; ++p) {
+      uint32_t d = (uint32_t)(k >> (start_bit + p * 8)) & 255;
+      atomicAdd(&cnt[wave][p * 256 + d], 1);
+    }
+  }
+  __syncthreads();
+  for (int i = tid; i < passes * 256; i += BLOCK) {
+    uint32_t s = 0;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) s += cnt[w][i];
+    if (s) atomicAdd(&totals[i], s);
+  }
+}
+
+template <bool HAS_VAL>
+__global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
+    const uint64_t* __restrict__ keys, const uint64_t* __restrict__ vals,
+    uint32_t n, int shift, uint32_t* __restrict__ desc /* [nb][256] */,
+    uint32_t* __restrict__ ticket,
+    const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst) {
+  constexpr int ND = 256;
+  extern __shared__ char smem_raw[];
+  uint64_t* exch = reinterpret_cast<uint64_t*>(smem_raw);
+  uint32_t* counters = reinterpret_cast<uint32_t*>(exch + TILE);
+  uint32_t* start = counters + NW * ND;
+  uint32_t* pref = start + ND;
+  uint32_t* sums = pref + ND;
+  uint32_t* vb_sh = sums + BLOCK;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (kWave - 1);
+  const int wave = tid >> 6;
+  if (tid == 0) *vb_sh = atomicAdd(ticket, 1);
+  for (int d = tid; d < NW * ND; d += BLOCK) counters[d] = 0;
+  __syncthreads();
+  const uint32_t b = *vb_sh;  // execution-ordered virtual block id
+  const uint64_t tile_start = (uint64_t)b * TILE;
+  const uint32_t tile_n =
+      (uint32_t)min((uint64_t)TILE, (uint64_t)n - tile_start);
+  uint32_t* my = counters + wave * ND;
+
+  // phase A: stable per-wave ranks (identical to radix_scatter_kernel)
+  uint64_t key_reg[ITEMS];
+  uint32_t rank_reg[ITEMS];
+  uint32_t dig_reg[ITEMS];
+  const uint64_t chunk = tile_start + (uint64_t)wave * (ITEMS * kWave);
+#pragma unroll 1
+  for (int i = 0; i < ITEMS; ++i) {
+    uint64_t e = chunk + (uint64_t)i * kWave + lane;
+    bool valid = e < n;
+    uint64_t k = valid ? keys[e] : 0;
+    key_reg[i] = k;
+    uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
+    dig_reg[i] = d;
+    uint64_t vm = __ballot(valid);
+    uint32_t r = 0;
+    if (valid) {
+      uint64_t match = match_lanes<8>(d, vm);
+      uint64_t lt = (1ull << lane) - 1;
+      uint32_t rank_in_iter = (uint32_t)__popcll(match & lt);
+      uint32_t c = my[d];
+      r = c + rank_in_iter;
+      if (rank_in_iter == 0) my[d] = c + (uint32_t)__popcll(match);
+    }
+    rank_reg[i] = r;
+  }
+  __syncthreads();
+
+  // per-digit wave-exclusive scan + block totals into start[]
+  for (int d = tid; d < ND; d += BLOCK) {
+    uint32_t run = 0;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) {
+      uint32_t t = counters[w * ND + d];
+      counters[w * ND + d] = run;
+      run += t;
+    }
+    start[d] = run;
+  }
+  __syncthreads();
+
+  // publish aggregate (or inclusive for block 0), then lookback
+  if (tid < ND) {
+    uint32_t total = start[tid];
+    uint32_t* slot = &desc[(uint64_t)b * ND + tid];
+    if (b == 0) {
+      __hip_atomic_store(slot, FLAG_INC | total, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      pref[tid] = 0;
+    } else {
+      __hip_atomic_store(slot, FLAG_AGG | total, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      uint32_t run = 0;
+      for (int64_t j = (int64_t)b - 1; j >= 0;) {
+        uint32_t w = __hip_atomic_load(&desc[(uint64_t)j * ND + tid],
+                                       __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+        if ((w & FLAG_MASK) == 0) {
+          __builtin_amdgcn_s_sleep(1);
+          continue;
+        }
+        run += w & VAL_MASK;
+        if ((w & FLAG_MASK) == FLAG_INC) break;
+        --j;
+      }
+      __hip_atomic_store(slot, FLAG_INC | (run + total), __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      pref[tid] = run;
+    }
+  }
+  // block-local digit starts (exclusive scan of totals)
+  block_exscan(start, sums, ND);  // includes the needed __syncthreads
+
+  // LDS exchange + linear write-out (same as radix_scatter_kernel)
+#pragma unroll 1
+  for (int i = 0; i < ITEMS; ++i) {
+    uint64_t e = chunk + (uint64_t)i * kWave + lane;
+    if (e < n) {
+      uint32_t d = dig_reg[i];
+      uint32_t j = start[d] + my[d] + rank_reg[i];
+      rank_reg[i] = j;
+      exch[j] = key_reg[i];
+    }
+  }
+  __syncthreads();
+  uint32_t out_d[ITEMS];
+  uint32_t out_off[ITEMS];
+#pragma unroll 1
+  for (int i = 0; i < ITEMS; ++i) {
+    uint32_t j = i * BLOCK + tid;
+    if (j < tile_n) {
+      uint64_t k = exch[j];
+      uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
+      uint32_t off = pref[d] + (j - start[d]);
+      out_d[i] = d;
+      out_off[i] = off;
+      reinterpret_cast<uint64_t*>(key_dst[d])[off] = k;
+    }
+  }
+  if (HAS_VAL) {
+    __syncthreads();
+#pragma unroll 1
+    for (int i = 0; i < ITEMS; ++i) {
+      uint64_t e = chunk + (uint64_t)i * kWave + lane;
+      if (e < n) exch[rank_reg[i]] = vals[e];
+    }
+    __syncthreads();
+#pragma unroll 1
+    for (int i = 0; i < ITEMS; ++i) {
+      uint32_t j = i * BLOCK + tid;
+      if (j < tile_n)
+        reinterpret_cast<uint64_t*>(val_dst[out_d[i]])[out_off[i]] = exch[j];
+    }
+  }
+}
+
+// digit bases for pass p from the all-pass totals
+__global__ void onesweep_digit_bases_kernel(
+    const uint32_t* __restrict__ totals /* [pass][256] */, int pass,
+    uint64_t out_keys, uint64_t out_vals, uint64_t* __restrict__ key_dst,
+    uint64_t* __restrict__ val_dst) {
+  __shared__ uint32_t arr[256];
+  __shared__ uint32_t sums[BLOCK];
+  const int tid = threadIdx.x;
+  if (tid < 256) arr[tid] = totals[pass * 256 + tid];
+  __syncthreads();
+  block_exscan(arr, sums, 256);
+  if (tid < 256) {
+    key_dst[tid] = out_keys + (uint64_t)arr[tid] * 8;
+    val_dst[tid] = out_vals ? out_vals + (uint64_t)arr[tid] * 8 : 0;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // host wrappers
 
 static inline uint32_t num_tiles(uint32_t n) {
@@ -461,6 +669,75 @@ void radix_scatter(uintptr_t keys, uintptr_t vals, uint32_t n, int shift,
 size_t sort_workspace_bytes(uint32_t n) {
   return radix_hist_bytes(n, 8) + radix_scan_ws_bytes(n, 8) + 256 * 4 +
          256 * 8 * 2;
+}
+
+// Onesweep sort: ws layout = totals u32[passes*256] | key_dst u64[256] |
+// val_dst u64[256] | ticket u32 (+pad) | desc u32[nb*256].
+size_t onesweep_workspace_bytes(uint32_t n, int passes) {
+  uint32_t nb = num_tiles(n);
+  return (size_t)passes * 256 * 4 + 256 * 8 * 2 + 16 +
+         (size_t)nb * 256 * 4;
+}
+
+int onesweep_sort_pairs_u64(uintptr_t keys, uintptr_t vals,
+                            uintptr_t tmp_keys, uintptr_t tmp_vals,
+                            uint32_t n, int start_bit, int end_bit,
+                            uintptr_t ws, uintptr_t stream) {
+  if (n >= (1u << 30))
+    throw std::runtime_error("onesweep requires n < 2^30");
+  auto s = reinterpret_cast<hipStream_t>(stream);
+  uint32_t nb = num_tiles(n);
+  int passes = (end_bit - start_bit + 7) / 8;
+  uint32_t* totals = reinterpret_cast<uint32_t*>(ws);
+  uint64_t* key_dst = reinterpret_cast<uint64_t*>(totals + (size_t)passes * 256);
+  uint64_t* val_dst = key_dst + 256;
+  uint32_t* ticket = reinterpret_cast<uint32_t*>(val_dst + 256);
+  uint32_t* desc = ticket + 4;
+  HIP_CHECK(hipMemsetAsync(totals, 0, (size_t)passes * 256 * 4, s));
+  uint32_t hist_grid = nb < 2048 ? (nb ? nb : 1) : 2048;
+  hipLaunchKernelGGL(onesweep_hist_all_kernel<8>, dim3(hist_grid),
+                     dim3(BLOCK), 0, s,
+                     reinterpret_cast<const uint64_t*>(keys), n, start_bit,
+                     passes, totals);
+  HIP_CHECK(hipGetLastError());
+  size_t lds = (size_t)TILE * 8 + (size_t)NW * 256 * 4 + 256 * 4 * 2 +
+               BLOCK * 4 + 16;
+  static bool attr_set = false;
+  if (!attr_set && lds > 64 * 1024) {
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&onesweep_pass_kernel<true>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&onesweep_pass_kernel<false>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+    attr_set = true;
+  }
+  uintptr_t src_k = keys, src_v = vals, dst_k = tmp_keys, dst_v = tmp_vals;
+  int cur = 0;
+  for (int p = 0; p < passes; ++p) {
+    hipLaunchKernelGGL(onesweep_digit_bases_kernel, dim3(1), dim3(BLOCK), 0,
+                       s, totals, p, (uint64_t)dst_k, (uint64_t)dst_v,
+                       key_dst, val_dst);
+    HIP_CHECK(hipGetLastError());
+    HIP_CHECK(hipMemsetAsync(ticket, 0, 16, s));
+    HIP_CHECK(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 4, s));
+    if (vals) {
+      hipLaunchKernelGGL(onesweep_pass_kernel<true>, dim3(nb), dim3(BLOCK),
+                         lds, s, reinterpret_cast<const uint64_t*>(src_k),
+                         reinterpret_cast<const uint64_t*>(src_v), n,
+                         start_bit + p * 8, desc, ticket, key_dst, val_dst);
+    } else {
+      hipLaunchKernelGGL(onesweep_pass_kernel<false>, dim3(nb), dim3(BLOCK),
+                         lds, s, reinterpret_cast<const uint64_t*>(src_k),
+                         nullptr, n, start_bit + p * 8, desc, ticket, key_dst,
+                         val_dst);
+    }
+    HIP_CHECK(hipGetLastError());
+    std::swap(src_k, dst_k);
+    std::swap(src_v, dst_v);
+    cur ^= 1;
+  }
+  return cur;
 }
 
 int sort_pairs_u64(uintptr_t keys, uintptr_t vals, uintptr_t tmp_keys,

@@ -69,12 +69,15 @@ def radix_partition(keys: torch.Tensor, vals: Optional[torch.Tensor],
 
 
 def sort_pairs(keys: torch.Tensor, vals: Optional[torch.Tensor] = None,
-               start_bit: int = 0, end_bit: int = 64
+               start_bit: int = 0, end_bit: int = 64,
+               onesweep: Optional[bool] = None
                ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
     """LSD radix sort of u64 keys (and optional u64 payload) on device.
 
     Sorts by bits [start_bit, end_bit) — callers that range-partitioned by
     the top bits pass end_bit = 64 - log2(R) and save whole passes.
+    Uses the decoupled-lookback onesweep path (one kernel per pass) when
+    n < 2^30; the 3-kernel hist/scan/scatter path otherwise.
     """
     m = load()
     n = keys.numel()
@@ -83,13 +86,25 @@ def sort_pairs(keys: torch.Tensor, vals: Optional[torch.Tensor] = None,
     dev = keys.device
     tmp_k = torch.empty_like(keys)
     tmp_v = torch.empty_like(vals) if vals is not None else None
-    ws = torch.empty(m.sort_workspace_bytes(n), dtype=torch.uint8, device=dev)
     end_bit = start_bit + ((end_bit - start_bit + 7) // 8) * 8  # whole digits
     end_bit = min(end_bit, 64)
-    res = m.sort_pairs_u64(
-        keys.data_ptr(), vals.data_ptr() if vals is not None else 0,
-        tmp_k.data_ptr(), tmp_v.data_ptr() if tmp_v is not None else 0,
-        n, start_bit, end_bit, ws.data_ptr(), _stream())
+    passes = (end_bit - start_bit) // 8
+    if onesweep is None:
+        onesweep = n < (1 << 30)
+    if onesweep:
+        ws = torch.empty(m.onesweep_workspace_bytes(n, passes),
+                         dtype=torch.uint8, device=dev)
+        res = m.onesweep_sort_pairs_u64(
+            keys.data_ptr(), vals.data_ptr() if vals is not None else 0,
+            tmp_k.data_ptr(), tmp_v.data_ptr() if tmp_v is not None else 0,
+            n, start_bit, end_bit, ws.data_ptr(), _stream())
+    else:
+        ws = torch.empty(m.sort_workspace_bytes(n), dtype=torch.uint8,
+                         device=dev)
+        res = m.sort_pairs_u64(
+            keys.data_ptr(), vals.data_ptr() if vals is not None else 0,
+            tmp_k.data_ptr(), tmp_v.data_ptr() if tmp_v is not None else 0,
+            n, start_bit, end_bit, ws.data_ptr(), _stream())
     if res == 0:
         return keys, vals
     return tmp_k, tmp_v

@@ -43,16 +43,32 @@ def test_radix_partition_counts_and_order(hs, n, nbits):
 
 
 @pytest.mark.parametrize("n", [1, 1000, 4096, 1_000_000, 4_000_000])
-def test_sort_pairs_full(hs, n):
+@pytest.mark.parametrize("onesweep", [False, True])
+def test_sort_pairs_full(hs, n, onesweep):
     from sparkrdma_amd.ops.radix import sort_pairs
     keys = rand_keys(n, seed=n)
     vals = keys.clone()
-    k_sorted, v_sorted = sort_pairs(keys, vals)
+    k_sorted, v_sorted = sort_pairs(keys, vals, onesweep=onesweep)
     torch.cuda.synchronize()
     got = k_sorted.cpu().numpy().view(np.uint64)
     want = np.sort(rand_keys(n, seed=n).cpu().numpy().view(np.uint64))
     assert np.array_equal(got, want)
     assert np.array_equal(v_sorted.cpu().numpy(), k_sorted.cpu().numpy())
+
+
+def test_onesweep_matches_3kernel_repeated(hs):
+    """Race screen: the lookback path must be deterministic across runs
+    and identical to the 3-kernel path."""
+    from sparkrdma_amd.ops.radix import sort_pairs
+    n = 2_000_000
+    keys = rand_keys(n, seed=77)
+    vals = rand_keys(n, seed=78)
+    ref_k, ref_v = sort_pairs(keys.clone(), vals.clone(), onesweep=False)
+    for trial in range(5):
+        k, v = sort_pairs(keys.clone(), vals.clone(), onesweep=True)
+        torch.cuda.synchronize()
+        assert torch.equal(k, ref_k), f"trial {trial}: keys diverge"
+        assert torch.equal(v, ref_v), f"trial {trial}: vals diverge"
 
 
 def test_sort_partial_bits(hs):
