@@ -385,7 +385,7 @@ __global__ __launch_bounds__(BLOCK) void onesweep_hist_all_kernel(
   }
 }
 
-template <bool HAS_VAL>
+template <bool HAS_VAL, int IT>
 __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
     const uint64_t* __restrict__ keys, const uint64_t* __restrict__ vals,
     uint32_t n, int shift, uint32_t* __restrict__ desc /* [nb][256] */,
@@ -393,12 +393,14 @@ __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
     const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst) {
   constexpr int ND = 256;
   extern __shared__ char smem_raw[];
+  constexpr int TILE_T = BLOCK * IT;
   uint64_t* exch = reinterpret_cast<uint64_t*>(smem_raw);
-  uint32_t* counters = reinterpret_cast<uint32_t*>(exch + TILE);
+  uint32_t* counters = reinterpret_cast<uint32_t*>(exch + TILE_T);
   uint32_t* start = counters + NW * ND;
   uint32_t* pref = start + ND;
   uint32_t* sums = pref + ND;
   uint32_t* vb_sh = sums + BLOCK;
+  uint8_t* dsort = reinterpret_cast<uint8_t*>(vb_sh + 4);  // [TILE_T]
 
   const int tid = threadIdx.x;
   const int lane = tid & (kWave - 1);
@@ -407,24 +409,22 @@ __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
   for (int d = tid; d < NW * ND; d += BLOCK) counters[d] = 0;
   __syncthreads();
   const uint32_t b = *vb_sh;  // execution-ordered virtual block id
-  const uint64_t tile_start = (uint64_t)b * TILE;
+  const uint64_t tile_start = (uint64_t)b * TILE_T;
   const uint32_t tile_n =
-      (uint32_t)min((uint64_t)TILE, (uint64_t)n - tile_start);
+      (uint32_t)min((uint64_t)TILE_T, (uint64_t)n - tile_start);
   uint32_t* my = counters + wave * ND;
 
-  // phase A: stable per-wave ranks (identical to radix_scatter_kernel)
-  uint64_t key_reg[ITEMS];
-  uint32_t rank_reg[ITEMS];
-  uint32_t dig_reg[ITEMS];
-  const uint64_t chunk = tile_start + (uint64_t)wave * (ITEMS * kWave);
-#pragma unroll 1
-  for (int i = 0; i < ITEMS; ++i) {
+  // phase A: stable per-wave ranks; digit(8b)|rank(16b) packed per elem
+  uint64_t key_reg[IT];
+  uint32_t digrank[IT];
+  const uint64_t chunk = tile_start + (uint64_t)wave * (IT * kWave);
+#pragma unroll
+  for (int i = 0; i < IT; ++i) {
     uint64_t e = chunk + (uint64_t)i * kWave + lane;
     bool valid = e < n;
     uint64_t k = valid ? keys[e] : 0;
     key_reg[i] = k;
     uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
-    dig_reg[i] = d;
     uint64_t vm = __ballot(valid);
     uint32_t r = 0;
     if (valid) {
@@ -435,7 +435,7 @@ __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
       r = c + rank_in_iter;
       if (rank_in_iter == 0) my[d] = c + (uint32_t)__popcll(match);
     }
-    rank_reg[i] = r;
+    digrank[i] = (d << 16) | r;
   }
   __syncthreads();
 
@@ -484,45 +484,46 @@ __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
   // block-local digit starts (exclusive scan of totals)
   block_exscan(start, sums, ND);  // includes the needed __syncthreads
 
-  // LDS exchange + linear write-out (same as radix_scatter_kernel)
-#pragma unroll 1
-  for (int i = 0; i < ITEMS; ++i) {
+  // LDS exchange + linear write-out; digit per sorted slot kept in a
+  // byte array so the val phase re-derives (d, off) without registers
+#pragma unroll
+  for (int i = 0; i < IT; ++i) {
     uint64_t e = chunk + (uint64_t)i * kWave + lane;
     if (e < n) {
-      uint32_t d = dig_reg[i];
-      uint32_t j = start[d] + my[d] + rank_reg[i];
-      rank_reg[i] = j;
+      uint32_t d = digrank[i] >> 16;
+      uint32_t j = start[d] + my[d] + (digrank[i] & 0xFFFF);
+      digrank[i] = j;  // reuse: local sorted position
       exch[j] = key_reg[i];
     }
   }
   __syncthreads();
-  uint32_t out_d[ITEMS];
-  uint32_t out_off[ITEMS];
-#pragma unroll 1
-  for (int i = 0; i < ITEMS; ++i) {
+#pragma unroll
+  for (int i = 0; i < IT; ++i) {
     uint32_t j = i * BLOCK + tid;
     if (j < tile_n) {
       uint64_t k = exch[j];
       uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
       uint32_t off = pref[d] + (j - start[d]);
-      out_d[i] = d;
-      out_off[i] = off;
+      dsort[j] = (uint8_t)d;
       reinterpret_cast<uint64_t*>(key_dst[d])[off] = k;
     }
   }
   if (HAS_VAL) {
     __syncthreads();
 #pragma unroll 1
-    for (int i = 0; i < ITEMS; ++i) {
+    for (int i = 0; i < IT; ++i) {
       uint64_t e = chunk + (uint64_t)i * kWave + lane;
-      if (e < n) exch[rank_reg[i]] = vals[e];
+      if (e < n) exch[digrank[i]] = vals[e];
     }
     __syncthreads();
 #pragma unroll 1
-    for (int i = 0; i < ITEMS; ++i) {
+    for (int i = 0; i < IT; ++i) {
       uint32_t j = i * BLOCK + tid;
-      if (j < tile_n)
-        reinterpret_cast<uint64_t*>(val_dst[out_d[i]])[out_off[i]] = exch[j];
+      if (j < tile_n) {
+        uint32_t d = dsort[j];
+        uint32_t off = pref[d] + (j - start[d]);
+        reinterpret_cast<uint64_t*>(val_dst[d])[off] = exch[j];
+      }
     }
   }
 }
@@ -673,8 +674,15 @@ size_t sort_workspace_bytes(uint32_t n) {
 
 // Onesweep sort: ws layout = totals u32[passes*256] | key_dst u64[256] |
 // val_dst u64[256] | ticket u32 (+pad) | desc u32[nb*256].
+constexpr int OS_ITEMS = 16;  // 4096-elem tiles (32 measured slower: occupancy loss beats longer digit runs)
+constexpr int OS_TILE = BLOCK * OS_ITEMS;
+
+static inline uint32_t os_num_tiles(uint32_t n) {
+  return (uint32_t)(((uint64_t)n + OS_TILE - 1) / OS_TILE);
+}
+
 size_t onesweep_workspace_bytes(uint32_t n, int passes) {
-  uint32_t nb = num_tiles(n);
+  uint32_t nb = os_num_tiles(n);
   return (size_t)passes * 256 * 4 + 256 * 8 * 2 + 16 +
          (size_t)nb * 256 * 4;
 }
@@ -686,7 +694,7 @@ int onesweep_sort_pairs_u64(uintptr_t keys, uintptr_t vals,
   if (n >= (1u << 30))
     throw std::runtime_error("onesweep requires n < 2^30");
   auto s = reinterpret_cast<hipStream_t>(stream);
-  uint32_t nb = num_tiles(n);
+  uint32_t nb = os_num_tiles(n);
   int passes = (end_bit - start_bit + 7) / 8;
   uint32_t* totals = reinterpret_cast<uint32_t*>(ws);
   uint64_t* key_dst = reinterpret_cast<uint64_t*>(totals + (size_t)passes * 256);
@@ -694,21 +702,21 @@ int onesweep_sort_pairs_u64(uintptr_t keys, uintptr_t vals,
   uint32_t* ticket = reinterpret_cast<uint32_t*>(val_dst + 256);
   uint32_t* desc = ticket + 4;
   HIP_CHECK(hipMemsetAsync(totals, 0, (size_t)passes * 256 * 4, s));
-  uint32_t hist_grid = nb < 2048 ? (nb ? nb : 1) : 2048;
+  uint32_t hist_grid = nb < 1024 ? (nb ? nb : 1) : 1024;
   hipLaunchKernelGGL(onesweep_hist_all_kernel<8>, dim3(hist_grid),
                      dim3(BLOCK), 0, s,
                      reinterpret_cast<const uint64_t*>(keys), n, start_bit,
                      passes, totals);
   HIP_CHECK(hipGetLastError());
-  size_t lds = (size_t)TILE * 8 + (size_t)NW * 256 * 4 + 256 * 4 * 2 +
-               BLOCK * 4 + 16;
+  size_t lds = (size_t)OS_TILE * 8 + (size_t)NW * 256 * 4 + 256 * 4 * 2 +
+               BLOCK * 4 + 16 + OS_TILE;
   static bool attr_set = false;
   if (!attr_set && lds > 64 * 1024) {
     (void)hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&onesweep_pass_kernel<true>),
+        reinterpret_cast<const void*>(&onesweep_pass_kernel<true, OS_ITEMS>),
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
     (void)hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&onesweep_pass_kernel<false>),
+        reinterpret_cast<const void*>(&onesweep_pass_kernel<false, OS_ITEMS>),
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
     attr_set = true;
   }
@@ -722,12 +730,12 @@ int onesweep_sort_pairs_u64(uintptr_t keys, uintptr_t vals,
     HIP_CHECK(hipMemsetAsync(ticket, 0, 16, s));
     HIP_CHECK(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 4, s));
     if (vals) {
-      hipLaunchKernelGGL(onesweep_pass_kernel<true>, dim3(nb), dim3(BLOCK),
+      hipLaunchKernelGGL((onesweep_pass_kernel<true, OS_ITEMS>), dim3(nb), dim3(BLOCK),
                          lds, s, reinterpret_cast<const uint64_t*>(src_k),
                          reinterpret_cast<const uint64_t*>(src_v), n,
                          start_bit + p * 8, desc, ticket, key_dst, val_dst);
     } else {
-      hipLaunchKernelGGL(onesweep_pass_kernel<false>, dim3(nb), dim3(BLOCK),
+      hipLaunchKernelGGL((onesweep_pass_kernel<false, OS_ITEMS>), dim3(nb), dim3(BLOCK),
                          lds, s, reinterpret_cast<const uint64_t*>(src_k),
                          nullptr, n, start_bit + p * 8, desc, ticket, key_dst,
                          val_dst);
