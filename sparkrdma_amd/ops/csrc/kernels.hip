@@ -78,6 +78,15 @@ __device__ void block_exscan(uint32_t* arr, uint32_t* sums, int nd) {
   __syncthreads();
 }
 
+// splitmix64 finalizer step — bit-identical to partitioner.HashPartitioner
+// so GPU hash partitioning matches the CPU oracle exactly.
+__device__ __forceinline__ uint64_t hash_mix64(uint64_t k) {
+  k ^= k >> 33;
+  k *= 0xFF51AFD7ED558CCDull;
+  k ^= k >> 33;
+  return k;
+}
+
 // 64-lane multi-split: lanes with equal `digit` (among `validmask` lanes).
 template <int NBITS>
 __device__ __forceinline__ uint64_t match_lanes(uint32_t digit,
@@ -99,7 +108,7 @@ __device__ __forceinline__ uint64_t match_lanes(uint32_t digit,
 template <int NBITS>
 __global__ __launch_bounds__(BLOCK) void radix_hist_kernel(
     const uint64_t* __restrict__ keys, uint32_t n, int shift,
-    uint32_t* __restrict__ hist, uint32_t nb) {
+    uint32_t* __restrict__ hist, uint32_t nb, int hash_mix) {
   constexpr int ND = 1 << NBITS;
   extern __shared__ char smem_raw[];
   uint32_t* counters = reinterpret_cast<uint32_t*>(smem_raw);  // [NW][ND]
@@ -118,6 +127,7 @@ __global__ __launch_bounds__(BLOCK) void radix_hist_kernel(
     uint64_t e = chunk + (uint64_t)i * kWave + lane;
     bool valid = e < n;
     uint64_t k = valid ? keys[e] : 0;
+    if (hash_mix) k = hash_mix64(k);
     uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
     uint64_t vm = __ballot(valid);
     if (valid) {
@@ -225,7 +235,8 @@ template <int NBITS, bool HAS_VAL>
 __global__ __launch_bounds__(BLOCK) void radix_scatter_kernel(
     const uint64_t* __restrict__ keys, const uint64_t* __restrict__ vals,
     uint32_t n, int shift, const uint32_t* __restrict__ hist, uint32_t nb,
-    const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst) {
+    const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst,
+    int hash_mix) {
   constexpr int ND = 1 << NBITS;
   extern __shared__ char smem_raw[];
   // layout: exchange u64[TILE] | counters u32[NW][ND] | start u32[ND]
@@ -261,7 +272,7 @@ __global__ __launch_bounds__(BLOCK) void radix_scatter_kernel(
     bool valid = e < n;
     uint64_t k = valid ? keys[e] : 0;
     key_reg[i] = k;
-    uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
+    uint32_t d = (uint32_t)((hash_mix ? hash_mix64(k) : k) >> shift) & (ND - 1);
     dig_reg[i] = d;
     uint64_t vm = __ballot(valid);
     uint32_t r = 0;
@@ -312,7 +323,7 @@ __global__ __launch_bounds__(BLOCK) void radix_scatter_kernel(
     uint32_t j = i * BLOCK + tid;
     if (j < tile_n) {
       uint64_t k = exch[j];
-      uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
+      uint32_t d = (uint32_t)((hash_mix ? hash_mix64(k) : k) >> shift) & (ND - 1);
       uint32_t off = pref[d] + (j - start[d]);
       out_d[i] = d;
       out_off[i] = off;
@@ -546,6 +557,97 @@ __global__ void onesweep_digit_bases_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Sort-merge join (merge kernel for SQL sort-merge join).
+// Both sides radix-sorted first; the merge is per-A-row binary search in
+// sorted B (log nb coherent loads — neighbouring threads probe
+// neighbouring B ranges, so the walk stays in L2), then pair emission at
+// scanned offsets.
+
+__device__ __forceinline__ uint32_t lower_bound_u64(
+    const uint64_t* __restrict__ arr, uint32_t n, uint64_t key) {
+  uint32_t lo = 0, hi = n;
+  while (lo < hi) {
+    uint32_t mid = (lo + hi) >> 1;
+    if (arr[mid] < key) lo = mid + 1; else hi = mid;
+  }
+  return lo;
+}
+
+__global__ __launch_bounds__(BLOCK) void join_count_kernel(
+    const uint64_t* __restrict__ a_keys, uint32_t na,
+    const uint64_t* __restrict__ b_keys, uint32_t nb_,
+    uint32_t* __restrict__ counts, uint32_t* __restrict__ lo_idx) {
+  const uint64_t stride = (uint64_t)gridDim.x * BLOCK;
+  for (uint64_t i = (uint64_t)blockIdx.x * BLOCK + threadIdx.x; i < na;
+       i += stride) {
+    uint64_t k = a_keys[i];
+    uint32_t lo = lower_bound_u64(b_keys, nb_, k);
+    uint32_t hi = lo;
+    while (hi < nb_ && b_keys[hi] == k) ++hi;  // runs are short for ~unique keys
+    counts[i] = hi - lo;
+    lo_idx[i] = lo;
+  }
+}
+
+__global__ __launch_bounds__(BLOCK) void join_emit_kernel(
+    const uint64_t* __restrict__ a_keys, const uint64_t* __restrict__ a_vals,
+    uint32_t na, const uint64_t* __restrict__ b_vals,
+    const uint32_t* __restrict__ counts, const uint32_t* __restrict__ lo_idx,
+    const uint64_t* __restrict__ offsets,
+    uint64_t* __restrict__ out_key, uint64_t* __restrict__ out_a,
+    uint64_t* __restrict__ out_b) {
+  const uint64_t stride = (uint64_t)gridDim.x * BLOCK;
+  for (uint64_t i = (uint64_t)blockIdx.x * BLOCK + threadIdx.x; i < na;
+       i += stride) {
+    uint32_t cnt = counts[i];
+    if (!cnt) continue;
+    uint64_t off = offsets[i];
+    uint32_t lo = lo_idx[i];
+    uint64_t k = a_keys[i];
+    uint64_t av = a_vals ? a_vals[i] : 0;
+    for (uint32_t j = 0; j < cnt; ++j) {
+      if (out_key) out_key[off + j] = k;
+      if (out_a) out_a[off + j] = av;
+      out_b[off + j] = b_vals[lo + j];
+    }
+  }
+}
+
+static inline uint32_t join_grid(uint32_t n) {
+  uint32_t g = (n + BLOCK - 1) / BLOCK;
+  return g > 2048 ? 2048 : (g ? g : 1);
+}
+
+void join_count(uintptr_t a_keys, uint32_t na, uintptr_t b_keys, uint32_t nb_,
+                uintptr_t counts, uintptr_t lo_idx, uintptr_t stream) {
+  auto s = reinterpret_cast<hipStream_t>(stream);
+  hipLaunchKernelGGL(join_count_kernel, dim3(join_grid(na)), dim3(BLOCK), 0,
+                     s, reinterpret_cast<const uint64_t*>(a_keys), na,
+                     reinterpret_cast<const uint64_t*>(b_keys), nb_,
+                     reinterpret_cast<uint32_t*>(counts),
+                     reinterpret_cast<uint32_t*>(lo_idx));
+  HIP_CHECK(hipGetLastError());
+}
+
+void join_emit(uintptr_t a_keys, uintptr_t a_vals, uint32_t na,
+               uintptr_t b_vals, uintptr_t counts, uintptr_t lo_idx,
+               uintptr_t offsets, uintptr_t out_key, uintptr_t out_a,
+               uintptr_t out_b, uintptr_t stream) {
+  auto s = reinterpret_cast<hipStream_t>(stream);
+  hipLaunchKernelGGL(join_emit_kernel, dim3(join_grid(na)), dim3(BLOCK), 0, s,
+                     reinterpret_cast<const uint64_t*>(a_keys),
+                     reinterpret_cast<const uint64_t*>(a_vals), na,
+                     reinterpret_cast<const uint64_t*>(b_vals),
+                     reinterpret_cast<const uint32_t*>(counts),
+                     reinterpret_cast<const uint32_t*>(lo_idx),
+                     reinterpret_cast<const uint64_t*>(offsets),
+                     reinterpret_cast<uint64_t*>(out_key),
+                     reinterpret_cast<uint64_t*>(out_a),
+                     reinterpret_cast<uint64_t*>(out_b));
+  HIP_CHECK(hipGetLastError());
+}
+
+// ---------------------------------------------------------------------------
 // host wrappers
 
 static inline uint32_t num_tiles(uint32_t n) {
@@ -558,11 +660,11 @@ size_t radix_hist_bytes(uint32_t n, int nbits) {
 
 template <int NBITS>
 static void hist_launch(const uint64_t* keys, uint32_t n, int shift,
-                        uint32_t* hist, hipStream_t s) {
+                        uint32_t* hist, hipStream_t s, int hash_mix = 0) {
   uint32_t nb = num_tiles(n);
   size_t lds = (size_t)NW * (1 << NBITS) * 4;
   hipLaunchKernelGGL(radix_hist_kernel<NBITS>, dim3(nb), dim3(BLOCK), lds, s,
-                     keys, n, shift, hist, nb);
+                     keys, n, shift, hist, nb, hash_mix);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -570,7 +672,7 @@ template <int NBITS, bool HAS_VAL>
 static void scatter_launch(const uint64_t* keys, const uint64_t* vals,
                            uint32_t n, int shift, const uint32_t* hist,
                            const uint64_t* key_dst, const uint64_t* val_dst,
-                           hipStream_t s) {
+                           hipStream_t s, int hash_mix = 0) {
   constexpr int ND = 1 << NBITS;
   uint32_t nb = num_tiles(n);
   size_t lds = (size_t)TILE * 8 + (size_t)NW * ND * 4 + (size_t)ND * 4 * 2 +
@@ -584,7 +686,7 @@ static void scatter_launch(const uint64_t* keys, const uint64_t* vals,
     attr_set[NBITS] = true;
   }
   hipLaunchKernelGGL(kfn, dim3(nb), dim3(BLOCK), lds, s, keys, vals, n, shift,
-                     hist, nb, key_dst, val_dst);
+                     hist, nb, key_dst, val_dst, hash_mix);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -603,10 +705,10 @@ static void scatter_launch(const uint64_t* keys, const uint64_t* vals,
   }
 
 void radix_hist(uintptr_t keys, uint32_t n, int shift, int nbits,
-                uintptr_t hist, uintptr_t stream) {
+                uintptr_t hist, uintptr_t stream, int hash_mix) {
   auto s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_NBITS(nbits, hist_launch, reinterpret_cast<const uint64_t*>(keys),
-                 n, shift, reinterpret_cast<uint32_t*>(hist), s);
+                 n, shift, reinterpret_cast<uint32_t*>(hist), s, hash_mix);
 }
 
 size_t radix_scan_ws_bytes(uint32_t n, int nbits) {
@@ -645,22 +747,24 @@ void radix_scan(uintptr_t hist, uint32_t n, int nbits, uintptr_t totals,
 template <int NBITS>
 static void scatter_hv(const uint64_t* keys, const uint64_t* vals, uint32_t n,
                        int shift, const uint32_t* hist, const uint64_t* kd,
-                       const uint64_t* vd, hipStream_t s) {
+                       const uint64_t* vd, hipStream_t s, int hash_mix = 0) {
   if (vals)
-    scatter_launch<NBITS, true>(keys, vals, n, shift, hist, kd, vd, s);
+    scatter_launch<NBITS, true>(keys, vals, n, shift, hist, kd, vd, s,
+                                hash_mix);
   else
-    scatter_launch<NBITS, false>(keys, nullptr, n, shift, hist, kd, nullptr, s);
+    scatter_launch<NBITS, false>(keys, nullptr, n, shift, hist, kd, nullptr,
+                                 s, hash_mix);
 }
 
 void radix_scatter(uintptr_t keys, uintptr_t vals, uint32_t n, int shift,
                    int nbits, uintptr_t hist, uintptr_t key_dst,
-                   uintptr_t val_dst, uintptr_t stream) {
+                   uintptr_t val_dst, uintptr_t stream, int hash_mix) {
   auto s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_NBITS(nbits, scatter_hv, reinterpret_cast<const uint64_t*>(keys),
                  reinterpret_cast<const uint64_t*>(vals), n, shift,
                  reinterpret_cast<const uint32_t*>(hist),
                  reinterpret_cast<const uint64_t*>(key_dst),
-                 reinterpret_cast<const uint64_t*>(val_dst), s);
+                 reinterpret_cast<const uint64_t*>(val_dst), s, hash_mix);
 }
 
 // Full LSD sort of (keys[, vals]) by bits [start_bit, end_bit).

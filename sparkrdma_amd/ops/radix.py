@@ -21,7 +21,8 @@ def _stream() -> int:
 def radix_partition(keys: torch.Tensor, vals: Optional[torch.Tensor],
                     nbits: int, shift: Optional[int] = None,
                     key_dst: Optional[torch.Tensor] = None,
-                    val_dst: Optional[torch.Tensor] = None
+                    val_dst: Optional[torch.Tensor] = None,
+                    hash_mix: bool = False
                     ) -> Tuple[torch.Tensor, torch.Tensor, Optional[torch.Tensor]]:
     """One bucket-scatter pass: digit = (key >> shift) & (2^nbits - 1).
 
@@ -46,8 +47,11 @@ def radix_partition(keys: torch.Tensor, vals: Optional[torch.Tensor],
     scan_ws = torch.empty(m.radix_scan_ws_bytes(n, nbits_eff) // 4,
                           dtype=torch.int32, device=dev)
     totals = torch.empty(nd, dtype=torch.int32, device=dev)
+    if hash_mix and nbits_eff != nbits:
+        raise ValueError("hash partitioning requires nbits >= 4")
     s = _stream()
-    m.radix_hist(keys.data_ptr(), n, shift, nbits_eff, hist.data_ptr(), s)
+    m.radix_hist(keys.data_ptr(), n, shift, nbits_eff, hist.data_ptr(), s,
+                 int(hash_mix))
     m.radix_scan(hist.data_ptr(), n, nbits_eff, totals.data_ptr(),
                  scan_ws.data_ptr(), s)
     keys_out = vals_out = None
@@ -64,7 +68,7 @@ def radix_partition(keys: torch.Tensor, vals: Optional[torch.Tensor],
     m.radix_scatter(keys.data_ptr(),
                     vals.data_ptr() if vals is not None else 0,
                     n, shift, nbits_eff, hist.data_ptr(),
-                    key_dst.data_ptr(), val_dst.data_ptr(), s)
+                    key_dst.data_ptr(), val_dst.data_ptr(), s, int(hash_mix))
     return totals[:1 << nbits], keys_out, vals_out
 
 

@@ -23,6 +23,12 @@ class HashPartitioner:
 
     def __init__(self, num_partitions: int):
         self.num_partitions = num_partitions
+        # GPU fast path: for pow2 R >= 16, digit = hash_mix64(k) & (R-1)
+        # in the kernel is bit-identical to partition_ids (% == & for pow2)
+        nbits = (num_partitions - 1).bit_length()
+        if (1 << nbits) == num_partitions and 4 <= nbits <= 12:
+            self.gpu_hash = True
+            self.gpu_shift = 0
 
     def partition_ids(self, keys: np.ndarray) -> np.ndarray:
         k = keys.astype(np.uint64, copy=False)

@@ -1,0 +1,141 @@
+"""PageRank over the one-sided shuffle — BASELINE config 4 (the
+reference's 19 GB PageRank, README.md:25-31).
+
+Spark-GraphX-style partitioning: edges live partitioned by src (each
+executor owns a contiguous vertex range and all edges whose src falls in
+it); ranks are co-partitioned, so the per-edge contribution
+rank[src]/outdeg[src] is computed locally and one shuffle per iteration
+routes (dst, contribution) to dst's owner, which aggregates into the new
+rank vector.
+
+Records are the framework's fixed-width pairs: key = dst vertex id (u64),
+value = contribution (f64 bit-pattern in the u64 payload).
+"""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass
+
+import numpy as np
+
+from ..engine import Engine
+from ..partitioner import RangePartitioner
+from ..writer import unpack_partition_segment
+
+DAMPING = 0.85
+
+
+@dataclass
+class PageRankResult:
+    seconds: float
+    iterations: int
+    edges: int
+    shuffle_bytes: int
+
+
+class PageRank:
+    def __init__(self, engine: Engine, num_vertices: int,
+                 edges_per_executor: int, partitions_per_executor: int = 64,
+                 device: str = "cpu", iterations: int = 5, seed: int = 0):
+        if num_vertices & (num_vertices - 1):
+            raise ValueError("num_vertices must be pow2")
+        self.engine = engine
+        self.V = num_vertices
+        self.iters = iterations
+        self.device = device
+        W = engine.world_size
+        R = W * partitions_per_executor
+        if R & (R - 1):
+            raise ValueError("total partitions must be pow2")
+        self.R = R
+        self.ppe = partitions_per_executor
+        # partitioner over the vertex id space (span = V, pow2)
+        self.part = RangePartitioner.uniform(R, key_min=0, key_max=self.V - 1)
+        rank = engine.rank
+        self.own_lo = rank * self.V // W
+        self.own_hi = (rank + 1) * self.V // W
+        span = self.own_hi - self.own_lo
+        rng = np.random.default_rng(seed * 7919 + rank)
+        src = rng.integers(self.own_lo, self.own_hi, edges_per_executor,
+                           dtype=np.uint64)
+        dst = rng.integers(0, self.V, edges_per_executor, dtype=np.uint64)
+        self.n_edges = edges_per_executor
+        outdeg = np.bincount((src - self.own_lo).astype(np.int64),
+                             minlength=span).astype(np.float64)
+        outdeg[outdeg == 0] = 1.0
+        if device == "cuda":
+            import torch
+            self.src_local = torch.from_numpy(
+                (src - self.own_lo).astype(np.int64)).cuda()
+            self.dst = torch.from_numpy(dst.view(np.int64)).cuda()
+            self.outdeg = torch.from_numpy(outdeg).cuda()
+            self.ranks = torch.full((span,), 1.0 / self.V,
+                                    dtype=torch.float64, device="cuda")
+        else:
+            self.src_local = (src - self.own_lo).astype(np.int64)
+            self.dst = dst
+            self.outdeg = outdeg
+            self.ranks = np.full(span, 1.0 / self.V, dtype=np.float64)
+        self._edges_u8 = None  # CPU value buffer cache
+
+    # ------------------------------------------------------------------
+
+    def run_step(self) -> PageRankResult:
+        t0 = time.perf_counter()
+        shuffle_bytes = 0
+        for _ in range(self.iters):
+            shuffle_bytes += self._iteration()
+        dt = time.perf_counter() - t0
+        return PageRankResult(dt, self.iters, self.n_edges, shuffle_bytes)
+
+    def _iteration(self) -> int:
+        eng = self.engine
+        handle = eng.register_shuffle(eng.world_size, self.R)
+        w = eng.manager.get_writer(handle, eng.rank)
+        if self.device == "cuda":
+            import torch
+            contrib = (self.ranks / self.outdeg)[self.src_local]
+            w.write_device_batch(self.dst, contrib.view(torch.int64))
+        else:
+            contrib = (self.ranks / self.outdeg)[self.src_local]
+            w.write_batch(self.dst,
+                          contrib.view(np.uint8).reshape(-1, 8).copy())
+        w.stop(True, partitioner=self.part)
+        eng.barrier()
+        lo, hi = eng.rank * self.ppe, (eng.rank + 1) * self.ppe - 1
+        reader = eng.manager.get_reader(handle, lo, hi)
+        span = self.own_hi - self.own_lo
+        if self.device == "cuda":
+            import torch
+            sums = torch.zeros(span, dtype=torch.float64, device="cuda")
+            for ref, data in reader:
+                t = data.view(torch.int64)
+                nrec = t.numel() // 2
+                idx = t[:nrec] - self.own_lo
+                sums.index_add_(0, idx, t[nrec:].contiguous().view(torch.float64))
+            self.ranks = (1.0 - DAMPING) / self.V + DAMPING * sums
+            torch.cuda.synchronize()
+        else:
+            sums = np.zeros(span, dtype=np.float64)
+            for ref, data in reader:
+                k, v = unpack_partition_segment(data, 8)
+                idx = (np.asarray(k) - self.own_lo).astype(np.int64)
+                np.add.at(sums, idx, np.ascontiguousarray(v).reshape(-1, 8)
+                          .view(np.float64).reshape(-1))
+            self.ranks = (1.0 - DAMPING) / self.V + DAMPING * sums
+        eng.unregister_shuffle(handle)
+        return reader.metrics.remote_bytes_read + reader.metrics.local_bytes_read
+
+    # reference implementation for validation (single process, full graph)
+    @staticmethod
+    def dense_reference(V, src, dst, iterations):
+        outdeg = np.bincount(src, minlength=V).astype(np.float64)
+        outdeg[outdeg == 0] = 1.0
+        ranks = np.full(V, 1.0 / V)
+        for _ in range(iterations):
+            contrib = ranks / outdeg
+            sums = np.zeros(V)
+            np.add.at(sums, dst, contrib[src])
+            ranks = (1.0 - DAMPING) / V + DAMPING * sums
+        return ranks

@@ -154,7 +154,10 @@ class ShuffleWriter:
             raise ValueError(
                 f"GPU partitioner requires pow2 partitions <= 4096, got {R}")
         shift = getattr(partitioner, "gpu_shift", 64 - nbits)
+        hash_mix = int(getattr(partitioner, "gpu_hash", False))
         nbits_eff = max(nbits, 4)  # kernel instantiations start at 4 bits
+        if hash_mix and nbits_eff != nbits:
+            raise ValueError("GPU hash partitioning requires >= 16 partitions")
         batches = self._gpu_batches
         keys = (batches[0][0] if len(batches) == 1
                 else torch.cat([b[0] for b in batches]))
@@ -171,7 +174,7 @@ class ShuffleWriter:
                               dtype=torch.int32, device=dev)
         totals = torch.empty(nd, dtype=torch.int32, device=dev)
         hs.radix_hist(keys.data_ptr(), n, shift, nbits_eff, hist.data_ptr(),
-                      stream)
+                      stream, hash_mix)
         hs.radix_scan(hist.data_ptr(), n, nbits_eff, totals.data_ptr(),
                       scan_ws.data_ptr(), stream)
         counts = totals.cpu().numpy().astype(np.int64)[:R]  # syncs the stream
@@ -220,7 +223,7 @@ class ShuffleWriter:
         hs.radix_scatter(keys.data_ptr(),
                          vals.data_ptr() if has_val else 0,
                          n, shift, nbits_eff, hist.data_ptr(),
-                         kd.data_ptr(), vd.data_ptr(), stream)
+                         kd.data_ptr(), vd.data_ptr(), stream, hash_mix)
         torch.cuda.synchronize()
         self.metrics.records_written += n
         mgr.keep_alive(self.handle, self.map_id, blocks)

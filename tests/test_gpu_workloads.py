@@ -1,0 +1,80 @@
+"""GPU workload paths: PageRank and sort-merge join end-to-end on one
+MI355X, validated against CPU oracles; plus the hash-mix kernel digit."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture
+def engine(tmp_path):
+    from sparkrdma_amd.conf import ShuffleConf
+    from sparkrdma_amd.engine import Engine
+    conf = ShuffleConf(shm_dir=str(tmp_path), transport="ipc",
+                       hbm_pool_size=2 << 30)
+    eng = Engine(conf, rank=0, world_size=1, driver_port=0)
+    yield eng
+    eng.shutdown()
+
+
+def test_pagerank_gpu_matches_dense(engine):
+    from sparkrdma_amd.workloads.pagerank import PageRank
+    V, E, iters = 1 << 12, 200_000, 3
+    pr = PageRank(engine, num_vertices=V, edges_per_executor=E,
+                  partitions_per_executor=16, device="cuda",
+                  iterations=iters, seed=5)
+    pr.run_step()
+    rng = np.random.default_rng(5 * 7919 + 0)
+    src = rng.integers(0, V, E, dtype=np.uint64).astype(np.int64)
+    dst = rng.integers(0, V, E, dtype=np.uint64).astype(np.int64)
+    want = PageRank.dense_reference(V, src, dst, iters)
+    got = pr.ranks.cpu().numpy()
+    np.testing.assert_allclose(got, want, rtol=1e-9)
+
+
+def test_sql_join_gpu_oracle(engine):
+    from sparkrdma_amd.workloads.sql_join import SortMergeJoin
+    j = SortMergeJoin(engine, rows_per_executor=500_000,
+                      partitions_per_executor=64, device="cuda",
+                      key_space_bits=18, validate=True)
+    r = j.run_step()
+    assert r.matches > 0
+
+
+def test_merge_join_kernel_oracle():
+    from sparkrdma_amd.ops.join import merge_join_sorted
+    rng = np.random.default_rng(1)
+    a = np.sort(rng.integers(0, 10_000, 100_000, dtype=np.uint64))
+    b = np.sort(rng.integers(0, 10_000, 80_000, dtype=np.uint64))
+    ak = torch.from_numpy(a.view(np.int64)).cuda()
+    bk = torch.from_numpy(b.view(np.int64)).cuda()
+    av = ak.clone()
+    bv = bk.clone()
+    jk, ja, jb = merge_join_sorted(ak, av, bk, bv)
+    torch.cuda.synchronize()
+    want = int((np.searchsorted(b, a, "right")
+                - np.searchsorted(b, a, "left")).sum())
+    assert jk.numel() == want
+    assert torch.equal(jk, ja) and torch.equal(jk, jb)
+
+
+def test_hash_mix_partition_matches_cpu():
+    from sparkrdma_amd.ops.radix import radix_partition
+    from sparkrdma_amd.partitioner import HashPartitioner
+    n, R = 300_000, 64
+    hp = HashPartitioner(R)
+    assert getattr(hp, "gpu_hash", False)
+    rng = np.random.default_rng(9)
+    k = rng.integers(0, 2 ** 63, n, dtype=np.uint64)
+    keys = torch.from_numpy(k.view(np.int64)).cuda()
+    counts, keys_out, _ = radix_partition(keys, None, 6, shift=0,
+                                          hash_mix=True)
+    torch.cuda.synchronize()
+    pids = hp.partition_ids(k)
+    want = np.bincount(pids, minlength=R)
+    assert np.array_equal(counts.cpu().numpy(), want)
+    # partition-ordered output matches CPU stable sort by hash pid
+    order = np.argsort(pids, kind="stable")
+    assert np.array_equal(keys_out.cpu().numpy().view(np.uint64), k[order])
