@@ -58,6 +58,7 @@ class Driver:
         self._server.listen(128)
         self.port = self._server.getsockname()[1]
         self._members: Dict[int, rpc.ExecutorInfo] = {}
+        self._conn_exec: Dict[rpc.MsgConnection, int] = {}
         self._conns: List[rpc.MsgConnection] = []
         self._shuffles: Dict[int, _ShuffleMeta] = {}
         self._next_shuffle_id = 0
@@ -95,17 +96,37 @@ class Driver:
         except (OSError, ValueError):
             pass
         finally:
+            # executor loss: prune membership and re-announce, the analog
+            # of SparkListenerBlockManagerRemoved pruning (reference
+            # RdmaShuffleManager.scala:155-165)
+            reannounce = None
             with self._lock:
                 if conn in self._conns:
                     self._conns.remove(conn)
                 if conn in self._barrier_waiters:
                     self._barrier_waiters.remove(conn)
+                exec_id = self._conn_exec.pop(conn, None)
+                if exec_id is not None and not self._stopped.is_set():
+                    self._members.pop(exec_id, None)
+                    log.warning("executor %d lost; %d members remain",
+                                exec_id, len(self._members))
+                    reannounce = (rpc.pack_announce(
+                        self.app_id, list(self._members.values())),
+                        list(self._conns))
+            if reannounce is not None:
+                payload, conns = reannounce
+                for c in conns:
+                    try:
+                        c.send(rpc.MSG_ANNOUNCE, payload)
+                    except OSError:
+                        pass
 
     def _dispatch(self, conn: rpc.MsgConnection, mtype: int, body: bytes) -> None:
         if mtype == rpc.MSG_HELLO:
             info = rpc.unpack_hello(body)
             with self._lock:
                 self._members[info.executor_id] = info
+                self._conn_exec[conn] = info.executor_id
                 members = list(self._members.values())
                 conns = list(self._conns)
             # fan out announce to every connected executor (reference :89-112)
