@@ -43,9 +43,12 @@ def main():
         import torch.distributed as tdist
         dist = tdist
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        dist.init_process_group("nccl" if use_cuda else "gloo")
+        backend = os.environ.get("BENCH_DIST_BACKEND",
+                                 "nccl" if use_cuda else "gloo")
+        dist.init_process_group(backend)
         if use_cuda:
-            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+            ngpu = torch.cuda.device_count()
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)) % ngpu)
 
     from sparkrdma_amd.conf import ShuffleConf
     from sparkrdma_amd.engine import Engine
@@ -54,9 +57,12 @@ def main():
     n_rec = int(args.gb_per_gpu * (1 << 30) / TeraSort.RECORD_BYTES)
     ppe = args.partitions_per_executor
     if ppe == 0:
-        ppe = 128  # world sizes are pow2 in the driver's sweep
+        # keep R = 256 total: 8-bit partition pass AND a 7-pass reduce sort
+        ppe = max(32, 256 // world)
     conf = ShuffleConf(transport="ipc" if use_cuda else "shm")
     if use_cuda:
+        conf.gpu_id = int(os.environ.get("LOCAL_RANK", rank)) % \
+            torch.cuda.device_count()
         # served blocks are pow2-rounded (buddy) -> up to 2x data, + slack
         conf.hbm_pool_size = int((args.gb_per_gpu * 2.5 + 2) * (1 << 30))
         conf.shuffle_write_block_size = 64 << 20
