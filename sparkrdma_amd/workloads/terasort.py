@@ -115,65 +115,59 @@ class TeraSort:
             sort_s=t_sort - t_fetch,
             remote_bytes=reader.metrics.remote_bytes_read)
 
-    def _reduce(self, parts: dict) -> dict:
-        """Sort each partition's fetched chunks over the low bits."""
-        out = {}
+    def _reduce(self, parts: dict):
+        """One batched sort per rank: all owned partitions share the top
+        log2(W) key bits (the rank prefix), so sorting the concatenated
+        fetches over bits [0, 64 - log2(W)) yields the rank's fully
+        sorted output in a single kernel sequence — avoiding per-partition
+        launch/workspace overhead (measured: R=64 beat R=256 at 8 GB
+        despite one extra radix pass)."""
+        wbits = (self.engine.world_size - 1).bit_length()
         if self.device == "cuda":
             import torch
             from ..ops.radix import sort_pairs
+            ks, vs = [], []
             for p, chunks in parts.items():
-                if not chunks:
-                    out[p] = (None, None)
-                    continue
-                ks, vs = [], []
                 for c in chunks:
                     t = c.view(torch.int64)
                     nrec = t.numel() // 2
                     ks.append(t[:nrec])
                     vs.append(t[nrec:])
-                k = torch.cat(ks) if len(ks) > 1 else ks[0].contiguous()
-                v = torch.cat(vs) if len(vs) > 1 else vs[0].contiguous()
-                out[p] = sort_pairs(k, v, 0, self.low_bits)
+            if not ks:
+                return None, None
+            k = torch.cat(ks) if len(ks) > 1 else ks[0].contiguous()
+            v = torch.cat(vs) if len(vs) > 1 else vs[0].contiguous()
+            out = sort_pairs(k, v, 0, 64 - wbits)
             torch.cuda.synchronize()
-        else:
-            for p, chunks in parts.items():
-                ks, vs = [], []
-                for c in chunks:
-                    k, v = unpack_partition_segment(c, 8)
-                    ks.append(np.array(k))
-                    vs.append(np.array(v))
-                if not ks:
-                    out[p] = (None, None)
-                    continue
-                k = np.concatenate(ks)
-                v = np.concatenate(vs)
-                order = np.argsort(k, kind="stable")
-                out[p] = (k[order], v[order])
-        return out
+            return out
+        ks, vs = [], []
+        for p, chunks in parts.items():
+            for c in chunks:
+                k, v = unpack_partition_segment(c, 8)
+                ks.append(np.array(k))
+                vs.append(np.array(v))
+        if not ks:
+            return None, None
+        k = np.concatenate(ks)
+        v = np.concatenate(vs)
+        order = np.argsort(k, kind="stable")
+        return k[order], v[order]
 
-    def _validate(self, sorted_parts: dict, lo: int) -> None:
-        prev_max = None
-        total = 0
-        for p in sorted(sorted_parts):
-            k, v = sorted_parts[p]
-            if k is None:
-                continue
-            if self.device == "cuda":
-                import torch
-                ku = k.cpu().numpy().view(np.uint64)
-                vu = v.cpu().numpy().view(np.uint64)
-            else:
-                ku, vu = k, v.view(np.uint64).reshape(-1)
-            assert np.all(ku[1:] >= ku[:-1]), f"partition {p} not sorted"
-            assert np.array_equal(ku, vu.reshape(-1)), "payload corrupted"
-            pids = self.part.partition_ids(ku)
-            assert np.all(pids == p), f"foreign keys in partition {p}"
-            if prev_max is not None and len(ku):
-                assert ku[0] >= prev_max
-            if len(ku):
-                prev_max = ku[-1]
-            total += len(ku)
-        log.info("validated %d records in partitions >= %d", total, lo)
+    def _validate(self, sorted_out, lo: int) -> None:
+        k, v = sorted_out
+        if k is None:
+            return
+        if self.device == "cuda":
+            ku = k.cpu().numpy().view(np.uint64)
+            vu = v.cpu().numpy().view(np.uint64)
+        else:
+            ku, vu = k, v.view(np.uint64).reshape(-1)
+        assert np.all(ku[1:] >= ku[:-1]), "rank output not sorted"
+        assert np.array_equal(ku, vu.reshape(-1)), "payload corrupted"
+        pids = self.part.partition_ids(ku)
+        hi = lo + self.ppe - 1
+        assert np.all((pids >= lo) & (pids <= hi)), "foreign keys in range"
+        log.info("validated %d records in partitions [%d, %d]", len(ku), lo, hi)
 
     # ------------------------------------------------------------------
 
