@@ -65,7 +65,11 @@ def main():
             torch.cuda.device_count()
         # served blocks are pow2-rounded (buddy) -> up to 2x data, + slack
         conf.hbm_pool_size = int((args.gb_per_gpu * 2.5 + 2) * (1 << 30))
-        conf.shuffle_write_block_size = 64 << 20
+        # few, large contiguous blocks: adjacent partitions of one map pack
+        # into one 512 MiB block, so coalesced fetches become ~GB-scale
+        # xGMI copies instead of per-partition ones
+        conf.shuffle_write_block_size = 512 << 20
+        conf.shuffle_read_block_size = 512 << 20
         conf.max_bytes_in_flight = 8 << 30
     eng = Engine(conf, rank=rank, world_size=world)
 
@@ -82,8 +86,13 @@ def main():
         if use_cuda:
             torch.cuda.synchronize()
 
+    debug = os.environ.get("BENCH_DEBUG")
     for _ in range(args.warmup):
-        ts.run_step()
+        r = ts.run_step()
+        if debug and rank == 0:
+            print(f"[warmup] total={r.seconds*1e3:.1f}ms write={r.write_s*1e3:.1f} "
+                  f"fetch={r.fetch_s*1e3:.1f} sort={r.sort_s*1e3:.1f}",
+                  file=sys.stderr)
     barrier_sync()
     t0 = time.perf_counter()
     results = [ts.run_step() for _ in range(args.steps)]
