@@ -56,13 +56,14 @@ PYBIND11_MODULE(_hipshuffle, m) {
   m.def("radix_scatter", &hs::radix_scatter, py::arg("keys"), py::arg("vals"),
         py::arg("n"), py::arg("shift"), py::arg("nbits"), py::arg("hist"),
         py::arg("key_dst"), py::arg("val_dst"), py::arg("stream") = 0,
-        py::arg("hash_mix") = 0);
+        py::arg("hash_mix") = 0, py::arg("aos_out") = 0);
   m.def("sort_workspace_bytes", &hs::sort_workspace_bytes);
   m.def("sort_pairs_u64", &hs::sort_pairs_u64, py::arg("keys"),
         py::arg("vals"), py::arg("tmp_keys"), py::arg("tmp_vals"),
         py::arg("n"), py::arg("start_bit"), py::arg("end_bit"), py::arg("ws"),
         py::arg("stream") = 0,
         py::call_guard<py::gil_scoped_release>());
+  m.def("probe_scatter_write", &hs::probe_scatter_write);
   m.def("join_count", &hs::join_count, py::arg("a_keys"), py::arg("na"),
         py::arg("b_keys"), py::arg("nb"), py::arg("counts"),
         py::arg("lo_idx"), py::arg("stream") = 0);
@@ -71,6 +72,10 @@ PYBIND11_MODULE(_hipshuffle, m) {
         py::arg("lo_idx"), py::arg("offsets"), py::arg("out_key"),
         py::arg("out_a"), py::arg("out_b"), py::arg("stream") = 0);
   m.def("onesweep_workspace_bytes", &hs::onesweep_workspace_bytes);
+  m.def("onesweep_sort_aos_u64", &hs::onesweep_sort_aos_u64,
+        py::arg("pairs"), py::arg("tmp_pairs"), py::arg("n"),
+        py::arg("start_bit"), py::arg("end_bit"), py::arg("ws"),
+        py::arg("stream") = 0, py::call_guard<py::gil_scoped_release>());
   m.def("onesweep_sort_pairs_u64", &hs::onesweep_sort_pairs_u64,
         py::arg("keys"), py::arg("vals"), py::arg("tmp_keys"),
         py::arg("tmp_vals"), py::arg("n"), py::arg("start_bit"),

@@ -33,11 +33,16 @@ void radix_scan(uintptr_t hist, uint32_t n, int nbits, uintptr_t totals,
                 uintptr_t scan_ws, uintptr_t stream);
 void radix_scatter(uintptr_t keys, uintptr_t vals, uint32_t n, int shift,
                    int nbits, uintptr_t hist, uintptr_t key_dst,
-                   uintptr_t val_dst, uintptr_t stream, int hash_mix = 0);
+                   uintptr_t val_dst, uintptr_t stream, int hash_mix = 0,
+                   int aos_out = 0);
 size_t sort_workspace_bytes(uint32_t n);
 int sort_pairs_u64(uintptr_t keys, uintptr_t vals, uintptr_t tmp_keys,
                    uintptr_t tmp_vals, uint32_t n, int start_bit, int end_bit,
                    uintptr_t ws, uintptr_t stream);
+void probe_scatter_write(uintptr_t out, uint64_t region_elems,
+                         uint32_t nregions, uint32_t burst_elems,
+                         uint32_t bursts_per_block, uint32_t grid,
+                         uintptr_t stream);
 void join_count(uintptr_t a_keys, uint32_t na, uintptr_t b_keys,
                 uint32_t nb_, uintptr_t counts, uintptr_t lo_idx,
                 uintptr_t stream);
@@ -46,6 +51,9 @@ void join_emit(uintptr_t a_keys, uintptr_t a_vals, uint32_t na,
                uintptr_t offsets, uintptr_t out_key, uintptr_t out_a,
                uintptr_t out_b, uintptr_t stream);
 size_t onesweep_workspace_bytes(uint32_t n, int passes);
+int onesweep_sort_aos_u64(uintptr_t pairs, uintptr_t tmp_pairs, uint32_t n,
+                          int start_bit, int end_bit, uintptr_t ws,
+                          uintptr_t stream);
 int onesweep_sort_pairs_u64(uintptr_t keys, uintptr_t vals,
                             uintptr_t tmp_keys, uintptr_t tmp_vals,
                             uint32_t n, int start_bit, int end_bit,

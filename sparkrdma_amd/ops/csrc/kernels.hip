@@ -236,7 +236,7 @@ __global__ __launch_bounds__(BLOCK) void radix_scatter_kernel(
     const uint64_t* __restrict__ keys, const uint64_t* __restrict__ vals,
     uint32_t n, int shift, const uint32_t* __restrict__ hist, uint32_t nb,
     const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst,
-    int hash_mix) {
+    int hash_mix, int aos_out) {
   constexpr int ND = 1 << NBITS;
   extern __shared__ char smem_raw[];
   // layout: exchange u64[TILE] | counters u32[NW][ND] | start u32[ND]
@@ -324,7 +324,7 @@ __global__ __launch_bounds__(BLOCK) void radix_scatter_kernel(
     if (j < tile_n) {
       uint64_t k = exch[j];
       uint32_t d = (uint32_t)((hash_mix ? hash_mix64(k) : k) >> shift) & (ND - 1);
-      uint32_t off = pref[d] + (j - start[d]);
+      uint32_t off = (pref[d] + (j - start[d])) << aos_out;
       out_d[i] = d;
       out_off[i] = off;
       reinterpret_cast<uint64_t*>(key_dst[d])[off] = k;
@@ -351,20 +351,19 @@ __global__ __launch_bounds__(BLOCK) void radix_scatter_kernel(
 // Onesweep path (8-bit digits): one kernel per pass with decoupled
 // lookback — removes the per-pass hist read and the scan kernels.
 //
-// Status word per (block, digit): 2-bit flag | 30-bit count/prefix packed
-// in ONE u32, so a single relaxed agent-scope atomic is a complete
+// Status word per (block, digit): 2-bit flag | 62-bit count/prefix packed
+// in ONE u64, so a single relaxed agent-scope atomic is a complete
 // publish (no multi-location ordering; the guide's G16 concerns collapse
-// to single-word atomicity). Agent-scope atomics are cache-coherent
-// across XCDs. Blocks take an atomic ticket at start, so a block's
-// predecessors in lookback order are already launched -> no deadlock,
-// and the ticket determines the tile, so results are placement- and
-// timing-independent. Requires n < 2^30 (reduce-side sorts are
-// per-partition, well under; larger calls fall back to the 3-kernel path).
+// to single-word atomicity; an 8-byte relaxed agent store lowers to an
+// sc1 write-through). Agent-scope atomics are cache-coherent across
+// XCDs. Blocks take an atomic ticket at start, so a block's predecessors
+// in lookback order are already launched -> no deadlock, and the ticket
+// determines the tile, so results are placement- and timing-independent.
 
-constexpr uint32_t FLAG_AGG = 1u << 30;
-constexpr uint32_t FLAG_INC = 2u << 30;
-constexpr uint32_t FLAG_MASK = 3u << 30;
-constexpr uint32_t VAL_MASK = (1u << 30) - 1;
+constexpr uint64_t FLAG_AGG = 1ull << 62;
+constexpr uint64_t FLAG_INC = 2ull << 62;
+constexpr uint64_t FLAG_MASK = 3ull << 62;
+constexpr uint64_t VAL_MASK = (1ull << 62) - 1;
 
 // Global digit totals of EVERY pass in one read (digit counts are
 // order-independent, so pass k's totals can be computed from pass 0's
@@ -372,7 +371,7 @@ constexpr uint32_t VAL_MASK = (1u << 30) - 1;
 template <int MAX_PASSES>
 __global__ __launch_bounds__(BLOCK) void onesweep_hist_all_kernel(
     const uint64_t* __restrict__ keys, uint32_t n, int start_bit, int passes,
-    uint32_t* __restrict__ totals /* [passes][256] */) {
+    uint32_t* __restrict__ totals /* [passes][256] */, int in_stride) {
   __shared__ uint32_t cnt[NW][MAX_PASSES * 256];
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -381,7 +380,7 @@ __global__ __launch_bounds__(BLOCK) void onesweep_hist_all_kernel(
   __syncthreads();
   const uint64_t stride = (uint64_t)gridDim.x * BLOCK;
   for (uint64_t e = (uint64_t)blockIdx.x * BLOCK + tid; e < n; e += stride) {
-    uint64_t k = keys[e];
+    uint64_t k = keys[e * in_stride];
     for (int p = 0; p < passes; ++p) {
       uint32_t d = (uint32_t)(k >> (start_bit + p * 8)) & 255;
       atomicAdd(&cnt[wave][p * 256 + d], 1);
@@ -396,22 +395,29 @@ __global__ __launch_bounds__(BLOCK) void onesweep_hist_all_kernel(
   }
 }
 
-template <bool HAS_VAL, int IT>
+// AOS=true: elements are interleaved (key, val) 16-byte records — one
+// dwordx4 load per element, a pair LDS exchange, ONE 16-byte store per
+// element, and 2x-longer digit write bursts (measured: scattered-write
+// bandwidth doubles from 128 B to 256 B bursts — profiles/).
+template <bool HAS_VAL, int IT, bool AOS>
 __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
     const uint64_t* __restrict__ keys, const uint64_t* __restrict__ vals,
-    uint32_t n, int shift, uint32_t* __restrict__ desc /* [nb][256] */,
+    uint32_t n, int shift, uint64_t* __restrict__ desc /* [nb][256] */,
     uint32_t* __restrict__ ticket,
     const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst) {
   constexpr int ND = 256;
   extern __shared__ char smem_raw[];
   constexpr int TILE_T = BLOCK * IT;
-  uint64_t* exch = reinterpret_cast<uint64_t*>(smem_raw);
-  uint32_t* counters = reinterpret_cast<uint32_t*>(exch + TILE_T);
+  using u64x2 = __attribute__((ext_vector_type(2))) unsigned long long;
+  uint64_t* exch = reinterpret_cast<uint64_t*>(smem_raw);  // [TILE_T] or
+  u64x2* exch2 = reinterpret_cast<u64x2*>(smem_raw);       // [TILE_T] pairs
+  uint32_t* counters = reinterpret_cast<uint32_t*>(
+      exch + (AOS ? 2 * TILE_T : TILE_T));
   uint32_t* start = counters + NW * ND;
   uint32_t* pref = start + ND;
   uint32_t* sums = pref + ND;
   uint32_t* vb_sh = sums + BLOCK;
-  uint8_t* dsort = reinterpret_cast<uint8_t*>(vb_sh + 4);  // [TILE_T]
+  uint8_t* dsort = reinterpret_cast<uint8_t*>(vb_sh + 4);  // [TILE_T], !AOS
 
   const int tid = threadIdx.x;
   const int lane = tid & (kWave - 1);
@@ -427,13 +433,22 @@ __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
 
   // phase A: stable per-wave ranks; digit(8b)|rank(16b) packed per elem
   uint64_t key_reg[IT];
+  uint64_t val_reg[AOS ? IT : 1];
   uint32_t digrank[IT];
   const uint64_t chunk = tile_start + (uint64_t)wave * (IT * kWave);
 #pragma unroll
   for (int i = 0; i < IT; ++i) {
     uint64_t e = chunk + (uint64_t)i * kWave + lane;
     bool valid = e < n;
-    uint64_t k = valid ? keys[e] : 0;
+    uint64_t k;
+    if (AOS) {
+      u64x2 kv = valid ? reinterpret_cast<const u64x2*>(keys)[e]
+                       : u64x2{0, 0};
+      k = kv.x;
+      val_reg[i] = kv.y;
+    } else {
+      k = valid ? keys[e] : 0;
+    }
     key_reg[i] = k;
     uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
     uint64_t vm = __ballot(valid);
@@ -465,8 +480,8 @@ __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
 
   // publish aggregate (or inclusive for block 0), then lookback
   if (tid < ND) {
-    uint32_t total = start[tid];
-    uint32_t* slot = &desc[(uint64_t)b * ND + tid];
+    uint64_t total = start[tid];
+    uint64_t* slot = &desc[(uint64_t)b * ND + tid];
     if (b == 0) {
       __hip_atomic_store(slot, FLAG_INC | total, __ATOMIC_RELAXED,
                          __HIP_MEMORY_SCOPE_AGENT);
@@ -474,9 +489,9 @@ __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
     } else {
       __hip_atomic_store(slot, FLAG_AGG | total, __ATOMIC_RELAXED,
                          __HIP_MEMORY_SCOPE_AGENT);
-      uint32_t run = 0;
+      uint64_t run = 0;
       for (int64_t j = (int64_t)b - 1; j >= 0;) {
-        uint32_t w = __hip_atomic_load(&desc[(uint64_t)j * ND + tid],
+        uint64_t w = __hip_atomic_load(&desc[(uint64_t)j * ND + tid],
                                        __ATOMIC_RELAXED,
                                        __HIP_MEMORY_SCOPE_AGENT);
         if ((w & FLAG_MASK) == 0) {
@@ -489,14 +504,38 @@ __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
       }
       __hip_atomic_store(slot, FLAG_INC | (run + total), __ATOMIC_RELAXED,
                          __HIP_MEMORY_SCOPE_AGENT);
-      pref[tid] = run;
+      pref[tid] = (uint32_t)run;
     }
   }
   // block-local digit starts (exclusive scan of totals)
   block_exscan(start, sums, ND);  // includes the needed __syncthreads
 
-  // LDS exchange + linear write-out; digit per sorted slot kept in a
-  // byte array so the val phase re-derives (d, off) without registers
+  if (AOS) {
+    // pair exchange + one dwordx4 store per element
+#pragma unroll
+    for (int i = 0; i < IT; ++i) {
+      uint64_t e = chunk + (uint64_t)i * kWave + lane;
+      if (e < n) {
+        uint32_t d = digrank[i] >> 16;
+        uint32_t j = start[d] + my[d] + (digrank[i] & 0xFFFF);
+        exch2[j] = u64x2{key_reg[i], val_reg[i]};
+      }
+    }
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < IT; ++i) {
+      uint32_t j = i * BLOCK + tid;
+      if (j < tile_n) {
+        u64x2 kv = exch2[j];
+        uint32_t d = (uint32_t)((uint64_t)kv.x >> shift) & (ND - 1);
+        uint32_t off = pref[d] + (j - start[d]);
+        reinterpret_cast<u64x2*>(key_dst[d])[off] = kv;
+      }
+    }
+    return;
+  }
+
+  // SoA path: key exchange + write-out, then val exchange + write-out
 #pragma unroll
   for (int i = 0; i < IT; ++i) {
     uint64_t e = chunk + (uint64_t)i * kWave + lane;
@@ -539,11 +578,12 @@ __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
   }
 }
 
-// digit bases for pass p from the all-pass totals
+// digit bases for pass p from the all-pass totals; rec_bytes = 8 (SoA)
+// or 16 (AoS interleaved records)
 __global__ void onesweep_digit_bases_kernel(
     const uint32_t* __restrict__ totals /* [pass][256] */, int pass,
     uint64_t out_keys, uint64_t out_vals, uint64_t* __restrict__ key_dst,
-    uint64_t* __restrict__ val_dst) {
+    uint64_t* __restrict__ val_dst, int rec_bytes) {
   __shared__ uint32_t arr[256];
   __shared__ uint32_t sums[BLOCK];
   const int tid = threadIdx.x;
@@ -551,9 +591,44 @@ __global__ void onesweep_digit_bases_kernel(
   __syncthreads();
   block_exscan(arr, sums, 256);
   if (tid < 256) {
-    key_dst[tid] = out_keys + (uint64_t)arr[tid] * 8;
-    val_dst[tid] = out_vals ? out_vals + (uint64_t)arr[tid] * 8 : 0;
+    key_dst[tid] = out_keys + (uint64_t)arr[tid] * rec_bytes;
+    val_dst[tid] = out_vals ? out_vals + (uint64_t)arr[tid] * rec_bytes : 0;
   }
+}
+
+// ---------------------------------------------------------------------------
+// Microbench probe: emulate the scatter's write pattern — each block
+// writes its tile as `nregions` bursts of `burst` bytes at region-strided
+// scattered offsets. Measures HBM efficiency vs burst length to size the
+// optimization headroom (not part of the data path).
+
+__global__ __launch_bounds__(BLOCK) void probe_scatter_write_kernel(
+    uint64_t* __restrict__ out, uint64_t region_elems, uint32_t nregions,
+    uint32_t burst_elems, uint32_t bursts_per_block) {
+  // block b writes bursts_per_block bursts; burst i goes to region
+  // (b*7 + i) % nregions at offset (b * bursts_per_block + i) scaled —
+  // scattered across regions like the radix scatter's digit streams.
+  const uint32_t b = blockIdx.x;
+  for (uint32_t i = 0; i < bursts_per_block; ++i) {
+    uint32_t region = (b * 7u + i) % nregions;
+    uint64_t off_in_region =
+        ((uint64_t)(b * 1315423911u + i * 2654435761u)) %
+        (region_elems - burst_elems);
+    uint64_t* dst = out + (uint64_t)region * region_elems + off_in_region;
+    for (uint32_t e = threadIdx.x; e < burst_elems; e += BLOCK)
+      dst[e] = (uint64_t)b * i + e;
+  }
+}
+
+void probe_scatter_write(uintptr_t out, uint64_t region_elems,
+                         uint32_t nregions, uint32_t burst_elems,
+                         uint32_t bursts_per_block, uint32_t grid,
+                         uintptr_t stream) {
+  hipLaunchKernelGGL(probe_scatter_write_kernel, dim3(grid), dim3(BLOCK), 0,
+                     reinterpret_cast<hipStream_t>(stream),
+                     reinterpret_cast<uint64_t*>(out), region_elems, nregions,
+                     burst_elems, bursts_per_block);
+  HIP_CHECK(hipGetLastError());
 }
 
 // ---------------------------------------------------------------------------
@@ -672,7 +747,7 @@ template <int NBITS, bool HAS_VAL>
 static void scatter_launch(const uint64_t* keys, const uint64_t* vals,
                            uint32_t n, int shift, const uint32_t* hist,
                            const uint64_t* key_dst, const uint64_t* val_dst,
-                           hipStream_t s, int hash_mix = 0) {
+                           hipStream_t s, int hash_mix = 0, int aos_out = 0) {
   constexpr int ND = 1 << NBITS;
   uint32_t nb = num_tiles(n);
   size_t lds = (size_t)TILE * 8 + (size_t)NW * ND * 4 + (size_t)ND * 4 * 2 +
@@ -686,7 +761,7 @@ static void scatter_launch(const uint64_t* keys, const uint64_t* vals,
     attr_set[NBITS] = true;
   }
   hipLaunchKernelGGL(kfn, dim3(nb), dim3(BLOCK), lds, s, keys, vals, n, shift,
-                     hist, nb, key_dst, val_dst, hash_mix);
+                     hist, nb, key_dst, val_dst, hash_mix, aos_out);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -747,24 +822,27 @@ void radix_scan(uintptr_t hist, uint32_t n, int nbits, uintptr_t totals,
 template <int NBITS>
 static void scatter_hv(const uint64_t* keys, const uint64_t* vals, uint32_t n,
                        int shift, const uint32_t* hist, const uint64_t* kd,
-                       const uint64_t* vd, hipStream_t s, int hash_mix = 0) {
+                       const uint64_t* vd, hipStream_t s, int hash_mix = 0,
+                       int aos_out = 0) {
   if (vals)
     scatter_launch<NBITS, true>(keys, vals, n, shift, hist, kd, vd, s,
-                                hash_mix);
+                                hash_mix, aos_out);
   else
     scatter_launch<NBITS, false>(keys, nullptr, n, shift, hist, kd, nullptr,
-                                 s, hash_mix);
+                                 s, hash_mix, aos_out);
 }
 
 void radix_scatter(uintptr_t keys, uintptr_t vals, uint32_t n, int shift,
                    int nbits, uintptr_t hist, uintptr_t key_dst,
-                   uintptr_t val_dst, uintptr_t stream, int hash_mix) {
+                   uintptr_t val_dst, uintptr_t stream, int hash_mix,
+                   int aos_out) {
   auto s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_NBITS(nbits, scatter_hv, reinterpret_cast<const uint64_t*>(keys),
                  reinterpret_cast<const uint64_t*>(vals), n, shift,
                  reinterpret_cast<const uint32_t*>(hist),
                  reinterpret_cast<const uint64_t*>(key_dst),
-                 reinterpret_cast<const uint64_t*>(val_dst), s, hash_mix);
+                 reinterpret_cast<const uint64_t*>(val_dst), s, hash_mix,
+                 aos_out);
 }
 
 // Full LSD sort of (keys[, vals]) by bits [start_bit, end_bit).
@@ -788,40 +866,46 @@ static inline uint32_t os_num_tiles(uint32_t n) {
 size_t onesweep_workspace_bytes(uint32_t n, int passes) {
   uint32_t nb = os_num_tiles(n);
   return (size_t)passes * 256 * 4 + 256 * 8 * 2 + 16 +
-         (size_t)nb * 256 * 4;
+         (size_t)nb * 256 * 8;
 }
 
-int onesweep_sort_pairs_u64(uintptr_t keys, uintptr_t vals,
-                            uintptr_t tmp_keys, uintptr_t tmp_vals,
-                            uint32_t n, int start_bit, int end_bit,
-                            uintptr_t ws, uintptr_t stream) {
-  if (n >= (1u << 30))
-    throw std::runtime_error("onesweep requires n < 2^30");
-  auto s = reinterpret_cast<hipStream_t>(stream);
+// aos = 0: keys/vals are separate u64 arrays (SoA).
+// aos = 1: keys/tmp_keys point at interleaved (key,val) 16-byte records;
+//          vals/tmp_vals ignored.
+static int onesweep_sort_impl(uintptr_t keys, uintptr_t vals,
+                              uintptr_t tmp_keys, uintptr_t tmp_vals,
+                              uint32_t n, int start_bit, int end_bit,
+                              uintptr_t ws, hipStream_t s, int aos) {
   uint32_t nb = os_num_tiles(n);
   int passes = (end_bit - start_bit + 7) / 8;
   uint32_t* totals = reinterpret_cast<uint32_t*>(ws);
   uint64_t* key_dst = reinterpret_cast<uint64_t*>(totals + (size_t)passes * 256);
   uint64_t* val_dst = key_dst + 256;
   uint32_t* ticket = reinterpret_cast<uint32_t*>(val_dst + 256);
-  uint32_t* desc = ticket + 4;
+  uint64_t* desc = reinterpret_cast<uint64_t*>(ticket + 4);
   HIP_CHECK(hipMemsetAsync(totals, 0, (size_t)passes * 256 * 4, s));
   uint32_t hist_grid = nb < 1024 ? (nb ? nb : 1) : 1024;
   hipLaunchKernelGGL(onesweep_hist_all_kernel<8>, dim3(hist_grid),
                      dim3(BLOCK), 0, s,
                      reinterpret_cast<const uint64_t*>(keys), n, start_bit,
-                     passes, totals);
+                     passes, totals, aos ? 2 : 1);
   HIP_CHECK(hipGetLastError());
-  size_t lds = (size_t)OS_TILE * 8 + (size_t)NW * 256 * 4 + 256 * 4 * 2 +
-               BLOCK * 4 + 16 + OS_TILE;
+  size_t lds_soa = (size_t)OS_TILE * 8 + (size_t)NW * 256 * 4 + 256 * 4 * 2 +
+                   BLOCK * 4 + 16 + OS_TILE;
+  size_t lds_aos = (size_t)OS_TILE * 16 + (size_t)NW * 256 * 4 +
+                   256 * 4 * 2 + BLOCK * 4 + 16;
+  size_t lds = aos ? lds_aos : lds_soa;
   static bool attr_set = false;
-  if (!attr_set && lds > 64 * 1024) {
-    (void)hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&onesweep_pass_kernel<true, OS_ITEMS>),
-        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
-    (void)hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&onesweep_pass_kernel<false, OS_ITEMS>),
-        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+  if (!attr_set) {
+    for (const void* f :
+         {reinterpret_cast<const void*>(
+              &onesweep_pass_kernel<true, OS_ITEMS, false>),
+          reinterpret_cast<const void*>(
+              &onesweep_pass_kernel<false, OS_ITEMS, false>),
+          reinterpret_cast<const void*>(
+              &onesweep_pass_kernel<true, OS_ITEMS, true>)})
+      (void)hipFuncSetAttribute(f, hipFuncAttributeMaxDynamicSharedMemorySize,
+                                (int)(lds_aos > lds_soa ? lds_aos : lds_soa));
     attr_set = true;
   }
   uintptr_t src_k = keys, src_v = vals, dst_k = tmp_keys, dst_v = tmp_vals;
@@ -829,20 +913,26 @@ int onesweep_sort_pairs_u64(uintptr_t keys, uintptr_t vals,
   for (int p = 0; p < passes; ++p) {
     hipLaunchKernelGGL(onesweep_digit_bases_kernel, dim3(1), dim3(BLOCK), 0,
                        s, totals, p, (uint64_t)dst_k, (uint64_t)dst_v,
-                       key_dst, val_dst);
+                       key_dst, val_dst, aos ? 16 : 8);
     HIP_CHECK(hipGetLastError());
     HIP_CHECK(hipMemsetAsync(ticket, 0, 16, s));
-    HIP_CHECK(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 4, s));
-    if (vals) {
-      hipLaunchKernelGGL((onesweep_pass_kernel<true, OS_ITEMS>), dim3(nb), dim3(BLOCK),
-                         lds, s, reinterpret_cast<const uint64_t*>(src_k),
+    HIP_CHECK(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
+    if (aos) {
+      hipLaunchKernelGGL((onesweep_pass_kernel<true, OS_ITEMS, true>),
+                         dim3(nb), dim3(BLOCK), lds, s,
+                         reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
+                         start_bit + p * 8, desc, ticket, key_dst, val_dst);
+    } else if (vals) {
+      hipLaunchKernelGGL((onesweep_pass_kernel<true, OS_ITEMS, false>),
+                         dim3(nb), dim3(BLOCK), lds, s,
+                         reinterpret_cast<const uint64_t*>(src_k),
                          reinterpret_cast<const uint64_t*>(src_v), n,
                          start_bit + p * 8, desc, ticket, key_dst, val_dst);
     } else {
-      hipLaunchKernelGGL((onesweep_pass_kernel<false, OS_ITEMS>), dim3(nb), dim3(BLOCK),
-                         lds, s, reinterpret_cast<const uint64_t*>(src_k),
-                         nullptr, n, start_bit + p * 8, desc, ticket, key_dst,
-                         val_dst);
+      hipLaunchKernelGGL((onesweep_pass_kernel<false, OS_ITEMS, false>),
+                         dim3(nb), dim3(BLOCK), lds, s,
+                         reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
+                         start_bit + p * 8, desc, ticket, key_dst, val_dst);
     }
     HIP_CHECK(hipGetLastError());
     std::swap(src_k, dst_k);
@@ -850,6 +940,22 @@ int onesweep_sort_pairs_u64(uintptr_t keys, uintptr_t vals,
     cur ^= 1;
   }
   return cur;
+}
+
+int onesweep_sort_pairs_u64(uintptr_t keys, uintptr_t vals,
+                            uintptr_t tmp_keys, uintptr_t tmp_vals,
+                            uint32_t n, int start_bit, int end_bit,
+                            uintptr_t ws, uintptr_t stream) {
+  return onesweep_sort_impl(keys, vals, tmp_keys, tmp_vals, n, start_bit,
+                            end_bit, ws,
+                            reinterpret_cast<hipStream_t>(stream), 0);
+}
+
+int onesweep_sort_aos_u64(uintptr_t pairs, uintptr_t tmp_pairs, uint32_t n,
+                          int start_bit, int end_bit, uintptr_t ws,
+                          uintptr_t stream) {
+  return onesweep_sort_impl(pairs, 0, tmp_pairs, 0, n, start_bit, end_bit,
+                            ws, reinterpret_cast<hipStream_t>(stream), 1);
 }
 
 int sort_pairs_u64(uintptr_t keys, uintptr_t vals, uintptr_t tmp_keys,

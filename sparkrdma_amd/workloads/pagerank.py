@@ -110,10 +110,9 @@ class PageRank:
             import torch
             sums = torch.zeros(span, dtype=torch.float64, device="cuda")
             for ref, data in reader:
-                t = data.view(torch.int64)
-                nrec = t.numel() // 2
-                idx = t[:nrec] - self.own_lo
-                sums.index_add_(0, idx, t[nrec:].contiguous().view(torch.float64))
+                t = data.view(torch.int64)  # AoS (dst, contrib) records
+                idx = t[0::2] - self.own_lo
+                sums.index_add_(0, idx, t[1::2].contiguous().view(torch.float64))
             self.ranks = (1.0 - DAMPING) / self.V + DAMPING * sums
             torch.cuda.synchronize()
         else:

@@ -131,16 +131,11 @@ class SortMergeJoin:
     @staticmethod
     def _concat_gpu(chunks):
         import torch
-        ks, vs = [], []
-        for c in chunks:
-            t = c.view(torch.int64)
-            nrec = t.numel() // 2
-            ks.append(t[:nrec])
-            vs.append(t[nrec:])
-        if not ks:
+        ts = [c.view(torch.int64) for c in chunks]  # AoS records
+        if not ts:
             return None, None
-        return (torch.cat(ks) if len(ks) > 1 else ks[0].contiguous(),
-                torch.cat(vs) if len(vs) > 1 else vs[0].contiguous())
+        pairs = torch.cat(ts) if len(ts) > 1 else ts[0].contiguous()
+        return pairs[0::2].contiguous(), pairs[1::2].contiguous()
 
     @staticmethod
     def _concat_cpu(chunks):

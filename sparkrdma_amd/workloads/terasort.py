@@ -125,21 +125,16 @@ class TeraSort:
         wbits = (self.engine.world_size - 1).bit_length()
         if self.device == "cuda":
             import torch
-            from ..ops.radix import sort_pairs
-            ks, vs = [], []
-            for p, chunks in parts.items():
-                for c in chunks:
-                    t = c.view(torch.int64)
-                    nrec = t.numel() // 2
-                    ks.append(t[:nrec])
-                    vs.append(t[nrec:])
-            if not ks:
+            from ..ops.radix import sort_pairs_aos
+            # fetched chunks are AoS (key,val) records: concat directly
+            ts = [c.view(torch.int64) for chunks in parts.values()
+                  for c in chunks]
+            if not ts:
                 return None, None
-            k = torch.cat(ks) if len(ks) > 1 else ks[0].contiguous()
-            v = torch.cat(vs) if len(vs) > 1 else vs[0].contiguous()
-            out = sort_pairs(k, v, 0, 64 - wbits)
+            pairs = torch.cat(ts) if len(ts) > 1 else ts[0].contiguous()
+            pairs = sort_pairs_aos(pairs, 0, 64 - wbits)
             torch.cuda.synchronize()
-            return out
+            return pairs[0::2], pairs[1::2]  # views; no copy until needed
         ks, vs = [], []
         for p, chunks in parts.items():
             for c in chunks:

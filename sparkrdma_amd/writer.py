@@ -36,18 +36,30 @@ HEADER_W = 8  # bytes per key
 
 
 def pack_partition_segment(keys: np.ndarray, values: Optional[np.ndarray]) -> bytes:
-    kb = keys.astype("<u8", copy=False).tobytes()
+    """AoS record layout: [key u64 | value bytes] interleaved per record.
+
+    Interleaving doubles the GPU scatter's per-digit write-burst length
+    (the partition kernel writes the identical layout) and lets the
+    reduce side consume fetched chunks without a split/concat step.
+    """
     if values is None or values.size == 0:
-        return kb
-    return kb + values.tobytes()
+        return keys.astype("<u8", copy=False).tobytes()
+    n = len(keys)
+    vw = values.shape[1]
+    rec = np.empty((n, HEADER_W + vw), dtype=np.uint8)
+    rec[:, :HEADER_W] = keys.astype("<u8", copy=False).view(np.uint8).reshape(n, 8)
+    rec[:, HEADER_W:] = values
+    return rec.tobytes()
 
 
 def unpack_partition_segment(buf, value_width: int):
-    n = len(buf) // (HEADER_W + value_width)
-    arr = np.frombuffer(buf, dtype=np.uint8)
-    keys = arr[:n * HEADER_W].view("<u8")
-    values = (arr[n * HEADER_W: n * (HEADER_W + value_width)]
-              .reshape(n, value_width) if value_width else None)
+    if value_width == 0:
+        return np.frombuffer(buf, dtype="<u8"), None
+    rec_w = HEADER_W + value_width
+    n = len(buf) // rec_w
+    arr = np.frombuffer(buf, dtype=np.uint8)[:n * rec_w].reshape(n, rec_w)
+    keys = arr[:, :HEADER_W].copy().view("<u8").reshape(n)
+    values = arr[:, HEADER_W:]
     return keys, values
 
 
@@ -213,8 +225,9 @@ class ShuffleWriter:
                 nb = int(seg_bytes[p])
                 table.put(p, off, nb, key)
                 if nb:
+                    # AoS segment: record r's key at +r*16, value at +r*16+8
                     key_dst[p] = base + off
-                    val_dst[p] = base + off + counts[p] * 8
+                    val_dst[p] = base + off + (8 if has_val else 0)
                 off += nb
                 self.metrics.bytes_written += nb
             blocks.append(blk)
@@ -223,7 +236,8 @@ class ShuffleWriter:
         hs.radix_scatter(keys.data_ptr(),
                          vals.data_ptr() if has_val else 0,
                          n, shift, nbits_eff, hist.data_ptr(),
-                         kd.data_ptr(), vd.data_ptr(), stream, hash_mix)
+                         kd.data_ptr(), vd.data_ptr(), stream, hash_mix,
+                         1 if has_val else 0)
         torch.cuda.synchronize()
         self.metrics.records_written += n
         mgr.keep_alive(self.handle, self.map_id, blocks)

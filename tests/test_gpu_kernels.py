@@ -56,6 +56,34 @@ def test_sort_pairs_full(hs, n, onesweep):
     assert np.array_equal(v_sorted.cpu().numpy(), k_sorted.cpu().numpy())
 
 
+@pytest.mark.parametrize("n", [1000, 4096, 3_000_000])
+def test_sort_pairs_aos(hs, n):
+    from sparkrdma_amd.ops.radix import sort_pairs_aos
+    keys = rand_keys(n, seed=n + 1)
+    pairs = torch.stack([keys, keys], dim=1).reshape(-1).contiguous()
+    out = sort_pairs_aos(pairs)
+    torch.cuda.synchronize()
+    got_k = out[0::2].cpu().numpy().view(np.uint64)
+    got_v = out[1::2].cpu().numpy().view(np.uint64)
+    want = np.sort(keys.cpu().numpy().view(np.uint64))
+    assert np.array_equal(got_k, want)
+    assert np.array_equal(got_v, got_k)
+
+
+def test_sort_aos_stability(hs):
+    from sparkrdma_amd.ops.radix import sort_pairs_aos
+    n = 400_000
+    rng = np.random.default_rng(4)
+    k = (rng.integers(0, 500, n, dtype=np.uint64))
+    keys = torch.from_numpy(k.view(np.int64)).cuda()
+    idx = torch.arange(n, dtype=torch.int64).cuda()
+    pairs = torch.stack([keys, idx], dim=1).reshape(-1).contiguous()
+    out = sort_pairs_aos(pairs, 0, 16)
+    torch.cuda.synchronize()
+    order = np.argsort(k, kind="stable")
+    assert np.array_equal(out[1::2].cpu().numpy(), order)
+
+
 def test_onesweep_matches_3kernel_repeated(hs):
     """Race screen: the lookback path must be deterministic across runs
     and identical to the 3-kernel path."""
