@@ -51,6 +51,14 @@ def main():
             print(f"sort {passes}p {tag} (bits 0..{end_bit}): {t*1e3:8.2f} ms  "
                   f"{n/t/1e9:6.2f} Grec/s  {n*rec_bytes/t/1e9:7.1f} GB/s payload")
 
+    from sparkrdma_amd.ops.radix import sort_pairs_aos
+    pairs = torch.stack([keys, vals], dim=1).reshape(-1).contiguous()
+    for passes, end_bit in ((7, 56), (8, 64)):
+        pp = pairs.clone()
+        t = bench(lambda: sort_pairs_aos(pp, 0, end_bit))
+        print(f"sort {passes}p AoS-onesweep (bits 0..{end_bit}): {t*1e3:8.2f} ms  "
+              f"{n/t/1e9:6.2f} Grec/s  {n*rec_bytes/t/1e9:7.1f} GB/s payload")
+
     # torch baseline for context
     t = bench(lambda: torch.sort(keys)[0])
     print(f"torch.sort keys-only baseline: {t*1e3:8.2f} ms  {n/t/1e9:6.2f} Grec/s")

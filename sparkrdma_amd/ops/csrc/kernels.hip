@@ -43,10 +43,11 @@ __device__ __forceinline__ uint32_t wave_inclusive_scan(uint32_t v) {
   return v;
 }
 
-// Exclusive scan of arr[nd] in LDS (nd pow2); sums: >=256 u32.
+// Exclusive scan of arr[nd] in LDS (nd pow2); sums: >=BS u32.
+template <int BS = BLOCK>
 __device__ void block_exscan(uint32_t* arr, uint32_t* sums, int nd) {
   const int tid = threadIdx.x;
-  const int per = (nd + BLOCK - 1) / BLOCK;  // >= 1
+  const int per = (nd + BS - 1) / BS;  // >= 1
   const int lo = tid * per;
   uint32_t run = 0;
 #pragma unroll 1
@@ -61,7 +62,7 @@ __device__ void block_exscan(uint32_t* arr, uint32_t* sums, int nd) {
   if (tid < kWave) {
     uint32_t off = 0;
 #pragma unroll 1
-    for (int c = 0; c < BLOCK / kWave; ++c) {
+    for (int c = 0; c < BS / kWave; ++c) {
       uint32_t v = sums[c * kWave + tid];
       uint32_t inc = wave_inclusive_scan(v);
       sums[c * kWave + tid] = off + inc - v;
@@ -399,31 +400,32 @@ __global__ __launch_bounds__(BLOCK) void onesweep_hist_all_kernel(
 // dwordx4 load per element, a pair LDS exchange, ONE 16-byte store per
 // element, and 2x-longer digit write bursts (measured: scattered-write
 // bandwidth doubles from 128 B to 256 B bursts — profiles/).
-template <bool HAS_VAL, int IT, bool AOS>
-__global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
+template <bool HAS_VAL, int IT, bool AOS, int BS = BLOCK>
+__global__ __launch_bounds__(BS) void onesweep_pass_kernel(
     const uint64_t* __restrict__ keys, const uint64_t* __restrict__ vals,
     uint32_t n, int shift, uint64_t* __restrict__ desc /* [nb][256] */,
     uint32_t* __restrict__ ticket,
     const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst) {
   constexpr int ND = 256;
   extern __shared__ char smem_raw[];
-  constexpr int TILE_T = BLOCK * IT;
+  constexpr int TILE_T = BS * IT;
+  constexpr int NWT = BS / kWave;
   using u64x2 = __attribute__((ext_vector_type(2))) unsigned long long;
   uint64_t* exch = reinterpret_cast<uint64_t*>(smem_raw);  // [TILE_T] or
   u64x2* exch2 = reinterpret_cast<u64x2*>(smem_raw);       // [TILE_T] pairs
   uint32_t* counters = reinterpret_cast<uint32_t*>(
       exch + (AOS ? 2 * TILE_T : TILE_T));
-  uint32_t* start = counters + NW * ND;
+  uint32_t* start = counters + NWT * ND;
   uint32_t* pref = start + ND;
-  uint32_t* sums = pref + ND;
-  uint32_t* vb_sh = sums + BLOCK;
+  uint32_t* sums = pref + ND;  // [BS]
+  uint32_t* vb_sh = sums + BS;
   uint8_t* dsort = reinterpret_cast<uint8_t*>(vb_sh + 4);  // [TILE_T], !AOS
 
   const int tid = threadIdx.x;
   const int lane = tid & (kWave - 1);
   const int wave = tid >> 6;
   if (tid == 0) *vb_sh = atomicAdd(ticket, 1);
-  for (int d = tid; d < NW * ND; d += BLOCK) counters[d] = 0;
+  for (int d = tid; d < NWT * ND; d += BS) counters[d] = 0;
   __syncthreads();
   const uint32_t b = *vb_sh;  // execution-ordered virtual block id
   const uint64_t tile_start = (uint64_t)b * TILE_T;
@@ -466,10 +468,10 @@ __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
   __syncthreads();
 
   // per-digit wave-exclusive scan + block totals into start[]
-  for (int d = tid; d < ND; d += BLOCK) {
+  for (int d = tid; d < ND; d += BS) {
     uint32_t run = 0;
 #pragma unroll
-    for (int w = 0; w < NW; ++w) {
+    for (int w = 0; w < NWT; ++w) {
       uint32_t t = counters[w * ND + d];
       counters[w * ND + d] = run;
       run += t;
@@ -508,7 +510,7 @@ __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
     }
   }
   // block-local digit starts (exclusive scan of totals)
-  block_exscan(start, sums, ND);  // includes the needed __syncthreads
+  block_exscan<BS>(start, sums, ND);  // includes the needed __syncthreads
 
   if (AOS) {
     // pair exchange + one dwordx4 store per element
@@ -524,7 +526,7 @@ __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
     __syncthreads();
 #pragma unroll
     for (int i = 0; i < IT; ++i) {
-      uint32_t j = i * BLOCK + tid;
+      uint32_t j = i * BS + tid;
       if (j < tile_n) {
         u64x2 kv = exch2[j];
         uint32_t d = (uint32_t)((uint64_t)kv.x >> shift) & (ND - 1);
@@ -549,7 +551,7 @@ __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
   __syncthreads();
 #pragma unroll
   for (int i = 0; i < IT; ++i) {
-    uint32_t j = i * BLOCK + tid;
+    uint32_t j = i * BS + tid;
     if (j < tile_n) {
       uint64_t k = exch[j];
       uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
@@ -568,7 +570,7 @@ __global__ __launch_bounds__(BLOCK) void onesweep_pass_kernel(
     __syncthreads();
 #pragma unroll 1
     for (int i = 0; i < IT; ++i) {
-      uint32_t j = i * BLOCK + tid;
+      uint32_t j = i * BS + tid;
       if (j < tile_n) {
         uint32_t d = dsort[j];
         uint32_t off = pref[d] + (j - start[d]);
@@ -892,8 +894,8 @@ static int onesweep_sort_impl(uintptr_t keys, uintptr_t vals,
   HIP_CHECK(hipGetLastError());
   size_t lds_soa = (size_t)OS_TILE * 8 + (size_t)NW * 256 * 4 + 256 * 4 * 2 +
                    BLOCK * 4 + 16 + OS_TILE;
-  size_t lds_aos = (size_t)OS_TILE * 16 + (size_t)NW * 256 * 4 +
-                   256 * 4 * 2 + BLOCK * 4 + 16;
+  size_t lds_aos = (size_t)OS_TILE * 16 + (size_t)(512 / kWave) * 256 * 4 +
+                   256 * 4 * 2 + 512 * 4 + 16;
   size_t lds = aos ? lds_aos : lds_soa;
   static bool attr_set = false;
   if (!attr_set) {
@@ -918,8 +920,8 @@ static int onesweep_sort_impl(uintptr_t keys, uintptr_t vals,
     HIP_CHECK(hipMemsetAsync(ticket, 0, 16, s));
     HIP_CHECK(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
     if (aos) {
-      hipLaunchKernelGGL((onesweep_pass_kernel<true, OS_ITEMS, true>),
-                         dim3(nb), dim3(BLOCK), lds, s,
+      hipLaunchKernelGGL((onesweep_pass_kernel<true, OS_ITEMS / 2, true, 512>),
+                         dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          start_bit + p * 8, desc, ticket, key_dst, val_dst);
     } else if (vals) {
