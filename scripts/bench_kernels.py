@@ -66,3 +66,21 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+def tile_ab():
+    import torch
+    from sparkrdma_amd.ops import load
+    from sparkrdma_amd.ops.radix import sort_pairs_aos
+    m = load()
+    n = 64_000_000
+    keys = torch.randint(-2**63, 2**63 - 1, (n,), dtype=torch.int64, device="cuda")
+    pairs = torch.stack([keys, keys], dim=1).reshape(-1).contiguous()
+    for tile in (4096, 8192):
+        m.set_aos_tile(tile)
+        pp = pairs.clone()
+        t = bench(lambda: sort_pairs_aos(pp, 0, 64))
+        print(f"AoS tile {tile}: {t*1e3:8.2f} ms  {n*16/t/1e9:7.1f} GB/s")
+    m.set_aos_tile(4096)
+
+if __name__ == "__main__" and os.environ.get("TILE_AB"):
+    tile_ab()

@@ -858,15 +858,21 @@ size_t sort_workspace_bytes(uint32_t n) {
 
 // Onesweep sort: ws layout = totals u32[passes*256] | key_dst u64[256] |
 // val_dst u64[256] | ticket u32 (+pad) | desc u32[nb*256].
-constexpr int OS_ITEMS = 16;  // 4096-elem tiles (32 measured slower: occupancy loss beats longer digit runs)
-constexpr int OS_TILE = BLOCK * OS_ITEMS;
+constexpr int OS_ITEMS = 16;  // SoA: 4096-elem tiles @ 256 threads
+// AoS tile selected at runtime: 4096 (512thr x IT8, 2 blocks/CU) or
+// 8192 (512thr x IT16, 1 block/CU, 512B write bursts)
+static int g_aos_tile = 4096;
+void set_aos_tile(int t) { g_aos_tile = t; }
 
+static inline uint32_t os_num_tiles_t(uint32_t n, int tile) {
+  return (uint32_t)(((uint64_t)n + tile - 1) / tile);
+}
 static inline uint32_t os_num_tiles(uint32_t n) {
-  return (uint32_t)(((uint64_t)n + OS_TILE - 1) / OS_TILE);
+  return os_num_tiles_t(n, BLOCK * OS_ITEMS);
 }
 
 size_t onesweep_workspace_bytes(uint32_t n, int passes) {
-  uint32_t nb = os_num_tiles(n);
+  uint32_t nb = os_num_tiles_t(n, 4096);  // smallest tile = max desc size
   return (size_t)passes * 256 * 4 + 256 * 8 * 2 + 16 +
          (size_t)nb * 256 * 8;
 }
@@ -878,7 +884,8 @@ static int onesweep_sort_impl(uintptr_t keys, uintptr_t vals,
                               uintptr_t tmp_keys, uintptr_t tmp_vals,
                               uint32_t n, int start_bit, int end_bit,
                               uintptr_t ws, hipStream_t s, int aos) {
-  uint32_t nb = os_num_tiles(n);
+  const int aos_tile = g_aos_tile;
+  uint32_t nb = aos ? os_num_tiles_t(n, aos_tile) : os_num_tiles(n);
   int passes = (end_bit - start_bit + 7) / 8;
   uint32_t* totals = reinterpret_cast<uint32_t*>(ws);
   uint64_t* key_dst = reinterpret_cast<uint64_t*>(totals + (size_t)passes * 256);
@@ -892,9 +899,9 @@ static int onesweep_sort_impl(uintptr_t keys, uintptr_t vals,
                      reinterpret_cast<const uint64_t*>(keys), n, start_bit,
                      passes, totals, aos ? 2 : 1);
   HIP_CHECK(hipGetLastError());
-  size_t lds_soa = (size_t)OS_TILE * 8 + (size_t)NW * 256 * 4 + 256 * 4 * 2 +
-                   BLOCK * 4 + 16 + OS_TILE;
-  size_t lds_aos = (size_t)OS_TILE * 16 + (size_t)(512 / kWave) * 256 * 4 +
+  size_t lds_soa = (size_t)BLOCK * OS_ITEMS * 8 + (size_t)NW * 256 * 4 +
+                   256 * 4 * 2 + BLOCK * 4 + 16 + BLOCK * OS_ITEMS;
+  size_t lds_aos = (size_t)aos_tile * 16 + (size_t)(512 / kWave) * 256 * 4 +
                    256 * 4 * 2 + 512 * 4 + 16;
   size_t lds = aos ? lds_aos : lds_soa;
   static bool attr_set = false;
@@ -919,8 +926,13 @@ static int onesweep_sort_impl(uintptr_t keys, uintptr_t vals,
     HIP_CHECK(hipGetLastError());
     HIP_CHECK(hipMemsetAsync(ticket, 0, 16, s));
     HIP_CHECK(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
-    if (aos) {
-      hipLaunchKernelGGL((onesweep_pass_kernel<true, OS_ITEMS / 2, true, 512>),
+    if (aos && aos_tile == 8192) {
+      hipLaunchKernelGGL((onesweep_pass_kernel<true, 16, true, 512>),
+                         dim3(nb), dim3(512), lds, s,
+                         reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
+                         start_bit + p * 8, desc, ticket, key_dst, val_dst);
+    } else if (aos) {
+      hipLaunchKernelGGL((onesweep_pass_kernel<true, 8, true, 512>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          start_bit + p * 8, desc, ticket, key_dst, val_dst);

@@ -99,30 +99,47 @@ class TeraSort:
         t_write = time.perf_counter()
         eng.barrier()
         lo, hi = rank * self.ppe, (rank + 1) * self.ppe - 1
-        reader = eng.manager.get_reader(handle, lo, hi)
-        parts = reader.collect_partitions()
-        t_fetch = time.perf_counter()
-        sorted_parts = self._reduce(parts)
-        t_sort = time.perf_counter()
+        # pipelined reduce: H chunk readers start fetching concurrently at
+        # construction; chunk h sorts while later chunks' one-sided copies
+        # are still in flight (overlaps xGMI fetch with the radix sort —
+        # only pays off when fetches are remote, i.e. multi-GPU)
+        H = min(4, self.ppe)   # ppe is pow2, so per = ppe/H is exact
+        per = self.ppe // H
+        spans = [(lo + h * per, lo + (h + 1) * per - 1) for h in range(H)]
+        readers = [eng.manager.get_reader(handle, a, b) for a, b in spans]
+        t_fetch_total = 0.0
+        t_sort_total = 0.0
+        outs = []
+        remote = 0
+        chunk_shared_bits = (self.R // per - 1).bit_length()
+        for reader in readers:
+            tf = time.perf_counter()
+            parts = reader.collect_partitions()
+            ts_ = time.perf_counter()
+            outs.append(self._reduce(parts, chunk_shared_bits))
+            t_fetch_total += ts_ - tf
+            t_sort_total += time.perf_counter() - ts_
+            remote += reader.metrics.remote_bytes_read
         if self.validate:
-            self._validate(sorted_parts, lo)
+            for (a, _b), out in zip(spans, outs):
+                self._validate(out, a, per)
         eng.unregister_shuffle(handle)
         dt = time.perf_counter() - t0
         return TeraSortResult(
             seconds=dt, records=self.n,
             bytes_sorted=self.n * self.RECORD_BYTES,
-            write_s=t_write - t0, fetch_s=t_fetch - t_write,
-            sort_s=t_sort - t_fetch,
-            remote_bytes=reader.metrics.remote_bytes_read)
+            write_s=t_write - t0, fetch_s=t_fetch_total,
+            sort_s=t_sort_total, remote_bytes=remote)
 
-    def _reduce(self, parts: dict):
-        """One batched sort per rank: all owned partitions share the top
-        log2(W) key bits (the rank prefix), so sorting the concatenated
-        fetches over bits [0, 64 - log2(W)) yields the rank's fully
-        sorted output in a single kernel sequence — avoiding per-partition
-        launch/workspace overhead (measured: R=64 beat R=256 at 8 GB
-        despite one extra radix pass)."""
-        wbits = (self.engine.world_size - 1).bit_length()
+    def _reduce(self, parts: dict, shared_bits: Optional[int] = None):
+        """One batched sort per fetched chunk: all partitions of a chunk
+        share their top `shared_bits` key bits (the chunk's partition
+        prefix), so sorting the concatenated fetches over the remaining
+        low bits yields the chunk's fully sorted output in a single kernel
+        sequence — avoiding per-partition launch/workspace overhead
+        (measured: R=64 beat R=256 at 8 GB despite one extra radix pass)."""
+        wbits = (shared_bits if shared_bits is not None
+                 else (self.engine.world_size - 1).bit_length())
         if self.device == "cuda":
             import torch
             from ..ops.radix import sort_pairs_aos
@@ -148,7 +165,7 @@ class TeraSort:
         order = np.argsort(k, kind="stable")
         return k[order], v[order]
 
-    def _validate(self, sorted_out, lo: int) -> None:
+    def _validate(self, sorted_out, lo: int, span: Optional[int] = None) -> None:
         k, v = sorted_out
         if k is None:
             return
@@ -157,10 +174,10 @@ class TeraSort:
             vu = v.cpu().numpy().view(np.uint64)
         else:
             ku, vu = k, v.view(np.uint64).reshape(-1)
-        assert np.all(ku[1:] >= ku[:-1]), "rank output not sorted"
+        assert np.all(ku[1:] >= ku[:-1]), "chunk output not sorted"
         assert np.array_equal(ku, vu.reshape(-1)), "payload corrupted"
         pids = self.part.partition_ids(ku)
-        hi = lo + self.ppe - 1
+        hi = lo + (span or self.ppe) - 1
         assert np.all((pids >= lo) & (pids <= hi)), "foreign keys in range"
         log.info("validated %d records in partitions [%d, %d]", len(ku), lo, hi)
 
