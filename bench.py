@@ -27,6 +27,11 @@ def main():
                     default=float(os.environ.get("TERASORT_GB_PER_GPU", 40)))
     ap.add_argument("--mode", choices=["framework", "rccl"],
                     default=os.environ.get("TERASORT_MODE", "framework"))
+    ap.add_argument("--workload",
+                    choices=["terasort", "pagerank", "join", "groupby"],
+                    default="terasort",
+                    help="terasort is the headline metric; others cover "
+                         "the remaining BASELINE configs")
     ap.add_argument("--partitions-per-executor", type=int, default=0,
                     help="0 = auto (pow2, ~128 per GPU)")
     ap.add_argument("--validate", action="store_true")
@@ -73,10 +78,27 @@ def main():
         conf.max_bytes_in_flight = 8 << 30
     eng = Engine(conf, rank=rank, world_size=world)
 
-    ts = TeraSort(eng, n_rec, partitions_per_executor=ppe,
-                  device="cuda" if use_cuda else "cpu",
-                  mode=args.mode if use_cuda else "framework",
-                  validate=args.validate)
+    device = "cuda" if use_cuda else "cpu"
+    if args.workload == "terasort":
+        ts = TeraSort(eng, n_rec, partitions_per_executor=ppe,
+                      device=device,
+                      mode=args.mode if use_cuda else "framework",
+                      validate=args.validate)
+    elif args.workload == "pagerank":
+        from sparkrdma_amd.workloads.pagerank import PageRank
+        # 19 GB edge list analog: 16 B records
+        edges = n_rec
+        ts = PageRank(eng, num_vertices=1 << 26, edges_per_executor=edges,
+                      partitions_per_executor=ppe, device=device,
+                      iterations=3)
+    elif args.workload == "join":
+        from sparkrdma_amd.workloads.sql_join import SortMergeJoin
+        ts = SortMergeJoin(eng, rows_per_executor=n_rec // 2,
+                           partitions_per_executor=ppe, device=device,
+                           key_space_bits=40, validate=args.validate)
+    else:
+        from sparkrdma_amd.workloads.groupby import GroupByKey
+        ts = GroupByKey(eng, rows_per_executor=min(n_rec, 1_000_000))
 
     def barrier_sync():
         if dist is not None:
@@ -89,7 +111,7 @@ def main():
     debug = os.environ.get("BENCH_DEBUG")
     for _ in range(args.warmup):
         r = ts.run_step()
-        if debug and rank == 0:
+        if debug and rank == 0 and hasattr(r, "write_s"):
             print(f"[warmup] total={r.seconds*1e3:.1f}ms write={r.write_s*1e3:.1f} "
                   f"fetch={r.fetch_s*1e3:.1f} sort={r.sort_s*1e3:.1f}",
                   file=sys.stderr)
@@ -108,13 +130,26 @@ def main():
         elapsed = float(t.item())
 
     ms_per_step = elapsed / args.steps * 1e3
-    total_bytes = n_rec * TeraSort.RECORD_BYTES * world
+    if args.workload == "terasort":
+        per_step_bytes = n_rec * TeraSort.RECORD_BYTES * world
+        metric = "terasort_sorted_gb_per_s"
+    elif args.workload == "pagerank":
+        per_step_bytes = results[0].iterations * n_rec * 16 * world
+        metric = "pagerank_edge_gb_per_s"
+    elif args.workload == "join":
+        per_step_bytes = n_rec * 16 * world  # both tables
+        metric = "join_row_gb_per_s"
+    else:
+        per_step_bytes = results[0].rows * world * 64
+        metric = "groupby_gb_per_s"
+    total_bytes = per_step_bytes
     value = total_bytes / (elapsed / args.steps) / 1e9  # GB/s whole job
-    remote_gb = sum(r.remote_bytes for r in results) / 1e9
+    remote_gb = sum(getattr(r, "remote_bytes", getattr(r, "shuffle_bytes", 0))
+                    for r in results) / 1e9
 
     if rank == 0:
         print(json.dumps({
-            "metric": "terasort_sorted_gb_per_s",
+            "metric": metric,
             "value": round(value, 3),
             "unit": "GB/s",
             "n_gpus": world,
@@ -127,12 +162,12 @@ def main():
             "dtype": "u64key+u64payload",
             "data": "synthetic",
             "config": {
-                "model": "terasort",
+                "model": args.workload,
                 "global_batch": n_rec * world,
                 "seq_len": TeraSort.RECORD_BYTES,
                 "dataset_gb": round(total_bytes / (1 << 30), 1),
                 "partitions": world * ppe,
-                "mode": ts.mode,
+                "mode": getattr(ts, "mode", args.mode),
                 "parallelism": f"shuffle{world}",
                 "wall_clock_s_per_job": round(elapsed / args.steps, 3),
                 "remote_gb_per_step": round(remote_gb / max(1, args.steps), 2),

@@ -109,8 +109,9 @@ class PageRank:
         if self.device == "cuda":
             import torch
             sums = torch.zeros(span, dtype=torch.float64, device="cuda")
+            from ..utils import as_device_i64
             for ref, data in reader:
-                t = data.view(torch.int64)  # AoS (dst, contrib) records
+                t = as_device_i64(data)  # AoS (dst, contrib) records
                 idx = t[0::2] - self.own_lo
                 sums.index_add_(0, idx, t[1::2].contiguous().view(torch.float64))
             self.ranks = (1.0 - DAMPING) / self.V + DAMPING * sums

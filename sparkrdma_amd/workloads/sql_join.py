@@ -131,7 +131,8 @@ class SortMergeJoin:
     @staticmethod
     def _concat_gpu(chunks):
         import torch
-        ts = [c.view(torch.int64) for c in chunks]  # AoS records
+        from ..utils import as_device_i64
+        ts = [as_device_i64(c) for c in chunks]  # AoS records
         if not ts:
             return None, None
         pairs = torch.cat(ts) if len(ts) > 1 else ts[0].contiguous()

@@ -143,8 +143,10 @@ class TeraSort:
         if self.device == "cuda":
             import torch
             from ..ops.radix import sort_pairs_aos
+            from ..utils import as_device_i64
             # fetched chunks are AoS (key,val) records: concat directly
-            ts = [c.view(torch.int64) for chunks in parts.values()
+            # (host-spilled chunks upload transparently)
+            ts = [as_device_i64(c) for chunks in parts.values()
                   for c in chunks]
             if not ts:
                 return None, None

@@ -211,13 +211,21 @@ class ShuffleWriter:
         if group:
             flushes.append(group)
         meta_key = make_key(mgr.executor_id, 1)
+        spill_groups = []   # (parts, total) that did not fit the HBM pool
         for parts in flushes:
             total = int(sum(seg_bytes[p] for p in parts))
             if total == 0:
                 for p in parts:
                     table.put(p, 0, 0, meta_key)
                 continue
-            blk = pool.get(total)
+            try:
+                blk = pool.get(total)
+            except MemoryError:
+                # pool pressure: this group spills to a host block after
+                # the scatter (reference keeps a disk file path for
+                # overflow — SURVEY §7.1 RdmaMappedFile row)
+                spill_groups.append((parts, total))
+                continue
             base = mgr.gpu.local_base(blk.segment_id)
             key = make_key(mgr.executor_id, blk.segment_id)
             off = blk.offset
@@ -231,6 +239,23 @@ class ShuffleWriter:
                 off += nb
                 self.metrics.bytes_written += nb
             blocks.append(blk)
+        spill_stage = None
+        if spill_groups:
+            # scatter spilled partitions into one transient device staging
+            # tensor, then copy each group to a pooled host block
+            stage_total = sum(t for _, t in spill_groups)
+            spill_stage = torch.empty(stage_total, dtype=torch.uint8,
+                                      device=dev)
+            soff = 0
+            for parts, total in spill_groups:
+                for p in parts:
+                    nb = int(seg_bytes[p])
+                    if nb:
+                        key_dst[p] = spill_stage.data_ptr() + soff
+                        val_dst[p] = spill_stage.data_ptr() + soff + \
+                            (8 if has_val else 0)
+                    # table filled after host placement below
+                    soff += nb
         kd = torch.from_numpy(key_dst).to(dev)
         vd = torch.from_numpy(val_dst).to(dev)
         hs.radix_scatter(keys.data_ptr(),
@@ -239,6 +264,22 @@ class ShuffleWriter:
                          kd.data_ptr(), vd.data_ptr(), stream, hash_mix,
                          1 if has_val else 0)
         torch.cuda.synchronize()
+        if spill_groups:
+            soff = 0
+            for parts, total in spill_groups:
+                hblk = mgr.pool.get(total)     # host pool
+                hseg = mgr.data_segment(hblk.segment_id)
+                hkey = make_key(mgr.executor_id, hblk.segment_id)
+                cpu_bytes = spill_stage[soff:soff + total].cpu().numpy()
+                hseg.write(hblk.offset, cpu_bytes.tobytes())
+                off = hblk.offset
+                for p in parts:
+                    nb = int(seg_bytes[p])
+                    table.put(p, off, nb, hkey)
+                    off += nb
+                    self.metrics.bytes_written += nb
+                soff += total
+                blocks.append(hblk)
         self.metrics.records_written += n
         mgr.keep_alive(self.handle, self.map_id, blocks)
         mgr.publish_map_output(self.handle, self.map_id, table_addr)
