@@ -93,14 +93,17 @@ class SortMergeJoin:
         if self.device == "cuda":
             import torch
             from ..ops.join import merge_join_sorted
-            from ..ops.radix import sort_pairs
-            for p in parts_a:
-                ak, av = self._concat_gpu(parts_a[p])
-                bk, bv = self._concat_gpu(parts_b[p])
-                if ak is None or bk is None:
-                    continue
-                ak, av = sort_pairs(ak, av, 0, self.low_bits)
-                bk, bv = sort_pairs(bk, bv, 0, self.low_bits)
+            from ..ops.radix import sort_pairs_aos
+            # batched: the whole owned partition range shares its top
+            # log2(W) key bits, so one AoS sort per side + one merge
+            # replaces ppe per-partition sorts (same trick as TeraSort)
+            wbits = (self.engine.world_size - 1).bit_length()
+            sort_bits = self.key_bits - wbits
+            a_chunks = [c for p in parts_a for c in parts_a[p]]
+            b_chunks = [c for p in parts_b for c in parts_b[p]]
+            ak, av = self._sort_side(a_chunks, sort_bits)
+            bk, bv = self._sort_side(b_chunks, sort_bits)
+            if ak is not None and bk is not None:
                 jk, ja, jb = merge_join_sorted(ak, av, bk, bv)
                 matches += jk.numel()
                 if self.validate and jk.numel():
@@ -129,13 +132,15 @@ class SortMergeJoin:
                           ma.remote_bytes_read + mb.remote_bytes_read)
 
     @staticmethod
-    def _concat_gpu(chunks):
+    def _sort_side(chunks, sort_bits):
         import torch
+        from ..ops.radix import sort_pairs_aos
         from ..utils import as_device_i64
         ts = [as_device_i64(c) for c in chunks]  # AoS records
         if not ts:
             return None, None
         pairs = torch.cat(ts) if len(ts) > 1 else ts[0].contiguous()
+        pairs = sort_pairs_aos(pairs, 0, sort_bits)
         return pairs[0::2].contiguous(), pairs[1::2].contiguous()
 
     @staticmethod
