@@ -68,8 +68,10 @@ def main():
     if use_cuda:
         conf.gpu_id = int(os.environ.get("LOCAL_RANK", rank)) % \
             torch.cuda.device_count()
-        # served blocks are pow2-rounded (buddy) -> up to 2x data, + slack
-        conf.hbm_pool_size = int((args.gb_per_gpu * 2.5 + 2) * (1 << 30))
+        # served blocks are pow2-rounded (buddy) -> up to 2x data, + slack;
+        # keep the pool tight enough that input + fetched + sort ping-pong
+        # still fit 288 GB at 40 GB/GPU
+        conf.hbm_pool_size = int((args.gb_per_gpu * 2.0 + 2) * (1 << 30))
         # few, large contiguous blocks: adjacent partitions of one map pack
         # into one 512 MiB block, so coalesced fetches become ~GB-scale
         # xGMI copies instead of per-partition ones

@@ -137,6 +137,9 @@ class GpuDataPlane:
         Blocking form (the fetcher's thread pool provides the async rim,
         like the reference's CQ threads)."""
         exec_id, _ = split_key(key)
+        # fetch runs on pool worker threads whose HIP current device
+        # defaults to 0 — pin it so per-peer streams land on OUR device
+        self.hs.set_device(self.device)
         base = self._resolve_base(key)
         dst = torch.empty(length, dtype=torch.uint8,
                           device=f"cuda:{self.device}")
