@@ -84,3 +84,18 @@ def tile_ab():
 
 if __name__ == "__main__" and os.environ.get("TILE_AB"):
     tile_ab()
+
+def digit_ab():
+    import torch
+    from sparkrdma_amd.ops.radix import sort_pairs_aos
+    n = 64_000_000
+    keys = torch.randint(-2**63, 2**63 - 1, (n,), dtype=torch.int64, device="cuda")
+    pairs = torch.stack([keys, keys], dim=1).reshape(-1).contiguous()
+    for bits_range in (56, 64):
+        for db in (7, 8):
+            pp = pairs.clone()
+            t = bench(lambda: sort_pairs_aos(pp, 0, bits_range, digit_bits=db))
+            print(f"AoS {bits_range}b digit={db}: {t*1e3:8.2f} ms  {n*16/t/1e9:7.1f} GB/s")
+
+if __name__ == "__main__" and os.environ.get("DIGIT_AB"):
+    digit_ab()

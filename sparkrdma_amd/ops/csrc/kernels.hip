@@ -369,26 +369,29 @@ constexpr uint64_t VAL_MASK = (1ull << 62) - 1;
 // Global digit totals of EVERY pass in one read (digit counts are
 // order-independent, so pass k's totals can be computed from pass 0's
 // input).
-template <int MAX_PASSES>
+template <int MAX_PASSES, int PBITS>
 __global__ __launch_bounds__(BLOCK) void onesweep_hist_all_kernel(
     const uint64_t* __restrict__ keys, uint32_t n, int start_bit, int passes,
-    uint32_t* __restrict__ totals /* [passes][256] */, int in_stride) {
-  __shared__ uint32_t cnt[NW][MAX_PASSES * 256];
+    uint32_t* __restrict__ totals /* [passes][1<<PBITS] */, int in_stride) {
+  constexpr int PD = 1 << PBITS;
+  __shared__ uint32_t cnt[NW][MAX_PASSES * PD];
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
-  for (int i = tid; i < NW * MAX_PASSES * 256; i += BLOCK)
+  for (int i = tid; i < NW * MAX_PASSES * PD; i += BLOCK)
     cnt[0][i] = 0;  // flat zero (cnt rows contiguous)
   __syncthreads();
   const uint64_t stride = (uint64_t)gridDim.x * BLOCK;
   for (uint64_t e = (uint64_t)blockIdx.x * BLOCK + tid; e < n; e += stride) {
     uint64_t k = keys[e * in_stride];
     for (int p = 0; p < passes; ++p) {
-      uint32_t d = (uint32_t)(k >> (start_bit + p * 8)) & 255;
-      atomicAdd(&cnt[wave][p * 256 + d], 1);
+      int sb = start_bit + p * PBITS;
+      if (sb > 64 - PBITS) sb = 64 - PBITS;  // last pass re-covers top bits
+      uint32_t d = (uint32_t)(k >> sb) & (PD - 1);
+      atomicAdd(&cnt[wave][p * PD + d], 1);
     }
   }
   __syncthreads();
-  for (int i = tid; i < passes * 256; i += BLOCK) {
+  for (int i = tid; i < passes * PD; i += BLOCK) {
     uint32_t s = 0;
 #pragma unroll
     for (int w = 0; w < NW; ++w) s += cnt[w][i];
@@ -400,13 +403,13 @@ __global__ __launch_bounds__(BLOCK) void onesweep_hist_all_kernel(
 // dwordx4 load per element, a pair LDS exchange, ONE 16-byte store per
 // element, and 2x-longer digit write bursts (measured: scattered-write
 // bandwidth doubles from 128 B to 256 B bursts — profiles/).
-template <bool HAS_VAL, int IT, bool AOS, int BS = BLOCK>
+template <bool HAS_VAL, int IT, bool AOS, int BS = BLOCK, int PBITS = 8>
 __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
     const uint64_t* __restrict__ keys, const uint64_t* __restrict__ vals,
-    uint32_t n, int shift, uint64_t* __restrict__ desc /* [nb][256] */,
+    uint32_t n, int shift, uint64_t* __restrict__ desc /* [nb][ND] */,
     uint32_t* __restrict__ ticket,
     const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst) {
-  constexpr int ND = 256;
+  constexpr int ND = 1 << PBITS;
   extern __shared__ char smem_raw[];
   constexpr int TILE_T = BS * IT;
   constexpr int NWT = BS / kWave;
@@ -456,7 +459,7 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
     uint64_t vm = __ballot(valid);
     uint32_t r = 0;
     if (valid) {
-      uint64_t match = match_lanes<8>(d, vm);
+      uint64_t match = match_lanes<PBITS>(d, vm);
       uint64_t lt = (1ull << lane) - 1;
       uint32_t rank_in_iter = (uint32_t)__popcll(match & lt);
       uint32_t c = my[d];
@@ -582,17 +585,18 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
 
 // digit bases for pass p from the all-pass totals; rec_bytes = 8 (SoA)
 // or 16 (AoS interleaved records)
+template <int ND = 256>
 __global__ void onesweep_digit_bases_kernel(
-    const uint32_t* __restrict__ totals /* [pass][256] */, int pass,
+    const uint32_t* __restrict__ totals /* [pass][ND] */, int pass,
     uint64_t out_keys, uint64_t out_vals, uint64_t* __restrict__ key_dst,
     uint64_t* __restrict__ val_dst, int rec_bytes) {
-  __shared__ uint32_t arr[256];
+  __shared__ uint32_t arr[ND];
   __shared__ uint32_t sums[BLOCK];
   const int tid = threadIdx.x;
-  if (tid < 256) arr[tid] = totals[pass * 256 + tid];
+  if (tid < ND) arr[tid] = totals[pass * ND + tid];
   __syncthreads();
-  block_exscan(arr, sums, 256);
-  if (tid < 256) {
+  block_exscan(arr, sums, ND);
+  if (tid < ND) {
     key_dst[tid] = out_keys + (uint64_t)arr[tid] * rec_bytes;
     val_dst[tid] = out_vals ? out_vals + (uint64_t)arr[tid] * rec_bytes : 0;
   }
@@ -880,73 +884,84 @@ size_t onesweep_workspace_bytes(uint32_t n, int passes) {
 // aos = 0: keys/vals are separate u64 arrays (SoA).
 // aos = 1: keys/tmp_keys point at interleaved (key,val) 16-byte records;
 //          vals/tmp_vals ignored.
-static int onesweep_sort_impl(uintptr_t keys, uintptr_t vals,
+// PBITS: digit width per pass. 7-bit digits double the per-digit write
+// burst (512 B at tile 4096) at the cost of more passes — the caller
+// picks per total bit count.
+template <int PBITS>
+static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
                               uintptr_t tmp_keys, uintptr_t tmp_vals,
                               uint32_t n, int start_bit, int end_bit,
                               uintptr_t ws, hipStream_t s, int aos) {
+  constexpr int PD = 1 << PBITS;
   const int aos_tile = g_aos_tile;
   uint32_t nb = aos ? os_num_tiles_t(n, aos_tile) : os_num_tiles(n);
-  int passes = (end_bit - start_bit + 7) / 8;
+  int passes = (end_bit - start_bit + PBITS - 1) / PBITS;
   uint32_t* totals = reinterpret_cast<uint32_t*>(ws);
-  uint64_t* key_dst = reinterpret_cast<uint64_t*>(totals + (size_t)passes * 256);
-  uint64_t* val_dst = key_dst + 256;
-  uint32_t* ticket = reinterpret_cast<uint32_t*>(val_dst + 256);
+  uint64_t* key_dst = reinterpret_cast<uint64_t*>(totals + (size_t)passes * PD);
+  uint64_t* val_dst = key_dst + PD;
+  uint32_t* ticket = reinterpret_cast<uint32_t*>(val_dst + PD);
   uint64_t* desc = reinterpret_cast<uint64_t*>(ticket + 4);
-  HIP_CHECK(hipMemsetAsync(totals, 0, (size_t)passes * 256 * 4, s));
+  HIP_CHECK(hipMemsetAsync(totals, 0, (size_t)passes * PD * 4, s));
   uint32_t hist_grid = nb < 1024 ? (nb ? nb : 1) : 1024;
-  hipLaunchKernelGGL(onesweep_hist_all_kernel<8>, dim3(hist_grid),
+  hipLaunchKernelGGL((onesweep_hist_all_kernel<10, PBITS>), dim3(hist_grid),
                      dim3(BLOCK), 0, s,
                      reinterpret_cast<const uint64_t*>(keys), n, start_bit,
                      passes, totals, aos ? 2 : 1);
   HIP_CHECK(hipGetLastError());
-  size_t lds_soa = (size_t)BLOCK * OS_ITEMS * 8 + (size_t)NW * 256 * 4 +
-                   256 * 4 * 2 + BLOCK * 4 + 16 + BLOCK * OS_ITEMS;
-  size_t lds_aos = (size_t)aos_tile * 16 + (size_t)(512 / kWave) * 256 * 4 +
-                   256 * 4 * 2 + 512 * 4 + 16;
+  size_t lds_soa = (size_t)BLOCK * OS_ITEMS * 8 + (size_t)NW * PD * 4 +
+                   PD * 4 * 2 + BLOCK * 4 + 16 + BLOCK * OS_ITEMS;
+  size_t lds_aos = (size_t)aos_tile * 16 + (size_t)(512 / kWave) * PD * 4 +
+                   PD * 4 * 2 + 512 * 4 + 16;
   size_t lds = aos ? lds_aos : lds_soa;
   static bool attr_set = false;
   if (!attr_set) {
     for (const void* f :
          {reinterpret_cast<const void*>(
-              &onesweep_pass_kernel<true, OS_ITEMS, false>),
+              &onesweep_pass_kernel<true, OS_ITEMS, false, BLOCK, PBITS>),
           reinterpret_cast<const void*>(
-              &onesweep_pass_kernel<false, OS_ITEMS, false>),
+              &onesweep_pass_kernel<false, OS_ITEMS, false, BLOCK, PBITS>),
           reinterpret_cast<const void*>(
-              &onesweep_pass_kernel<true, OS_ITEMS, true>)})
+              &onesweep_pass_kernel<true, 8, true, 512, PBITS>),
+          reinterpret_cast<const void*>(
+              &onesweep_pass_kernel<true, 16, true, 512, PBITS>)})
       (void)hipFuncSetAttribute(f, hipFuncAttributeMaxDynamicSharedMemorySize,
-                                (int)(lds_aos > lds_soa ? lds_aos : lds_soa));
+                                160 * 1024 - 1024);
     attr_set = true;
   }
   uintptr_t src_k = keys, src_v = vals, dst_k = tmp_keys, dst_v = tmp_vals;
   int cur = 0;
   for (int p = 0; p < passes; ++p) {
-    hipLaunchKernelGGL(onesweep_digit_bases_kernel, dim3(1), dim3(BLOCK), 0,
-                       s, totals, p, (uint64_t)dst_k, (uint64_t)dst_v,
-                       key_dst, val_dst, aos ? 16 : 8);
+    hipLaunchKernelGGL((onesweep_digit_bases_kernel<PD>), dim3(1),
+                       dim3(BLOCK), 0, s, totals, p, (uint64_t)dst_k,
+                       (uint64_t)dst_v, key_dst, val_dst, aos ? 16 : 8);
     HIP_CHECK(hipGetLastError());
     HIP_CHECK(hipMemsetAsync(ticket, 0, 16, s));
-    HIP_CHECK(hipMemsetAsync(desc, 0, (size_t)nb * 256 * 8, s));
+    HIP_CHECK(hipMemsetAsync(desc, 0, (size_t)nb * PD * 8, s));
+    int sb = start_bit + p * PBITS;
+    if (sb > 64 - PBITS) sb = 64 - PBITS;  // same clamp as hist_all
     if (aos && aos_tile == 8192) {
-      hipLaunchKernelGGL((onesweep_pass_kernel<true, 16, true, 512>),
+      hipLaunchKernelGGL((onesweep_pass_kernel<true, 16, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
-                         start_bit + p * 8, desc, ticket, key_dst, val_dst);
+                         sb, desc, ticket, key_dst, val_dst);
     } else if (aos) {
-      hipLaunchKernelGGL((onesweep_pass_kernel<true, 8, true, 512>),
+      hipLaunchKernelGGL((onesweep_pass_kernel<true, 8, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
-                         start_bit + p * 8, desc, ticket, key_dst, val_dst);
+                         sb, desc, ticket, key_dst, val_dst);
     } else if (vals) {
-      hipLaunchKernelGGL((onesweep_pass_kernel<true, OS_ITEMS, false>),
-                         dim3(nb), dim3(BLOCK), lds, s,
-                         reinterpret_cast<const uint64_t*>(src_k),
-                         reinterpret_cast<const uint64_t*>(src_v), n,
-                         start_bit + p * 8, desc, ticket, key_dst, val_dst);
+      hipLaunchKernelGGL(
+          (onesweep_pass_kernel<true, OS_ITEMS, false, BLOCK, PBITS>),
+          dim3(nb), dim3(BLOCK), lds, s,
+          reinterpret_cast<const uint64_t*>(src_k),
+          reinterpret_cast<const uint64_t*>(src_v), n, sb, desc, ticket,
+          key_dst, val_dst);
     } else {
-      hipLaunchKernelGGL((onesweep_pass_kernel<false, OS_ITEMS, false>),
-                         dim3(nb), dim3(BLOCK), lds, s,
-                         reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
-                         start_bit + p * 8, desc, ticket, key_dst, val_dst);
+      hipLaunchKernelGGL(
+          (onesweep_pass_kernel<false, OS_ITEMS, false, BLOCK, PBITS>),
+          dim3(nb), dim3(BLOCK), lds, s,
+          reinterpret_cast<const uint64_t*>(src_k), nullptr, n, sb, desc,
+          ticket, key_dst, val_dst);
     }
     HIP_CHECK(hipGetLastError());
     std::swap(src_k, dst_k);
@@ -960,16 +975,23 @@ int onesweep_sort_pairs_u64(uintptr_t keys, uintptr_t vals,
                             uintptr_t tmp_keys, uintptr_t tmp_vals,
                             uint32_t n, int start_bit, int end_bit,
                             uintptr_t ws, uintptr_t stream) {
-  return onesweep_sort_impl(keys, vals, tmp_keys, tmp_vals, n, start_bit,
-                            end_bit, ws,
-                            reinterpret_cast<hipStream_t>(stream), 0);
+  return onesweep_sort_tmpl<8>(keys, vals, tmp_keys, tmp_vals, n, start_bit,
+                               end_bit, ws,
+                               reinterpret_cast<hipStream_t>(stream), 0);
 }
 
 int onesweep_sort_aos_u64(uintptr_t pairs, uintptr_t tmp_pairs, uint32_t n,
                           int start_bit, int end_bit, uintptr_t ws,
                           uintptr_t stream) {
-  return onesweep_sort_impl(pairs, 0, tmp_pairs, 0, n, start_bit, end_bit,
-                            ws, reinterpret_cast<hipStream_t>(stream), 1);
+  return onesweep_sort_tmpl<8>(pairs, 0, tmp_pairs, 0, n, start_bit, end_bit,
+                               ws, reinterpret_cast<hipStream_t>(stream), 1);
+}
+
+int onesweep_sort_aos7_u64(uintptr_t pairs, uintptr_t tmp_pairs, uint32_t n,
+                           int start_bit, int end_bit, uintptr_t ws,
+                           uintptr_t stream) {
+  return onesweep_sort_tmpl<7>(pairs, 0, tmp_pairs, 0, n, start_bit, end_bit,
+                               ws, reinterpret_cast<hipStream_t>(stream), 1);
 }
 
 int sort_pairs_u64(uintptr_t keys, uintptr_t vals, uintptr_t tmp_keys,

@@ -73,7 +73,8 @@ def radix_partition(keys: torch.Tensor, vals: Optional[torch.Tensor],
 
 
 def sort_pairs_aos(pairs: torch.Tensor, start_bit: int = 0,
-                   end_bit: int = 64) -> torch.Tensor:
+                   end_bit: int = 64,
+                   digit_bits: Optional[int] = None) -> torch.Tensor:
     """Sort interleaved (key u64, val u64) 16-byte records by key bits
     [start_bit, end_bit). ``pairs`` is an int64 tensor of 2n elements.
     One dwordx4 load + one dwordx4 store per record per pass, and digit
@@ -84,15 +85,19 @@ def sort_pairs_aos(pairs: torch.Tensor, start_bit: int = 0,
     n = pairs.numel() // 2
     if n == 0:
         return pairs
+    bits = min(end_bit, 64) - start_bit
+    if digit_bits is None:
+        # 7-bit digits halve the scattered-write cost (512 B bursts) but
+        # add passes; pick whichever needs fewer weighted passes
+        digit_bits = 7 if -(-bits // 7) * 0.70 < -(-bits // 8) else 8
+    passes = -(-bits // digit_bits)
     tmp = torch.empty_like(pairs)
-    end_bit = start_bit + ((end_bit - start_bit + 7) // 8) * 8
-    end_bit = min(end_bit, 64)
-    passes = (end_bit - start_bit) // 8
     ws = torch.empty(m.onesweep_workspace_bytes(n, passes),
                      dtype=torch.uint8, device=pairs.device)
-    res = m.onesweep_sort_aos_u64(pairs.data_ptr(), tmp.data_ptr(), n,
-                                  start_bit, end_bit, ws.data_ptr(),
-                                  _stream())
+    fn = (m.onesweep_sort_aos7_u64 if digit_bits == 7
+          else m.onesweep_sort_aos_u64)
+    res = fn(pairs.data_ptr(), tmp.data_ptr(), n, start_bit,
+             start_bit + bits, ws.data_ptr(), _stream())
     return pairs if res == 0 else tmp
 
 

@@ -103,7 +103,10 @@ class TeraSort:
         # construction; chunk h sorts while later chunks' one-sided copies
         # are still in flight (overlaps xGMI fetch with the radix sort —
         # only pays off when fetches are remote, i.e. multi-GPU)
-        H = min(4, self.ppe)   # ppe is pow2, so per = ppe/H is exact
+        # chunking shrinks each chunk's shared top bits, which can add a
+        # whole radix pass; overlap only pays when fetches are remote
+        H = 1 if eng.world_size == 1 else min(4, self.ppe)
+        # ppe is pow2, so per = ppe/H is exact
         per = self.ppe // H
         spans = [(lo + h * per, lo + (h + 1) * per - 1) for h in range(H)]
         readers = [eng.manager.get_reader(handle, a, b) for a, b in spans]
