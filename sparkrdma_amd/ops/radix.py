@@ -87,9 +87,9 @@ def sort_pairs_aos(pairs: torch.Tensor, start_bit: int = 0,
         return pairs
     bits = min(end_bit, 64) - start_bit
     if digit_bits is None:
-        # 7-bit digits halve the scattered-write cost (512 B bursts) but
-        # add passes; pick whichever needs fewer weighted passes
-        digit_bits = 7 if -(-bits // 7) * 0.70 < -(-bits // 8) else 8
+        # measured: 7-bit digits (512 B bursts) do NOT beat 8-bit — the
+        # write-only burst model overestimates; pass floor is elsewhere
+        digit_bits = 8
     passes = -(-bits // digit_bits)
     tmp = torch.empty_like(pairs)
     ws = torch.empty(m.onesweep_workspace_bytes(n, passes),
