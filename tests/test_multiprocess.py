@@ -52,6 +52,69 @@ def _worker(rank, world_size, driver_port, shm_dir, q):
         raise
 
 
+def _pagerank_worker(rank, world, driver_port, shm_dir, q):
+    try:
+        import sys
+        sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+        from sparkrdma_amd.conf import ShuffleConf
+        from sparkrdma_amd.engine import Engine
+        from sparkrdma_amd.workloads.pagerank import PageRank
+
+        conf = ShuffleConf(shm_dir=shm_dir, max_buffer_allocation_size=1 << 30)
+        eng = Engine(conf, rank=rank, world_size=world, driver_port=driver_port)
+        pr = PageRank(eng, num_vertices=1 << 10, edges_per_executor=5000,
+                      partitions_per_executor=8, device="cpu",
+                      iterations=3, seed=11)
+        pr.run_step()
+        q.put((rank, pr.ranks.tobytes()))
+        eng.barrier()
+        eng.shutdown()
+    except BaseException as e:
+        import traceback
+        q.put((rank, f"ERROR: {e}\n{traceback.format_exc()}"))
+        raise
+
+
+def test_multiprocess_pagerank_matches_dense(tmp_path):
+    """2-process PageRank must equal the single-machine dense oracle —
+    validates cross-process contribution routing + aggregation."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    procs = [ctx.Process(target=_pagerank_worker,
+                         args=(r, world, port, str(tmp_path), q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    parts = {}
+    for _ in range(world):
+        rank, payload = q.get(timeout=120)
+        assert not isinstance(payload, str), f"rank {rank}: {payload}"
+        parts[rank] = np.frombuffer(payload, dtype=np.float64)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    got = np.concatenate([parts[r] for r in range(world)])
+    # rebuild the full edge list exactly as the workers did
+    from sparkrdma_amd.workloads.pagerank import PageRank
+    V = 1 << 10
+    srcs, dsts = [], []
+    for r in range(world):
+        lo, hi = r * V // world, (r + 1) * V // world
+        rng = np.random.default_rng(11 * 7919 + r)
+        srcs.append(rng.integers(lo, hi, 5000, dtype=np.uint64))
+        dsts.append(rng.integers(0, V, 5000, dtype=np.uint64))
+    want = PageRank.dense_reference(
+        V, np.concatenate(srcs).astype(np.int64),
+        np.concatenate(dsts).astype(np.int64), 3)
+    np.testing.assert_allclose(got, want, rtol=1e-12)
+
+
 @pytest.mark.parametrize("world_size", [2, 4])
 def test_multiprocess_shuffle(tmp_path, world_size):
     ctx = mp.get_context("spawn")
