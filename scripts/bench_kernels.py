@@ -75,7 +75,7 @@ def tile_ab():
     n = 64_000_000
     keys = torch.randint(-2**63, 2**63 - 1, (n,), dtype=torch.int64, device="cuda")
     pairs = torch.stack([keys, keys], dim=1).reshape(-1).contiguous()
-    for tile in (4096, 8192):
+    for tile in (2048, 4096, 8192):
         m.set_aos_tile(tile)
         pp = pairs.clone()
         t = bench(lambda: sort_pairs_aos(pp, 0, 64))

@@ -22,7 +22,7 @@ from __future__ import annotations
 
 import threading
 from dataclasses import dataclass, field
-from typing import Callable, Dict, List, Optional, Tuple
+from typing import Callable, Dict, List, Optional
 
 MIN_BLOCK = 16 << 10   # reference MIN_BLOCK_SIZE (RdmaBufferManager.java:93)
 

@@ -22,7 +22,6 @@ on the CQ poller thread there — RdmaChannel.java:731).
 from __future__ import annotations
 
 import logging
-import os
 import socket
 import threading
 import uuid

@@ -31,12 +31,10 @@ import time
 from dataclasses import dataclass
 from typing import Dict, List, Optional
 
-import numpy as np
-
 from .block_pool import BlockPool
 from .conf import ShuffleConf
 from .map_output import (DriverTable, MAP_ENTRY_SIZE, MapTaskOutput,
-                         make_key, split_key)
+                         make_key)
 from . import rpc
 from .segments import (FIRST_DATA_SEGMENT_ID, HostSegment, META_SEGMENT_ID,
                        META_TABLE_REGION_OFF, SegmentRegistry, segment_path)

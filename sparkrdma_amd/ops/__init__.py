@@ -8,8 +8,6 @@ paths silently (the HIP path must be the one that runs).
 from __future__ import annotations
 
 import importlib
-import os
-import sys
 
 _mod = None
 

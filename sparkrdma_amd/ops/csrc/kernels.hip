@@ -921,6 +921,8 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
           reinterpret_cast<const void*>(
               &onesweep_pass_kernel<false, OS_ITEMS, false, BLOCK, PBITS>),
           reinterpret_cast<const void*>(
+              &onesweep_pass_kernel<true, 4, true, 512, PBITS>),
+          reinterpret_cast<const void*>(
               &onesweep_pass_kernel<true, 8, true, 512, PBITS>),
           reinterpret_cast<const void*>(
               &onesweep_pass_kernel<true, 16, true, 512, PBITS>)})
@@ -941,6 +943,11 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
     if (sb > 64 - PBITS) sb = 64 - PBITS;  // same clamp as hist_all
     if (aos && aos_tile == 8192) {
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 16, true, 512, PBITS>),
+                         dim3(nb), dim3(512), lds, s,
+                         reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
+                         sb, desc, ticket, key_dst, val_dst);
+    } else if (aos && aos_tile == 2048) {
+      hipLaunchKernelGGL((onesweep_pass_kernel<true, 4, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          sb, desc, ticket, key_dst, val_dst);

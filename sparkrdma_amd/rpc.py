@@ -26,7 +26,7 @@ import socket
 import struct
 import threading
 from dataclasses import dataclass
-from typing import Callable, Dict, List, Optional
+from typing import List, Optional
 
 MSG_HELLO = 0
 MSG_ANNOUNCE = 1

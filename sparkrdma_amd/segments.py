@@ -25,9 +25,9 @@ from __future__ import annotations
 import mmap
 import os
 import threading
-from typing import Dict, Optional
+from typing import Dict
 
-from .map_output import make_key, split_key
+from .map_output import split_key
 
 # well-known segment ids (segment_id 0 is reserved == "unpublished")
 META_SEGMENT_ID = 1      # metadata: map-output tables + HBM slab handle table

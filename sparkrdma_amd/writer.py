@@ -24,7 +24,7 @@ from __future__ import annotations
 
 import pickle
 import time
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 import numpy as np
 
