@@ -99,3 +99,25 @@ def digit_ab():
 
 if __name__ == "__main__" and os.environ.get("DIGIT_AB"):
     digit_ab()
+
+def stage_ab():
+    import torch
+    from sparkrdma_amd.ops import load
+    from sparkrdma_amd.ops.radix import sort_pairs_aos
+    m = load()
+    n = 64_000_000
+    i = torch.arange(n, dtype=torch.int64, device="cuda")
+    keys = i * 0x9E3779B97F4A7C15
+    keys ^= keys >> 31
+    pairs = torch.stack([keys, keys], dim=1).reshape(-1).contiguous()
+    for stage, label in ((0, "rank only, no lookback walk"),
+                         (1, "rank+lookback only"), (2, "+exch+dummy-store"),
+                         (3, "full")):
+        m.set_pass_stage(stage)
+        pp = pairs.clone()
+        t = bench(lambda: sort_pairs_aos(pp, 0, 64))
+        print(f"stage {stage} ({label}): {t*1e3:8.2f} ms  ({t/8*1e3:.3f} ms/pass)")
+    m.set_pass_stage(3)
+
+if __name__ == "__main__" and os.environ.get("STAGE_AB"):
+    stage_ab()

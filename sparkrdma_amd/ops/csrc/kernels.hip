@@ -366,6 +366,35 @@ constexpr uint64_t FLAG_INC = 2ull << 62;
 constexpr uint64_t FLAG_MASK = 3ull << 62;
 constexpr uint64_t VAL_MASK = (1ull << 62) - 1;
 
+// Serial decoupled-lookback walk for digit `tid`: sum predecessors'
+// aggregates back to the nearest inclusive prefix, publish our own
+// inclusive, record the exclusive prefix in pref[tid]. (A windowed
+// 8-wide variant measured NO faster — the cost is waiting on the
+// publish frontier, not load throughput — so the caller overlaps this
+// wait with the LDS exchange instead.)
+template <int ND>
+__device__ __forceinline__ void lookback_walk(
+    uint64_t* __restrict__ desc, uint32_t b, int tid, uint64_t my_total,
+    uint32_t* __restrict__ pref) {
+  uint64_t run = 0;
+  for (int64_t j = (int64_t)b - 1; j >= 0;) {
+    uint64_t v = __hip_atomic_load(&desc[(uint64_t)j * ND + tid],
+                                   __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+    if ((v & FLAG_MASK) == 0) {
+      __builtin_amdgcn_s_sleep(2);
+      continue;
+    }
+    run += v & VAL_MASK;
+    if ((v & FLAG_MASK) == FLAG_INC) break;
+    --j;
+  }
+  __hip_atomic_store(&desc[(uint64_t)b * ND + tid],
+                     FLAG_INC | (run + my_total), __ATOMIC_RELAXED,
+                     __HIP_MEMORY_SCOPE_AGENT);
+  pref[tid] = (uint32_t)run;
+}
+
 // Global digit totals of EVERY pass in one read (digit counts are
 // order-independent, so pass k's totals can be computed from pass 0's
 // input).
@@ -408,7 +437,8 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
     const uint64_t* __restrict__ keys, const uint64_t* __restrict__ vals,
     uint32_t n, int shift, uint64_t* __restrict__ desc /* [nb][ND] */,
     uint32_t* __restrict__ ticket,
-    const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst) {
+    const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst,
+    int stage = 3) {
   constexpr int ND = 1 << PBITS;
   extern __shared__ char smem_raw[];
   constexpr int TILE_T = BS * IT;
@@ -483,62 +513,57 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
   }
   __syncthreads();
 
-  // publish aggregate (or inclusive for block 0), then lookback
+  // publish this block's AGGREGATE as early as possible — consumers'
+  // lookbacks are gated on it; our own walk is DEFERRED to overlap the
+  // LDS exchange below (the walk is wait-dominated: it costs ~30% of the
+  // pass when run serially here — profiles/r01)
+  uint64_t my_total = 0;
   if (tid < ND) {
-    uint64_t total = start[tid];
+    my_total = start[tid];
     uint64_t* slot = &desc[(uint64_t)b * ND + tid];
-    if (b == 0) {
-      __hip_atomic_store(slot, FLAG_INC | total, __ATOMIC_RELAXED,
+    if (stage == 0 || b == 0) {  // stage 0: ablation, WRONG results
+      __hip_atomic_store(slot, FLAG_INC | my_total, __ATOMIC_RELAXED,
                          __HIP_MEMORY_SCOPE_AGENT);
       pref[tid] = 0;
     } else {
-      __hip_atomic_store(slot, FLAG_AGG | total, __ATOMIC_RELAXED,
+      __hip_atomic_store(slot, FLAG_AGG | my_total, __ATOMIC_RELAXED,
                          __HIP_MEMORY_SCOPE_AGENT);
-      uint64_t run = 0;
-      for (int64_t j = (int64_t)b - 1; j >= 0;) {
-        uint64_t w = __hip_atomic_load(&desc[(uint64_t)j * ND + tid],
-                                       __ATOMIC_RELAXED,
-                                       __HIP_MEMORY_SCOPE_AGENT);
-        if ((w & FLAG_MASK) == 0) {
-          __builtin_amdgcn_s_sleep(1);
-          continue;
-        }
-        run += w & VAL_MASK;
-        if ((w & FLAG_MASK) == FLAG_INC) break;
-        --j;
-      }
-      __hip_atomic_store(slot, FLAG_INC | (run + total), __ATOMIC_RELAXED,
-                         __HIP_MEMORY_SCOPE_AGENT);
-      pref[tid] = (uint32_t)run;
     }
   }
   // block-local digit starts (exclusive scan of totals)
   block_exscan<BS>(start, sums, ND);  // includes the needed __syncthreads
 
   if (AOS) {
-    // pair exchange + one dwordx4 store per element
+    if (stage >= 2) {
+      // pair exchange first: the deferred lookback's wait overlaps it
 #pragma unroll
-    for (int i = 0; i < IT; ++i) {
-      uint64_t e = chunk + (uint64_t)i * kWave + lane;
-      if (e < n) {
-        uint32_t d = digrank[i] >> 16;
-        uint32_t j = start[d] + my[d] + (digrank[i] & 0xFFFF);
-        exch2[j] = u64x2{key_reg[i], val_reg[i]};
+      for (int i = 0; i < IT; ++i) {
+        uint64_t e = chunk + (uint64_t)i * kWave + lane;
+        if (e < n) {
+          uint32_t d = digrank[i] >> 16;
+          uint32_t j = start[d] + my[d] + (digrank[i] & 0xFFFF);
+          exch2[j] = u64x2{key_reg[i], val_reg[i]};
+        }
       }
     }
+    if (stage != 0 && b != 0 && tid < ND)
+      lookback_walk<ND>(desc, b, tid, my_total, pref);
+    if (stage < 2) return;  // ablation: rank/publish/lookback only
     __syncthreads();
+    const uint32_t off_mask = stage < 3 ? 1023u : 0xFFFFFFFFu;
 #pragma unroll
     for (int i = 0; i < IT; ++i) {
       uint32_t j = i * BS + tid;
       if (j < tile_n) {
         u64x2 kv = exch2[j];
         uint32_t d = (uint32_t)((uint64_t)kv.x >> shift) & (ND - 1);
-        uint32_t off = pref[d] + (j - start[d]);
+        uint32_t off = (pref[d] + (j - start[d])) & off_mask;
         reinterpret_cast<u64x2*>(key_dst[d])[off] = kv;
       }
     }
     return;
   }
+  if (b != 0 && tid < ND) lookback_walk<ND>(desc, b, tid, my_total, pref);
 
   // SoA path: key exchange + write-out, then val exchange + write-out
 #pragma unroll
@@ -867,6 +892,10 @@ constexpr int OS_ITEMS = 16;  // SoA: 4096-elem tiles @ 256 threads
 // 8192 (512thr x IT16, 1 block/CU, 512B write bursts)
 static int g_aos_tile = 4096;
 void set_aos_tile(int t) { g_aos_tile = t; }
+// ablation: 1 = rank+lookback only, 2 = full work but stores land in a
+// compact dummy window (isolates scattered-store cost), 3 = normal
+static int g_pass_stage = 3;
+void set_pass_stage(int s) { g_pass_stage = s; }
 
 static inline uint32_t os_num_tiles_t(uint32_t n, int tile) {
   return (uint32_t)(((uint64_t)n + tile - 1) / tile);
@@ -945,17 +974,17 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 16, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
-                         sb, desc, ticket, key_dst, val_dst);
+                         sb, desc, ticket, key_dst, val_dst, g_pass_stage);
     } else if (aos && aos_tile == 2048) {
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 4, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
-                         sb, desc, ticket, key_dst, val_dst);
+                         sb, desc, ticket, key_dst, val_dst, g_pass_stage);
     } else if (aos) {
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 8, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
-                         sb, desc, ticket, key_dst, val_dst);
+                         sb, desc, ticket, key_dst, val_dst, g_pass_stage);
     } else if (vals) {
       hipLaunchKernelGGL(
           (onesweep_pass_kernel<true, OS_ITEMS, false, BLOCK, PBITS>),
