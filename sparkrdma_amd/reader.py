@@ -21,6 +21,7 @@ was in the reference (it is pure byte-budget accounting).
 
 from __future__ import annotations
 
+import os
 import queue
 import random
 import threading
@@ -69,6 +70,23 @@ class FetchFailedError(RuntimeError):
     (reference RdmaShuffleFetcherIterator.scala:167)."""
 
 
+def parse_cpu_list(spec: str) -> list:
+    """Parse the reference's cpuList format '0-3,8,10-11'
+    (RdmaShuffleConf.scala:89) into a CPU id list for fetch-thread
+    affinity (the CQ-poller placement analog, RdmaThread.java:45-47)."""
+    cpus = []
+    for part in (spec or "").split(","):
+        part = part.strip()
+        if not part:
+            continue
+        if "-" in part:
+            a, b = part.split("-")
+            cpus.extend(range(int(a), int(b) + 1))
+        else:
+            cpus.append(int(part))
+    return cpus
+
+
 def coalesce_blocks(blocks: List[BlockRef], max_bytes: int,
                     max_blocks: int = 1 << 30) -> List[CoalescedFetch]:
     """Group blocks into minimal one-sided reads.
@@ -107,6 +125,7 @@ class FetcherIterator:
         self.end_partition = end_partition  # inclusive
         self.metrics = TaskMetrics()
         conf = manager.conf
+        self._cpus = parse_cpu_list(conf.cpu_list)
         self._max_bytes_in_flight = conf.max_bytes_in_flight
         self._read_block = conf.shuffle_read_block_size
         self._reqs_limit = conf.resolved_read_requests_limit()
@@ -117,12 +136,22 @@ class FetcherIterator:
         self._reqs_in_flight = 0
         self._outstanding = 0     # fetches not yet surfaced to the consumer
         self._pool = ThreadPoolExecutor(max_workers=num_workers,
-                                        thread_name_prefix="sparkrdma-fetch")
+                                        thread_name_prefix="sparkrdma-fetch",
+                                        initializer=self._pin_worker)
         self._rng = random.Random(seed)
         self._failed: Optional[BaseException] = None
         self._start()
 
     # ------------------------------------------------------------------
+
+    def _pin_worker(self) -> None:
+        """Spread fetch workers over conf.cpuList (poller affinity parity,
+        RdmaNode.java:222-279). No-op when the list is empty."""
+        if self._cpus:
+            try:
+                os.sched_setaffinity(0, set(self._cpus))
+            except OSError:
+                pass
 
     def _start(self) -> None:
         mgr = self.manager

@@ -69,3 +69,10 @@ def test_range_partitioner_terasort_property():
     # searchsorted(side=right) puts key == bounds[i] into partition i+1
     assert rp.partition_ids(np.array([rp.bounds[0]], dtype=np.uint64))[0] == 1
     assert rp.partition_ids(np.array([rp.bounds[0] - 1], dtype=np.uint64))[0] == 0
+
+
+def test_parse_cpu_list():
+    from sparkrdma_amd.reader import parse_cpu_list
+    assert parse_cpu_list("") == []
+    assert parse_cpu_list("0-3,8,10-11") == [0, 1, 2, 3, 8, 10, 11]
+    assert parse_cpu_list("5") == [5]

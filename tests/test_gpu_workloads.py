@@ -78,3 +78,13 @@ def test_hash_mix_partition_matches_cpu():
     # partition-ordered output matches CPU stable sort by hash pid
     order = np.argsort(pids, kind="stable")
     assert np.array_equal(keys_out.cpu().numpy().view(np.uint64), k[order])
+
+
+def test_terasort_rccl_mode_single_rank(engine):
+    """Stage-mode path at W=1 (pure AoS-less sort branch)."""
+    from sparkrdma_amd.workloads.terasort import TeraSort
+    ts = TeraSort(engine, records_per_executor=1_000_000,
+                  partitions_per_executor=32, device="cuda",
+                  mode="rccl", validate=True)
+    r = ts.run_step()
+    assert r.records == 1_000_000
