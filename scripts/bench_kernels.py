@@ -101,11 +101,11 @@ if __name__ == "__main__" and os.environ.get("DIGIT_AB"):
     digit_ab()
 
 def stage_ab():
-    import torch
+    import torch, sys
     from sparkrdma_amd.ops import load
     from sparkrdma_amd.ops.radix import sort_pairs_aos
     m = load()
-    n = 64_000_000
+    n = int(float(sys.argv[1]) * 1e6) if len(sys.argv) > 1 else 64_000_000
     i = torch.arange(n, dtype=torch.int64, device="cuda")
     keys = i * 0x9E3779B97F4A7C15
     keys ^= keys >> 31
