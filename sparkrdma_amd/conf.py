@@ -106,8 +106,9 @@ class ShuffleConf:
                              4 << 10, 1 << 32)
         self._validate_range("max_bytes_in_flight", self.max_bytes_in_flight,
                              self.shuffle_read_block_size, 1 << 40)
-        if self.transport not in ("auto", "shm", "ipc", "rccl"):
-            raise ConfError(f"transport must be auto|shm|ipc|rccl, got {self.transport!r}")
+        if self.transport not in ("auto", "shm", "ipc", "rccl", "tcp"):
+            raise ConfError(
+                f"transport must be auto|shm|ipc|rccl|tcp, got {self.transport!r}")
 
     @staticmethod
     def _validate_range(name: str, value: int, lo: int, hi: int) -> None:

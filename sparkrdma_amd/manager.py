@@ -89,8 +89,14 @@ class ShuffleManager:
         self.reader_stats = (ShuffleReaderStats(conf)
                              if conf.collect_shuffle_reader_stats else None)
         self.gpu = None
+        self._data_server = None
+        self._data_client = None
+        self._my_host = socket.gethostname()
         self._stopped = False
         if is_executor:
+            from .data_server import DataClient, DataServer
+            self._data_server = DataServer(self)
+            self._data_client = DataClient()
             self._connect_and_hello()
 
     # ------------------------------------------------------------------
@@ -119,8 +125,9 @@ class ShuffleManager:
             target=self._recv_loop, name=f"sparkrdma-exec{self.executor_id}-recv",
             daemon=True)
         self._recv_thread.start()
-        meta_path = ""  # path known only after app_id arrives; re-hello below
-        info = rpc.ExecutorInfo(self.executor_id, self.conf.driver_host, 0,
+        meta_path = ""  # path known only after app_id arrives
+        data_port = self._data_server.port if self._data_server else 0
+        info = rpc.ExecutorInfo(self.executor_id, self._my_host, data_port,
                                 self.conf.resolved_gpu_id(), meta_path)
         self._conn.send(rpc.MSG_HELLO, rpc.pack_hello(info))
         if not self._announce_evt.wait(self.conf.rdma_cm_event_timeout_ms / 1000):
@@ -174,7 +181,7 @@ class ShuffleManager:
             self._pool.preallocate(size, count)
         self.gpu = None
         if self.conf.transport in ("ipc", "rccl") or (
-                self.conf.transport == "auto" and _cuda_available()):
+                self.conf.transport in ("auto", "tcp") and _cuda_available()):
             from .gpu_plane import GpuDataPlane
             self.gpu = GpuDataPlane(self.conf, self.executor_id,
                                     self._meta_segment, self._registry)
@@ -253,6 +260,10 @@ class ShuffleManager:
         for mm in self._driver_tables.values():
             mm.close()
         self._driver_tables.clear()
+        if self._data_server is not None:
+            self._data_server.stop()
+        if self._data_client is not None:
+            self._data_client.close()
         if self._registry is not None:
             self._registry.close()
         for seg in self._data_segments.values():
@@ -337,6 +348,25 @@ class ShuffleManager:
         if self.gpu is None:
             raise RuntimeError("GPU data plane not initialized")
         return self.gpu.read_device(key, addr, length)
+
+    # --- cross-host fallback lane (TCP data servers) -------------------
+
+    def is_remote_host(self, exec_id: int) -> bool:
+        """True when exec_id's memory is NOT reachable one-sidedly from
+        this process (different host — or transport=tcp forcing, used by
+        tests and as a soft-RoCE-free multi-node mode)."""
+        if exec_id == self.executor_id:
+            return False
+        if self.conf.transport == "tcp":
+            return True
+        m = self._members.get(exec_id)
+        return m is not None and m.host != self._my_host
+
+    def tcp_read(self, exec_id: int, key: int, addr: int, length: int) -> bytes:
+        m = self._members.get(exec_id)
+        if m is None or m.port == 0:
+            raise RuntimeError(f"no data server known for executor {exec_id}")
+        return self._data_client.read(m.host, m.port, key, addr, length)
 
     def data_segment(self, seg_id: int) -> HostSegment:
         return self._data_segments[seg_id]

@@ -197,7 +197,10 @@ class FetcherIterator:
         from .gpu_plane import is_gpu_key
         t0 = time.perf_counter()
         try:
-            if is_gpu_key(f.key):
+            owner = split_key(f.key)[0]
+            if self.manager.is_remote_host(owner):
+                data = self.manager.tcp_read(owner, f.key, f.addr, f.length)
+            elif is_gpu_key(f.key):
                 data = self.manager.remote_read_device(f.key, f.addr, f.length)
             else:
                 data = self.manager.remote_read(f.key, f.addr, f.length)

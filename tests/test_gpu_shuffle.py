@@ -135,3 +135,43 @@ def test_cross_process_ipc_shuffle(tmp_path):
         assert p.exitcode == 0
     # each rank must have fetched ~half its data from the peer process
     assert remote_total > 0
+
+
+def test_tcp_serves_hbm_blocks(tmp_path):
+    """transport=tcp with CUDA: the writer lands data in HBM slabs; the
+    peer's read goes over the data server, which stages D2H — the
+    multi-node lane serving GPU memory."""
+    import numpy as np
+    import torch
+    from sparkrdma_amd.conf import ShuffleConf
+    from sparkrdma_amd.driver import Driver
+    from sparkrdma_amd.manager import ShuffleManager
+    from sparkrdma_amd.partitioner import RangePartitioner
+    from sparkrdma_amd.writer import unpack_partition_segment
+
+    conf = ShuffleConf(shm_dir=str(tmp_path), transport="tcp",
+                       hbm_pool_size=1 << 30, gpu_id=0)
+    driver = Driver(conf)
+    m0 = ShuffleManager(conf, executor_id=0, driver_port=driver.port)
+    m1 = ShuffleManager(conf, executor_id=1, driver_port=driver.port)
+    try:
+        R = 16
+        part = RangePartitioner.uniform(R)
+        handle = m0.register_shuffle(num_maps=1, num_partitions=R)
+        keys = torch.randint(-2**63, 2**63 - 1, (200_000,),
+                             dtype=torch.int64, device="cuda")
+        w = m0.get_writer(handle, 0)
+        w.write_device_batch(keys, keys.clone())
+        w.stop(True, partitioner=part)
+        reader = m1.get_reader(handle, 0, R - 1)
+        got = []
+        for ref, data in reader:
+            k, v = unpack_partition_segment(data, 8)
+            got.append(np.array(k))
+        want = np.sort(keys.cpu().numpy().view(np.uint64))
+        assert np.array_equal(np.sort(np.concatenate(got)), want)
+        assert reader.metrics.remote_bytes_read == 200_000 * 16
+    finally:
+        m0.stop()
+        m1.stop()
+        driver.stop()
