@@ -95,6 +95,9 @@ class _FakeManager:
     def get_map_task_output_table(self, handle):
         return [(0, 99)] * handle.num_maps  # table key 99 (executor 0)
 
+    def is_remote_host(self, exec_id):
+        return False
+
     def remote_read(self, key, addr, length):
         import time
         if key == 99:  # hop-2 table read: fabricate location entries
