@@ -6,6 +6,7 @@ Prints partition + sort throughput (records/s and effective GB/s).
 
 import os
 import sys
+import sys
 import time
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
@@ -121,3 +122,30 @@ def stage_ab():
 
 if __name__ == "__main__" and os.environ.get("STAGE_AB"):
     stage_ab()
+
+def timing_probe():
+    import torch
+    from sparkrdma_amd.ops import load
+    from sparkrdma_amd.ops.radix import sort_pairs_aos
+    m = load()
+    n = int(float(sys.argv[1]) * 1e6) if len(sys.argv) > 1 else 64_000_000
+    i = torch.arange(n, dtype=torch.int64, device="cuda")
+    keys = i * 0x9E3779B97F4A7C15
+    keys ^= keys >> 31
+    pairs = torch.stack([keys, keys], dim=1).reshape(-1).contiguous()
+    tbuf = torch.zeros(4, dtype=torch.int64, device="cuda")
+    m.set_timing_buf(tbuf.data_ptr())
+    pp = pairs.clone()
+    t = bench(lambda: sort_pairs_aos(pp, 0, 64), iters=1, warmup=1)
+    m.set_timing_buf(0)
+    torch.cuda.synchronize()
+    vals = tbuf.cpu().numpy() / 2  # 2 runs (warmup+timed)
+    nb = (n + 4095) // 4096
+    per_block_us = vals / (nb * 8) / 100.0  # 100 MHz realtime clock, 8 passes
+    names = ["phaseA+scan+publish", "exchange", "lookback+walk", "writeout"]
+    print(f"full sort: {t*1e3:.2f} ms; per-block phase means (us):")
+    for nm, v in zip(names, per_block_us):
+        print(f"  {nm:22s} {v:8.2f} us")
+
+if __name__ == "__main__" and os.environ.get("TIMING_PROBE"):
+    timing_probe()
