@@ -149,3 +149,30 @@ def timing_probe():
 
 if __name__ == "__main__" and os.environ.get("TIMING_PROBE"):
     timing_probe()
+
+def mode_ab():
+    import torch
+    from sparkrdma_amd.ops import load
+    from sparkrdma_amd.ops.radix import sort_pairs_aos
+    m = load()
+    n = int(float(sys.argv[1]) * 1e6) if len(sys.argv) > 1 else 64_000_000
+    i = torch.arange(n, dtype=torch.int64, device="cuda")
+    keys = i * 0x9E3779B97F4A7C15
+    keys ^= keys >> 31
+    pairs = torch.stack([keys, keys], dim=1).reshape(-1).contiguous()
+    ref = None
+    for mode, label in ((0, "lookback"), (1, "hist+scan")):
+        m.set_sort_mode(mode)
+        pp = pairs.clone()
+        t = bench(lambda: sort_pairs_aos(pp, 0, 64))
+        out = sort_pairs_aos(pairs.clone(), 0, 64)
+        torch.cuda.synchronize()
+        if ref is None:
+            ref = out
+        else:
+            assert torch.equal(out, ref), f"mode {mode} output differs!"
+        print(f"mode {mode} ({label}): {t*1e3:8.2f} ms")
+    m.set_sort_mode(1)
+
+if __name__ == "__main__" and os.environ.get("MODE_AB"):
+    mode_ab()

@@ -109,7 +109,8 @@ __device__ __forceinline__ uint64_t match_lanes(uint32_t digit,
 template <int NBITS>
 __global__ __launch_bounds__(BLOCK) void radix_hist_kernel(
     const uint64_t* __restrict__ keys, uint32_t n, int shift,
-    uint32_t* __restrict__ hist, uint32_t nb, int hash_mix) {
+    uint32_t* __restrict__ hist, uint32_t nb, int hash_mix,
+    int in_stride = 1) {
   constexpr int ND = 1 << NBITS;
   extern __shared__ char smem_raw[];
   uint32_t* counters = reinterpret_cast<uint32_t*>(smem_raw);  // [NW][ND]
@@ -127,7 +128,7 @@ __global__ __launch_bounds__(BLOCK) void radix_hist_kernel(
   for (int i = 0; i < ITEMS; ++i) {
     uint64_t e = chunk + (uint64_t)i * kWave + lane;
     bool valid = e < n;
-    uint64_t k = valid ? keys[e] : 0;
+    uint64_t k = valid ? keys[e * in_stride] : 0;
     if (hash_mix) k = hash_mix64(k);
     uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
     uint64_t vm = __ballot(valid);
@@ -438,7 +439,8 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
     uint32_t n, int shift, uint64_t* __restrict__ desc /* [nb][ND] */,
     uint32_t* __restrict__ ticket,
     const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst,
-    int stage = 3, uint64_t* __restrict__ timing = nullptr) {
+    int stage = 3, uint64_t* __restrict__ timing = nullptr,
+    const uint32_t* __restrict__ hist_pref = nullptr) {
   constexpr int ND = 1 << PBITS;
   extern __shared__ char smem_raw[];
   constexpr int TILE_T = BS * IT;
@@ -457,7 +459,8 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
   const int tid = threadIdx.x;
   const int lane = tid & (kWave - 1);
   const int wave = tid >> 6;
-  if (tid == 0) *vb_sh = atomicAdd(ticket, 1);
+  if (tid == 0)
+    *vb_sh = hist_pref ? blockIdx.x : atomicAdd(ticket, 1);
   for (int d = tid; d < NWT * ND; d += BS) counters[d] = 0;
   __syncthreads();
   const uint32_t b = *vb_sh;  // execution-ordered virtual block id
@@ -527,7 +530,11 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
   // LDS exchange below (the walk is wait-dominated: it costs ~30% of the
   // pass when run serially here — profiles/r01)
   uint64_t my_total = 0;
-  if (tid < ND) {
+  if (hist_pref) {
+    // no-lookback mode: cross-block prefixes were precomputed by the
+    // hist+scan pre-pass (kills the ~10us/block lookback wait)
+    if (tid < ND) pref[tid] = hist_pref[(uint64_t)b * ND + tid];
+  } else if (tid < ND) {
     my_total = start[tid];
     uint64_t* slot = &desc[(uint64_t)b * ND + tid];
     if (stage == 0 || b == 0) {  // stage 0: ablation, WRONG results
@@ -557,7 +564,7 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
       }
     }
     uint64_t t2 = timing ? __builtin_amdgcn_s_memrealtime() : 0;
-    if (stage != 0 && b != 0 && tid < ND)
+    if (!hist_pref && stage != 0 && b != 0 && tid < ND)
       lookback_walk<ND>(desc, b, tid, my_total, pref);
     if (stage < 2 && !timing) return;  // ablation: rank/publish/lookback
     __syncthreads();
@@ -582,7 +589,8 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
     }
     return;
   }
-  if (b != 0 && tid < ND) lookback_walk<ND>(desc, b, tid, my_total, pref);
+  if (!hist_pref && b != 0 && tid < ND)
+    lookback_walk<ND>(desc, b, tid, my_total, pref);
 
   // SoA path: key exchange + write-out, then val exchange + write-out
 #pragma unroll
@@ -785,11 +793,12 @@ size_t radix_hist_bytes(uint32_t n, int nbits) {
 
 template <int NBITS>
 static void hist_launch(const uint64_t* keys, uint32_t n, int shift,
-                        uint32_t* hist, hipStream_t s, int hash_mix = 0) {
+                        uint32_t* hist, hipStream_t s, int hash_mix = 0,
+                        int in_stride = 1) {
   uint32_t nb = num_tiles(n);
   size_t lds = (size_t)NW * (1 << NBITS) * 4;
   hipLaunchKernelGGL(radix_hist_kernel<NBITS>, dim3(nb), dim3(BLOCK), lds, s,
-                     keys, n, shift, hist, nb, hash_mix);
+                     keys, n, shift, hist, nb, hash_mix, in_stride);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -915,6 +924,11 @@ void set_aos_tile(int t) { g_aos_tile = t; }
 // compact dummy window (isolates scattered-store cost), 3 = normal
 static int g_pass_stage = 3;
 void set_pass_stage(int s) { g_pass_stage = s; }
+// sort cross-block-prefix mode: 0 = decoupled lookback (one kernel per
+// pass; measured fastest: 6.2 vs 7.0 ms for the hist+scan pre-pass at
+// 64M — the pre-pass re-read outweighs the lookback wait), 1 = hist+scan
+static int g_sort_mode = 0;
+void set_sort_mode(int m) { g_sort_mode = m; }
 // optional phase-timing accumulator: u64[4] = {phaseA, exch, lookback, writeout}
 static uint64_t* g_timing_buf = nullptr;
 void set_timing_buf(uintptr_t p) { g_timing_buf = reinterpret_cast<uint64_t*>(p); }
@@ -952,13 +966,15 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
   uint64_t* val_dst = key_dst + PD;
   uint32_t* ticket = reinterpret_cast<uint32_t*>(val_dst + PD);
   uint64_t* desc = reinterpret_cast<uint64_t*>(ticket + 4);
-  HIP_CHECK(hipMemsetAsync(totals, 0, (size_t)passes * PD * 4, s));
-  uint32_t hist_grid = nb < 1024 ? (nb ? nb : 1) : 1024;
-  hipLaunchKernelGGL((onesweep_hist_all_kernel<10, PBITS>), dim3(hist_grid),
-                     dim3(BLOCK), 0, s,
-                     reinterpret_cast<const uint64_t*>(keys), n, start_bit,
-                     passes, totals, aos ? 2 : 1);
-  HIP_CHECK(hipGetLastError());
+  if (!(g_sort_mode == 1 && aos && g_aos_tile == 4096)) {
+    HIP_CHECK(hipMemsetAsync(totals, 0, (size_t)passes * PD * 4, s));
+    uint32_t hist_grid = nb < 1024 ? (nb ? nb : 1) : 1024;
+    hipLaunchKernelGGL((onesweep_hist_all_kernel<10, PBITS>), dim3(hist_grid),
+                       dim3(BLOCK), 0, s,
+                       reinterpret_cast<const uint64_t*>(keys), n, start_bit,
+                       passes, totals, aos ? 2 : 1);
+    HIP_CHECK(hipGetLastError());
+  }
   size_t lds_soa = (size_t)BLOCK * OS_ITEMS * 8 + (size_t)NW * PD * 4 +
                    PD * 4 * 2 + BLOCK * 4 + 16 + BLOCK * OS_ITEMS;
   size_t lds_aos = (size_t)aos_tile * 16 + (size_t)(512 / kWave) * PD * 4 +
@@ -981,35 +997,48 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
                                 160 * 1024 - 1024);
     attr_set = true;
   }
+  // hist mode: reuse the desc area as [hist u32[nb*PD] | scan partials]
+  const bool use_hist = g_sort_mode == 1 && aos && aos_tile == 4096;
+  uint32_t* hist32 = reinterpret_cast<uint32_t*>(desc);
+  uint32_t* hist_scan_ws = hist32 + (size_t)nb * PD;
   uintptr_t src_k = keys, src_v = vals, dst_k = tmp_keys, dst_v = tmp_vals;
   int cur = 0;
   for (int p = 0; p < passes; ++p) {
+    int sb = start_bit + p * PBITS;
+    if (sb > 64 - PBITS) sb = 64 - PBITS;  // same clamp as hist_all
+    if (use_hist) {
+      hist_launch<PBITS>(reinterpret_cast<const uint64_t*>(src_k), n, sb,
+                         hist32, s, 0, 2);
+      scan_launch<PBITS>(hist32, nb, hist_scan_ws, totals + (size_t)p * PD,
+                         s);
+    }
     hipLaunchKernelGGL((onesweep_digit_bases_kernel<PD>), dim3(1),
                        dim3(BLOCK), 0, s, totals, p, (uint64_t)dst_k,
                        (uint64_t)dst_v, key_dst, val_dst, aos ? 16 : 8);
     HIP_CHECK(hipGetLastError());
-    HIP_CHECK(hipMemsetAsync(ticket, 0, 16, s));
-    HIP_CHECK(hipMemsetAsync(desc, 0, (size_t)nb * PD * 8, s));
-    int sb = start_bit + p * PBITS;
-    if (sb > 64 - PBITS) sb = 64 - PBITS;  // same clamp as hist_all
+    if (!use_hist) {
+      HIP_CHECK(hipMemsetAsync(ticket, 0, 16, s));
+      HIP_CHECK(hipMemsetAsync(desc, 0, (size_t)nb * PD * 8, s));
+    }
+    const uint32_t* pass_pref = use_hist ? hist32 : nullptr;
     if (aos && aos_tile == 8192) {
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 16, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          sb, desc, ticket, key_dst, val_dst, g_pass_stage,
-                         g_timing_buf);
+                         g_timing_buf, pass_pref);
     } else if (aos && aos_tile == 2048) {
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 4, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          sb, desc, ticket, key_dst, val_dst, g_pass_stage,
-                         g_timing_buf);
+                         g_timing_buf, pass_pref);
     } else if (aos) {
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 8, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          sb, desc, ticket, key_dst, val_dst, g_pass_stage,
-                         g_timing_buf);
+                         g_timing_buf, pass_pref);
     } else if (vals) {
       hipLaunchKernelGGL(
           (onesweep_pass_kernel<true, OS_ITEMS, false, BLOCK, PBITS>),
