@@ -173,6 +173,12 @@ def main():
                 "parallelism": f"shuffle{world}",
                 "wall_clock_s_per_job": round(elapsed / args.steps, 3),
                 "remote_gb_per_step": round(remote_gb / max(1, args.steps), 2),
+                # BASELINE metric names shuffle-read GB/s explicitly
+                "shuffle_read_gb_per_s": round(
+                    sum(getattr(r, "fetch_s", 0) and
+                        (n_rec * 16 / getattr(r, "fetch_s")) / 1e9
+                        for r in results) / max(1, len(results)), 2)
+                if args.workload == "terasort" else None,
             },
         }))
     eng.shutdown()

@@ -124,11 +124,20 @@ __global__ __launch_bounds__(BLOCK) void radix_hist_kernel(
   __syncthreads();
 
   const uint64_t chunk = tile_start + (uint64_t)wave * (ITEMS * kWave);
-#pragma unroll 1
+  // loads issued in their own unrolled loop so they are all in flight
+  // before the serial LDS counter chain (same fix as the sort pass:
+  // phase A 11.4 -> 8.2 us/block)
+  uint64_t key_reg[ITEMS];
+#pragma unroll
+  for (int i = 0; i < ITEMS; ++i) {
+    uint64_t e = chunk + (uint64_t)i * kWave + lane;
+    key_reg[i] = e < n ? keys[e * in_stride] : 0;
+  }
+#pragma unroll
   for (int i = 0; i < ITEMS; ++i) {
     uint64_t e = chunk + (uint64_t)i * kWave + lane;
     bool valid = e < n;
-    uint64_t k = valid ? keys[e * in_stride] : 0;
+    uint64_t k = key_reg[i];
     if (hash_mix) k = hash_mix64(k);
     uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
     uint64_t vm = __ballot(valid);
@@ -263,17 +272,22 @@ __global__ __launch_bounds__(BLOCK) void radix_scatter_kernel(
   for (int d = tid; d < ND; d += BLOCK) pref[d] = hist[(uint64_t)b * ND + d];
   __syncthreads();
 
-  // phase A: per-element (digit, rank within wave&digit), stable
+  // phase A: per-element (digit, rank within wave&digit), stable;
+  // loads split from the serial LDS counter chain (see sort pass)
   uint64_t key_reg[ITEMS];
   uint32_t rank_reg[ITEMS];
   uint32_t dig_reg[ITEMS];
   const uint64_t chunk = tile_start + (uint64_t)wave * (ITEMS * kWave);
-#pragma unroll 1
+#pragma unroll
+  for (int i = 0; i < ITEMS; ++i) {
+    uint64_t e = chunk + (uint64_t)i * kWave + lane;
+    key_reg[i] = e < n ? keys[e] : 0;
+  }
+#pragma unroll
   for (int i = 0; i < ITEMS; ++i) {
     uint64_t e = chunk + (uint64_t)i * kWave + lane;
     bool valid = e < n;
-    uint64_t k = valid ? keys[e] : 0;
-    key_reg[i] = k;
+    uint64_t k = key_reg[i];
     uint32_t d = (uint32_t)((hash_mix ? hash_mix64(k) : k) >> shift) & (ND - 1);
     dig_reg[i] = d;
     uint64_t vm = __ballot(valid);
