@@ -141,3 +141,28 @@ def sort_pairs(keys: torch.Tensor, vals: Optional[torch.Tensor] = None,
     if res == 0:
         return keys, vals
     return tmp_k, tmp_v
+
+
+class AosSorter:
+    """Persistent-buffer AoS sorter: workspace and ping-pong buffer are
+    allocated once, so a sort issues ONLY kernel launches and stream
+    memsets — hipGraph-capturable (see tests/test_gpu_graphs.py)."""
+
+    def __init__(self, n: int, device="cuda", start_bit: int = 0,
+                 end_bit: int = 64):
+        m = load()
+        self.n = n
+        self.start_bit = start_bit
+        self.end_bit = min(start_bit + ((end_bit - start_bit + 7) // 8) * 8, 64)
+        passes = (self.end_bit - start_bit) // 8
+        self.tmp = torch.empty(2 * n, dtype=torch.int64, device=device)
+        self.ws = torch.empty(m.onesweep_workspace_bytes(n, passes),
+                              dtype=torch.uint8, device=device)
+        self._m = m
+
+    def sort_(self, pairs: torch.Tensor) -> torch.Tensor:
+        assert pairs.numel() == 2 * self.n
+        res = self._m.onesweep_sort_aos_u64(
+            pairs.data_ptr(), self.tmp.data_ptr(), self.n, self.start_bit,
+            self.end_bit, self.ws.data_ptr(), _stream())
+        return pairs if res == 0 else self.tmp
