@@ -111,7 +111,8 @@ def stage_ab():
     keys = i * 0x9E3779B97F4A7C15
     keys ^= keys >> 31
     pairs = torch.stack([keys, keys], dim=1).reshape(-1).contiguous()
-    for stage, label in ((0, "rank only, no lookback walk"),
+    for stage, label in ((9, "no LDS counter chain (timing only)"),
+                         (0, "rank only, no lookback walk"),
                          (1, "rank+lookback only"), (2, "+exch+dummy-store"),
                          (3, "full")):
         m.set_pass_stage(stage)

@@ -518,9 +518,13 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
       uint64_t match = match_lanes<PBITS>(d, vm);
       uint64_t lt = (1ull << lane) - 1;
       uint32_t rank_in_iter = (uint32_t)__popcll(match & lt);
-      uint32_t c = my[d];
-      r = c + rank_in_iter;
-      if (rank_in_iter == 0) my[d] = c + (uint32_t)__popcll(match);
+      if (stage == 9) {  // ablation: no LDS counter chain (WRONG results)
+        r = rank_in_iter;
+      } else {
+        uint32_t c = my[d];
+        r = c + rank_in_iter;
+        if (rank_in_iter == 0) my[d] = c + (uint32_t)__popcll(match);
+      }
     }
     digrank[i] = (d << 16) | r;
   }
