@@ -148,6 +148,18 @@ class GpuDataPlane:
         self.hs.wait_event(ev)
         return dst
 
+    def read_device_into(self, key: int, addr: int, length: int,
+                         dst_ptr: int) -> None:
+        """One-sided read landing at a caller-chosen device address —
+        the arena fast path (no per-fetch allocation, no reduce-side
+        concat)."""
+        exec_id, _ = split_key(key)
+        self.hs.set_device(self.device)
+        base = self._resolve_base(key)
+        ev = self.hs.read_batch(exec_id % 64, [dst_ptr], [base + addr],
+                                [length])
+        self.hs.wait_event(ev)
+
     def stop(self) -> None:
         for key, base in list(self._peer_bases.items()):
             exec_id, _ = split_key(key)

@@ -81,13 +81,14 @@ class SortMergeJoin:
         lo, hi = eng.rank * self.ppe, (eng.rank + 1) * self.ppe - 1
         reader = eng.manager.get_reader(handle, lo, hi)
         parts = reader.collect_partitions()
-        return handle, parts, reader.metrics
+        return handle, parts, reader.metrics, \
+            getattr(reader.fetcher, "arena", None)
 
     def run_step(self) -> JoinResult:
         eng = self.engine
         t0 = time.perf_counter()
-        ha, parts_a, ma = self._shuffle(self.a_keys, self.a_vals)
-        hb, parts_b, mb = self._shuffle(self.b_keys, self.b_vals)
+        ha, parts_a, ma, arena_a = self._shuffle(self.a_keys, self.a_vals)
+        hb, parts_b, mb, arena_b = self._shuffle(self.b_keys, self.b_vals)
         matches = 0
         checksum = 0
         if self.device == "cuda":
@@ -99,8 +100,10 @@ class SortMergeJoin:
             # replaces ppe per-partition sorts (same trick as TeraSort)
             wbits = (self.engine.world_size - 1).bit_length()
             sort_bits = self.key_bits - wbits
-            a_chunks = [c for p in parts_a for c in parts_a[p]]
-            b_chunks = [c for p in parts_b for c in parts_b[p]]
+            a_chunks = ([arena_a.view(torch.int64)] if arena_a is not None
+                        else [c for p in parts_a for c in parts_a[p]])
+            b_chunks = ([arena_b.view(torch.int64)] if arena_b is not None
+                        else [c for p in parts_b for c in parts_b[p]])
             ak, av = self._sort_side(a_chunks, sort_bits)
             bk, bv = self._sort_side(b_chunks, sort_bits)
             if ak is not None and bk is not None:

@@ -119,7 +119,18 @@ class TeraSort:
             tf = time.perf_counter()
             parts = reader.collect_partitions()
             ts_ = time.perf_counter()
-            outs.append(self._reduce(parts, chunk_shared_bits))
+            arena = getattr(reader.fetcher, "arena", None)
+            if self.device == "cuda" and arena is not None:
+                # fetches landed pre-placed in one device buffer: sort it
+                # directly (no concat pass)
+                import torch
+                from ..ops.radix import sort_pairs_aos
+                pairs = sort_pairs_aos(arena.view(torch.int64), 0,
+                                       64 - chunk_shared_bits)
+                torch.cuda.synchronize()
+                outs.append((pairs[0::2], pairs[1::2]))
+            else:
+                outs.append(self._reduce(parts, chunk_shared_bits))
             t_fetch_total += ts_ - tf
             t_sort_total += time.perf_counter() - ts_
             remote += reader.metrics.remote_bytes_read

@@ -98,6 +98,8 @@ class _FakeManager:
     def is_remote_host(self, exec_id):
         return False
 
+    gpu = None  # no device plane -> no arena mode
+
     def remote_read(self, key, addr, length):
         import time
         if key == 99:  # hop-2 table read: fabricate location entries
