@@ -216,9 +216,10 @@ class ShuffleManager:
         return ShuffleWriter(self, handle, map_id)
 
     def get_reader(self, handle: ShuffleHandle, start_partition: int,
-                   end_partition: int):
+                   end_partition: int, arena=None):
         from .reader import ShuffleReader
-        return ShuffleReader(self, handle, start_partition, end_partition)
+        return ShuffleReader(self, handle, start_partition, end_partition,
+                             arena=arena)
 
     def unregister_shuffle(self, shuffle_id: int, notify_driver: bool = True) -> None:
         # release served blocks (liveness discipline: blocks stay alive until

@@ -74,7 +74,9 @@ def radix_partition(keys: torch.Tensor, vals: Optional[torch.Tensor],
 
 def sort_pairs_aos(pairs: torch.Tensor, start_bit: int = 0,
                    end_bit: int = 64,
-                   digit_bits: Optional[int] = None) -> torch.Tensor:
+                   digit_bits: Optional[int] = None,
+                   tmp: Optional[torch.Tensor] = None,
+                   ws: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Sort interleaved (key u64, val u64) 16-byte records by key bits
     [start_bit, end_bit). ``pairs`` is an int64 tensor of 2n elements.
     One dwordx4 load + one dwordx4 store per record per pass, and digit
@@ -91,9 +93,16 @@ def sort_pairs_aos(pairs: torch.Tensor, start_bit: int = 0,
         # write-only burst model overestimates; pass floor is elsewhere
         digit_bits = 8
     passes = -(-bits // digit_bits)
-    tmp = torch.empty_like(pairs)
-    ws = torch.empty(m.onesweep_workspace_bytes(n, passes),
-                     dtype=torch.uint8, device=pairs.device)
+    if tmp is None:
+        tmp = torch.empty_like(pairs)
+    else:
+        assert tmp.numel() >= pairs.numel(), "tmp too small"
+        tmp = tmp[:pairs.numel()]
+    need_ws = m.onesweep_workspace_bytes(n, passes)
+    if ws is None:
+        ws = torch.empty(need_ws, dtype=torch.uint8, device=pairs.device)
+    else:
+        assert ws.numel() >= need_ws, "ws too small"
     fn = (m.onesweep_sort_aos7_u64 if digit_bits == 7
           else m.onesweep_sort_aos_u64)
     res = fn(pairs.data_ptr(), tmp.data_ptr(), n, start_bit,

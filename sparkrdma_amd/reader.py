@@ -118,7 +118,8 @@ class FetcherIterator:
 
     def __init__(self, manager: ShuffleManager, handle: ShuffleHandle,
                  start_partition: int, end_partition: int,
-                 num_workers: int = 8, seed: Optional[int] = None):
+                 num_workers: int = 8, seed: Optional[int] = None,
+                 arena: Optional[object] = None):
         self.manager = manager
         self.handle = handle
         self.start_partition = start_partition
@@ -140,6 +141,7 @@ class FetcherIterator:
                                         initializer=self._pin_worker)
         self._rng = random.Random(seed)
         self._failed: Optional[BaseException] = None
+        self._arena_hint = arena   # caller-provided reusable device buffer
         self._start()
 
     # ------------------------------------------------------------------
@@ -180,9 +182,13 @@ class FetcherIterator:
             for f in fetches:       # block order, not arrival order
                 self._arena_off[id(f)] = off
                 off += f.length
-            self.arena = torch.empty(
-                total, dtype=torch.uint8,
-                device=f"cuda:{mgr.gpu.device}")
+            hint = self._arena_hint
+            if hint is not None and hint.numel() >= total:
+                self.arena = hint[:total]   # reuse caller's buffer
+            else:
+                self.arena = torch.empty(
+                    total, dtype=torch.uint8,
+                    device=f"cuda:{mgr.gpu.device}")
         # randomize to spread load over source executors (reference :74-79)
         self._rng.shuffle(fetches)
         with self._lock:
@@ -295,13 +301,14 @@ class ShuffleReader:
     """Public reader: iterate raw blocks or aggregate per partition."""
 
     def __init__(self, manager: ShuffleManager, handle: ShuffleHandle,
-                 start_partition: int, end_partition: int):
+                 start_partition: int, end_partition: int, arena=None):
         self.manager = manager
         self.handle = handle
         self.start_partition = start_partition
         self.end_partition = end_partition
         self.fetcher = FetcherIterator(manager, handle,
-                                       start_partition, end_partition)
+                                       start_partition, end_partition,
+                                       arena=arena)
 
     def __iter__(self):
         return iter(self.fetcher)

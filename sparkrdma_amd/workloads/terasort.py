@@ -109,7 +109,9 @@ class TeraSort:
         # ppe is pow2, so per = ppe/H is exact
         per = self.ppe // H
         spans = [(lo + h * per, lo + (h + 1) * per - 1) for h in range(H)]
-        readers = [eng.manager.get_reader(handle, a, b) for a, b in spans]
+        arenas = self._arenas(H)
+        readers = [eng.manager.get_reader(handle, a, b, arena=ar)
+                   for (a, b), ar in zip(spans, arenas)]
         t_fetch_total = 0.0
         t_sort_total = 0.0
         outs = []
@@ -122,11 +124,14 @@ class TeraSort:
             arena = getattr(reader.fetcher, "arena", None)
             if self.device == "cuda" and arena is not None:
                 # fetches landed pre-placed in one device buffer: sort it
-                # directly (no concat pass)
+                # directly (no concat pass); tmp/ws persist across steps
+                # so a 40 GB step performs no large allocations at all
                 import torch
                 from ..ops.radix import sort_pairs_aos
                 pairs = sort_pairs_aos(arena.view(torch.int64), 0,
-                                       64 - chunk_shared_bits)
+                                       64 - chunk_shared_bits,
+                                       tmp=self._sort_tmp(arena.numel() // 8),
+                                       ws=self._sort_ws())
                 torch.cuda.synchronize()
                 outs.append((pairs[0::2], pairs[1::2]))
             else:
@@ -144,6 +149,38 @@ class TeraSort:
             bytes_sorted=self.n * self.RECORD_BYTES,
             write_s=t_write - t0, fetch_s=t_fetch_total,
             sort_s=t_sort_total, remote_bytes=remote)
+
+    def _arenas(self, H: int):
+        """Reusable per-chunk fetch arenas (slack for partition skew)."""
+        if self.device != "cuda":
+            return [None] * H
+        import torch
+        cap = int(self.n * self.RECORD_BYTES // H * 1.25) + (64 << 10)
+        cur = getattr(self, "_arena_cache", None)
+        if cur is None or len(cur) != H or cur[0].numel() < cap:
+            self._arena_cache = [torch.empty(cap, dtype=torch.uint8,
+                                             device="cuda")
+                                 for _ in range(H)]
+        return self._arena_cache
+
+    def _sort_tmp(self, n_i64: int):
+        import torch
+        cur = getattr(self, "_tmp_cache", None)
+        if cur is None or cur.numel() < n_i64:
+            self._tmp_cache = torch.empty(int(n_i64 * 1.05) + 1024,
+                                          dtype=torch.int64, device="cuda")
+        return self._tmp_cache
+
+    def _sort_ws(self):
+        from ..ops import load
+        import torch
+        cur = getattr(self, "_ws_cache", None)
+        need = load().onesweep_workspace_bytes(
+            int(self.n * 1.25) + 4096, 8)
+        if cur is None or cur.numel() < need:
+            self._ws_cache = torch.empty(need, dtype=torch.uint8,
+                                         device="cuda")
+        return self._ws_cache
 
     def _reduce(self, parts: dict, shared_bits: Optional[int] = None):
         """One batched sort per fetched chunk: all partitions of a chunk
