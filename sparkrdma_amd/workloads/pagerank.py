@@ -104,7 +104,16 @@ class PageRank:
         w.stop(True, partitioner=self.part)
         eng.barrier()
         lo, hi = eng.rank * self.ppe, (eng.rank + 1) * self.ppe - 1
-        reader = eng.manager.get_reader(handle, lo, hi)
+        arena = None
+        if self.device == "cuda":
+            import torch
+            cap = int(self.n_edges * 16 * 1.25) + (64 << 10)
+            arena = getattr(self, "_arena_cache", None)
+            if arena is None or arena.numel() < cap:
+                self._arena_cache = torch.empty(cap, dtype=torch.uint8,
+                                                device="cuda")
+                arena = self._arena_cache
+        reader = eng.manager.get_reader(handle, lo, hi, arena=arena)
         span = self.own_hi - self.own_lo
         if self.device == "cuda":
             import torch

@@ -79,8 +79,24 @@ class SortMergeJoin:
         w.stop(True, partitioner=self.part)
         eng.barrier()
         lo, hi = eng.rank * self.ppe, (eng.rank + 1) * self.ppe - 1
-        reader = eng.manager.get_reader(handle, lo, hi)
+        arena_hint = None
+        if self.device == "cuda":
+            import torch
+            cap = int(self.n * 16 * 1.25) + (64 << 10)
+            self._arena_pool = getattr(self, "_arena_pool", [])
+            for t in self._arena_pool:
+                if not getattr(t, "_in_use", False) and t.numel() >= cap:
+                    arena_hint = t
+                    break
+            if arena_hint is None:
+                arena_hint = torch.empty(cap, dtype=torch.uint8,
+                                         device="cuda")
+                self._arena_pool.append(arena_hint)
+            arena_hint._in_use = True
+        reader = eng.manager.get_reader(handle, lo, hi, arena=arena_hint)
         parts = reader.collect_partitions()
+        # arena stays marked in-use until run_step() ends — table A's data
+        # must survive table B's shuffle
         return handle, parts, reader.metrics, \
             getattr(reader.fetcher, "arena", None)
 
@@ -128,6 +144,8 @@ class SortMergeJoin:
                 matches += int((hi - lo).sum())
         eng.unregister_shuffle(ha)
         eng.unregister_shuffle(hb)
+        for t in getattr(self, "_arena_pool", []):
+            t._in_use = False
         dt = time.perf_counter() - t0
         if self.validate:
             self._validate_counts(matches)
