@@ -90,6 +90,9 @@ class PageRank:
         return PageRankResult(dt, self.iters, self.n_edges, shuffle_bytes)
 
     def _iteration(self) -> int:
+        import os
+        dbg = os.environ.get("BENCH_DEBUG")
+        t0 = time.perf_counter()
         eng = self.engine
         handle = eng.register_shuffle(eng.world_size, self.R)
         w = eng.manager.get_writer(handle, eng.rank)
@@ -125,6 +128,10 @@ class PageRank:
                 sums.index_add_(0, idx, t[1::2].contiguous().view(torch.float64))
             self.ranks = (1.0 - DAMPING) / self.V + DAMPING * sums
             torch.cuda.synchronize()
+            if dbg:
+                import sys
+                print(f"[pr-iter] write+fetch+agg={time.perf_counter()-t0:.3f}s",
+                      file=sys.stderr)
         else:
             sums = np.zeros(span, dtype=np.float64)
             for ref, data in reader:
