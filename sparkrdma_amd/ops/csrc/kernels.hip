@@ -88,6 +88,17 @@ __device__ __forceinline__ uint64_t hash_mix64(uint64_t k) {
   return k;
 }
 
+// Match from PRE-COMPUTED per-bit ballots (register votes) — lets rank
+// logic reuse one iteration's ballots against another iteration's digits.
+template <int PBITS>
+__device__ __forceinline__ uint64_t match_votes(const uint64_t* v,
+                                                uint64_t vm, uint32_t d) {
+  uint64_t m = vm;
+#pragma unroll
+  for (int b = 0; b < PBITS; ++b) m &= ((d >> b) & 1) ? v[b] : ~v[b];
+  return m;
+}
+
 // 64-lane multi-split: lanes with equal `digit` (among `validmask` lanes).
 template <int NBITS>
 __device__ __forceinline__ uint64_t match_lanes(uint32_t digit,
@@ -506,27 +517,48 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
       key_reg[i] = valid ? keys[e] : 0;
     }
   }
+  // rank in iteration PAIRS: both iterations' per-bit ballots are held
+  // in registers, so the pair's ranks and the single LDS counter update
+  // per digit need only ONE serial LDS round trip per two iterations —
+  // the serial chain was 13% of a pass (stage-9 ablation).
+  static_assert(IT % 2 == 0, "pair ranking needs even IT");
+  const uint64_t lt = (1ull << lane) - 1;
 #pragma unroll
-  for (int i = 0; i < IT; ++i) {
-    uint64_t e = chunk + (uint64_t)i * kWave + lane;
-    bool valid = e < n;
-    uint64_t k = key_reg[i];
-    uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
-    uint64_t vm = __ballot(valid);
-    uint32_t r = 0;
-    if (valid) {
-      uint64_t match = match_lanes<PBITS>(d, vm);
-      uint64_t lt = (1ull << lane) - 1;
-      uint32_t rank_in_iter = (uint32_t)__popcll(match & lt);
-      if (stage == 9) {  // ablation: no LDS counter chain (WRONG results)
-        r = rank_in_iter;
-      } else {
-        uint32_t c = my[d];
-        r = c + rank_in_iter;
-        if (rank_in_iter == 0) my[d] = c + (uint32_t)__popcll(match);
+  for (int c = 0; c < IT / 2; ++c) {
+    const int i0 = 2 * c, i1 = 2 * c + 1;
+    uint64_t e0 = chunk + (uint64_t)i0 * kWave + lane;
+    uint64_t e1 = chunk + (uint64_t)i1 * kWave + lane;
+    bool val0 = e0 < n, val1 = e1 < n;
+    uint32_t d0 = (uint32_t)(key_reg[i0] >> shift) & (ND - 1);
+    uint32_t d1 = (uint32_t)(key_reg[i1] >> shift) & (ND - 1);
+    uint64_t vm0 = __ballot(val0), vm1 = __ballot(val1);
+    uint64_t v0[PBITS], v1[PBITS];
+#pragma unroll
+    for (int b = 0; b < PBITS; ++b) v0[b] = __ballot((d0 >> b) & 1);
+#pragma unroll
+    for (int b = 0; b < PBITS; ++b) v1[b] = __ballot((d1 >> b) & 1);
+    // per-chunk bases read BEFORE this chunk's single update round
+    uint32_t base0 = (val0 && stage != 9) ? my[d0] : 0;
+    uint32_t base1 = (val1 && stage != 9) ? my[d1] : 0;
+    uint64_t m0 = match_votes<PBITS>(v0, vm0, d0);
+    uint64_t m1 = match_votes<PBITS>(v1, vm1, d1);
+    uint64_t cross1 = match_votes<PBITS>(v0, vm0, d1);  // i0 elems == d1
+    uint32_t r0 = base0 + (uint32_t)__popcll(m0 & lt);
+    uint32_t r1 = base1 + (uint32_t)__popcll(cross1)
+                  + (uint32_t)__popcll(m1 & lt);
+    digrank[i0] = (d0 << 16) | (val0 ? r0 : 0);
+    digrank[i1] = (d1 << 16) | (val1 ? r1 : 0);
+    if (stage != 9) {
+      // one LDS update per digit per PAIR: the i0 leader of a digit adds
+      // both iterations' counts; digits present only in i1 are added by
+      // their i1 leader
+      if (val0 && (m0 & lt) == 0) {
+        uint64_t cross0 = match_votes<PBITS>(v1, vm1, d0);
+        my[d0] = base0 + (uint32_t)__popcll(m0) + (uint32_t)__popcll(cross0);
       }
+      if (val1 && (m1 & lt) == 0 && cross1 == 0)
+        my[d1] = base1 + (uint32_t)__popcll(m1);
     }
-    digrank[i] = (d << 16) | r;
   }
   __syncthreads();
 
