@@ -167,3 +167,35 @@ def test_barrier(cluster):
     for t in threads:
         t.join(timeout=10)
     assert sorted(results) == [0, 1, 2]
+
+
+def test_concurrent_shuffles(cluster):
+    """Two shuffles in flight: registrations, writes and reads interleave
+    without cross-talk (per-shuffle tables and block liveness)."""
+    _, managers = cluster
+    part = HashPartitioner(4)
+    h1 = managers[0].register_shuffle(num_maps=3, num_partitions=4)
+    h2 = managers[0].register_shuffle(num_maps=3, num_partitions=4)
+    rng = np.random.default_rng(21)
+    data = {h.shuffle_id: [] for h in (h1, h2)}
+    for mid, mgr in enumerate(managers):
+        for h, base in ((h1, 0), (h2, 1 << 32)):
+            keys = rng.integers(base, base + (1 << 31), 3000, dtype=np.uint64)
+            data[h.shuffle_id].append(keys)
+            w = mgr.get_writer(h, mid)
+            w.write_batch(keys)
+            w.stop(True, partitioner=part)
+    for h in (h1, h2):
+        reader = managers[1].get_reader(h, 0, 3)
+        got = []
+        for ref, chunk in reader:
+            k, _ = unpack_partition_segment(chunk, 0)
+            got.append(np.array(k))
+        want = np.sort(np.concatenate(data[h.shuffle_id]))
+        assert np.array_equal(np.sort(np.concatenate(got)), want)
+    managers[0].unregister_shuffle(h1.shuffle_id)
+    # h2 must still be readable after h1 is gone
+    reader = managers[2].get_reader(h2, 0, 3)
+    total = sum(len(unpack_partition_segment(c, 0)[0]) for _, c in reader)
+    assert total == 9000
+    managers[0].unregister_shuffle(h2.shuffle_id)
