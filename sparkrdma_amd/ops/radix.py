@@ -72,6 +72,42 @@ def radix_partition(keys: torch.Tensor, vals: Optional[torch.Tensor],
     return totals[:1 << nbits], keys_out, vals_out
 
 
+def partition_aos(keys: torch.Tensor, vals: torch.Tensor, nbits: int,
+                  shift: Optional[int] = None,
+                  hash_mix: bool = False):
+    """Partition SoA (keys, vals) into ONE interleaved AoS buffer with
+    buckets laid out contiguously — the stage-mode (RCCL alltoallv) send
+    buffer in a single scatter pass.
+
+    Returns (counts int32[2^nbits] device, pairs int64[2n] device)."""
+    m = load()
+    n = keys.numel()
+    if shift is None:
+        shift = 64 - nbits
+    nbits_eff = max(nbits, 4)
+    nd = 1 << nbits_eff
+    dev = keys.device
+    hist = torch.empty(m.radix_hist_bytes(n, nbits_eff) // 4,
+                       dtype=torch.int32, device=dev)
+    scan_ws = torch.empty(m.radix_scan_ws_bytes(n, nbits_eff) // 4,
+                          dtype=torch.int32, device=dev)
+    totals = torch.empty(nd, dtype=torch.int32, device=dev)
+    s = _stream()
+    m.radix_hist(keys.data_ptr(), n, shift, nbits_eff, hist.data_ptr(), s,
+                 int(hash_mix))
+    m.radix_scan(hist.data_ptr(), n, nbits_eff, totals.data_ptr(),
+                 scan_ws.data_ptr(), s)
+    pairs = torch.empty(2 * n, dtype=torch.int64, device=dev)
+    counts64 = totals.to(torch.int64)
+    bases = torch.cumsum(counts64, 0) - counts64
+    key_dst = pairs.data_ptr() + bases * 16
+    val_dst = key_dst + 8
+    m.radix_scatter(keys.data_ptr(), vals.data_ptr(), n, shift, nbits_eff,
+                    hist.data_ptr(), key_dst.data_ptr(), val_dst.data_ptr(),
+                    s, int(hash_mix), 1)
+    return totals[:1 << nbits], pairs
+
+
 def sort_pairs_aos(pairs: torch.Tensor, start_bit: int = 0,
                    end_bit: int = 64,
                    digit_bits: Optional[int] = None,

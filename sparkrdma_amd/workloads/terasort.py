@@ -237,39 +237,44 @@ class TeraSort:
     # ------------------------------------------------------------------
 
     def _step_rccl(self) -> TeraSortResult:
-        """Stage-mode shuffle: RCCL all_to_all_single over xGMI
-        (SURVEY §7.1 'collective-form option')."""
+        """Stage-mode shuffle: ONE RCCL all_to_all_single over xGMI of
+        AoS records (SURVEY §7.1 'collective-form option') — the partition
+        pass scatters straight into the send buffer."""
         import torch
         import torch.distributed as dist
-        from ..ops.radix import radix_partition, sort_pairs
+        from ..ops.radix import partition_aos, sort_pairs_aos
         eng = self.engine
         W = eng.world_size
         t0 = time.perf_counter()
         wbits = (W - 1).bit_length()
         if W == 1:
-            k, v = sort_pairs(self.keys.clone(), self.vals.clone(), 0, 64)
+            pairs = torch.empty(2 * self.n, dtype=torch.int64, device="cuda")
+            pairs[0::2] = self.keys
+            pairs[1::2] = self.vals
+            pairs = sort_pairs_aos(pairs, 0, 64)
             torch.cuda.synchronize()
             dt = time.perf_counter() - t0
+            if self.validate:
+                ku = pairs[0::2].cpu().numpy().view(np.uint64)
+                assert np.all(ku[1:] >= ku[:-1])
             return TeraSortResult(dt, self.n, self.n * 16, 0, 0, dt, 0)
-        counts, k_part, v_part = radix_partition(
-            self.keys, self.vals, wbits, shift=64 - wbits)
-        in_splits = counts.to(torch.int64)
+        counts, send_pairs = partition_aos(self.keys, self.vals, wbits,
+                                           shift=64 - wbits)
+        in_splits = counts.to(torch.int64) * 2   # i64 elements per record: 2
         out_splits = torch.empty_like(in_splits)
         dist.all_to_all_single(out_splits, in_splits)
         in_l = in_splits.cpu().tolist()
         out_l = out_splits.cpu().tolist()
-        recv_k = torch.empty(sum(out_l), dtype=torch.int64, device="cuda")
-        recv_v = torch.empty_like(recv_k)
-        dist.all_to_all_single(recv_k, k_part, out_l, in_l)
-        dist.all_to_all_single(recv_v, v_part, out_l, in_l)
+        recv = torch.empty(sum(out_l), dtype=torch.int64, device="cuda")
+        dist.all_to_all_single(recv, send_pairs, out_l, in_l)
         t_fetch = time.perf_counter()
-        k_sorted, v_sorted = sort_pairs(recv_k, recv_v, 0, 64 - wbits)
+        pairs = sort_pairs_aos(recv, 0, 64 - wbits)
         torch.cuda.synchronize()
         t_sort = time.perf_counter()
         if self.validate:
-            ku = k_sorted.cpu().numpy().view(np.uint64)
+            ku = pairs[0::2].cpu().numpy().view(np.uint64)
             assert np.all(ku[1:] >= ku[:-1])
         dt = time.perf_counter() - t0
         return TeraSortResult(dt, self.n, self.n * 16,
                               t_fetch - t0, 0, t_sort - t_fetch,
-                              int(sum(out_l)) * 16)
+                              int(sum(out_l)) * 8)
