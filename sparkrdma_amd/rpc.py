@@ -130,11 +130,7 @@ def pack_handle(shuffle_id: int, num_maps: int, num_partitions: int,
 def unpack_handle(body: bytes) -> tuple:
     shuffle_id, num_maps, num_partitions, table_addr = struct.unpack_from("<iiiQ", body, 0)
     uri, _ = _unpack_str(body, 20)
-    return shuffle_id, num_maps, num_partitions, table_uri_fix(uri), table_addr
-
-
-def table_uri_fix(uri: str) -> str:
-    return uri
+    return shuffle_id, num_maps, num_partitions, uri, table_addr
 
 
 def pack_unregister(shuffle_id: int) -> bytes:

@@ -77,7 +77,6 @@ class PageRank:
             self.dst = dst
             self.outdeg = outdeg
             self.ranks = np.full(span, 1.0 / self.V, dtype=np.float64)
-        self._edges_u8 = None  # CPU value buffer cache
 
     # ------------------------------------------------------------------
 

@@ -106,7 +106,6 @@ class SortMergeJoin:
         ha, parts_a, ma, arena_a = self._shuffle(self.a_keys, self.a_vals)
         hb, parts_b, mb, arena_b = self._shuffle(self.b_keys, self.b_vals)
         matches = 0
-        checksum = 0
         if self.device == "cuda":
             import torch
             from ..ops.join import merge_join_sorted
