@@ -169,3 +169,10 @@ class GpuDataPlane:
                 except Exception:
                     pass
         self._peer_bases.clear()
+        for slot, sid in list(self._slab_ids.items()):
+            try:
+                self.hs.slab_free(sid)
+            except Exception:
+                pass
+        self._slab_ids.clear()
+        self._slab_bases.clear()

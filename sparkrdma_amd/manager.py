@@ -250,8 +250,6 @@ class ShuffleManager:
             self.reader_stats.print_histograms(log)
         if self._pool is not None:
             log.info("%s", self._pool.format_stats())
-        if self.gpu is not None:
-            self.gpu.stop()
         if self._conn is not None:
             try:
                 self._conn.send(rpc.MSG_BYE, b"")
@@ -262,9 +260,11 @@ class ShuffleManager:
             mm.close()
         self._driver_tables.clear()
         if self._data_server is not None:
-            self._data_server.stop()
+            self._data_server.stop()   # before freeing the slabs it serves
         if self._data_client is not None:
             self._data_client.close()
+        if self.gpu is not None:
+            self.gpu.stop()
         if self._registry is not None:
             self._registry.close()
         for seg in self._data_segments.values():
