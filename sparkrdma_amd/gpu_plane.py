@@ -16,8 +16,9 @@ READ machinery:
   owner's slab-table slot. Addresses in BlockLocations are slab OFFSETS,
   valid in any importing process.
 
-Fetch destinations are transient torch tensors (consumed by the reduce
-task); only *served* memory needs the IPC-exported slabs.
+Fetch destinations land in a caller-reusable device *arena* (each
+coalesced fetch pre-assigned an offset — reader.py), or in transient
+torch tensors; only *served* memory needs the IPC-exported slabs.
 """
 
 from __future__ import annotations

@@ -88,3 +88,24 @@ def test_partitioners_total_and_in_range(R, keys):
     rp = RangePartitioner.uniform(R)
     order = np.argsort(k)
     assert np.all(np.diff(rp.partition_ids(k[order])) >= 0)
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.integers(0, 2**64 - 1), st.integers(-2**31, 2**31 - 1),
+       st.integers(0, 2**32 - 1))
+def test_block_location_pack_roundtrip(addr, length, key):
+    from sparkrdma_amd.map_output import BlockLocation
+    loc = BlockLocation(addr, length, key)
+    assert BlockLocation.unpack(loc.pack()) == loc
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.lists(st.tuples(st.integers(0, 2**63), st.integers(0, 2**31 - 1),
+                          st.integers(1, 2**32 - 1)), min_size=1, max_size=64))
+def test_map_output_table_roundtrip(entries):
+    from sparkrdma_amd.map_output import BlockLocation, MapTaskOutput
+    t = MapTaskOutput(len(entries))
+    for i, (a, l, k) in enumerate(entries):
+        t.put(i, a, l, k)
+    parsed = MapTaskOutput.parse_locations(t.tobytes())
+    assert parsed == [BlockLocation(a, l, k) for a, l, k in entries]
