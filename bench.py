@@ -25,7 +25,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--gb-per-gpu", type=float,
                     default=float(os.environ.get("TERASORT_GB_PER_GPU", 40)))
-    ap.add_argument("--mode", choices=["framework", "rccl"],
+    ap.add_argument("--mode",
+                    choices=["framework", "rccl", "shuffleread"],
                     default=os.environ.get("TERASORT_MODE", "framework"))
     ap.add_argument("--workload",
                     choices=["terasort", "pagerank", "join", "groupby"],
@@ -82,10 +83,10 @@ def main():
 
     device = "cuda" if use_cuda else "cpu"
     if args.workload == "terasort":
+        mode = args.mode if (use_cuda or args.mode != "rccl") \
+            else "framework"
         ts = TeraSort(eng, n_rec, partitions_per_executor=ppe,
-                      device=device,
-                      mode=args.mode if use_cuda else "framework",
-                      validate=args.validate)
+                      device=device, mode=mode, validate=args.validate)
     elif args.workload == "pagerank":
         from sparkrdma_amd.workloads.pagerank import PageRank
         # 19 GB edge list analog: 16 B records
@@ -132,7 +133,10 @@ def main():
         elapsed = float(t.item())
 
     ms_per_step = elapsed / args.steps * 1e3
-    if args.workload == "terasort":
+    if args.workload == "terasort" and args.mode == "shuffleread":
+        per_step_bytes = n_rec * TeraSort.RECORD_BYTES * world
+        metric = "shuffle_read_gb_per_s"
+    elif args.workload == "terasort":
         per_step_bytes = n_rec * TeraSort.RECORD_BYTES * world
         metric = "terasort_sorted_gb_per_s"
     elif args.workload == "pagerank":

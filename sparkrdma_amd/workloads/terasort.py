@@ -83,7 +83,45 @@ class TeraSort:
     def run_step(self) -> TeraSortResult:
         if self.mode == "rccl":
             return self._step_rccl()
+        if self.mode == "shuffleread":
+            return self._step_shuffleread()
         return self._step_framework()
+
+    def _step_shuffleread(self) -> TeraSortResult:
+        """Pure one-sided read bandwidth: the map output is written once
+        (first step); every step then re-fetches this rank's partition
+        range over xGMI/shm — BASELINE's 'shuffle-read GB/s at 1/2/4/8'
+        measured directly, no sort in the timed path."""
+        eng = self.engine
+        rank = eng.rank
+        t0 = time.perf_counter()
+        if getattr(self, "_sr_handle", None) is None:
+            handle = eng.register_shuffle(eng.world_size, self.R)
+            w = eng.manager.get_writer(handle, rank)
+            if self.device == "cuda":
+                w.write_device_batch(self.keys, self.vals)
+            else:
+                w.write_batch(self.keys, self.vals)
+            w.stop(True, partitioner=self.part)
+            eng.barrier()
+            self._sr_handle = handle
+        t_write = time.perf_counter()
+        lo, hi = rank * self.ppe, (rank + 1) * self.ppe - 1
+        arena = self._arenas(1)[0]
+        reader = eng.manager.get_reader(self._sr_handle, lo, hi, arena=arena)
+        n_blocks = sum(1 for _ in reader)
+        if self.device == "cuda":
+            import torch
+            torch.cuda.synchronize()
+        t_fetch = time.perf_counter()
+        eng.barrier()
+        dt = time.perf_counter() - t0
+        fetched = (reader.metrics.remote_bytes_read
+                   + reader.metrics.local_bytes_read)
+        return TeraSortResult(
+            seconds=dt, records=self.n, bytes_sorted=fetched,
+            write_s=t_write - t0, fetch_s=t_fetch - t_write, sort_s=0.0,
+            remote_bytes=reader.metrics.remote_bytes_read)
 
     def _step_framework(self) -> TeraSortResult:
         eng = self.engine
