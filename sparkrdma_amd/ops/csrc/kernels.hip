@@ -17,11 +17,14 @@
 //   * scattered global writes are coalesced through an LDS exchange
 //     (elements digit-sorted in LDS, then written out linearly — each
 //     digit's run is a contiguous global write burst).
-//   * 3-kernel pass: hist -> column scan -> scatter (classic upsweep/
-//     downsweep; digit-major [ND][nb] hist layout so the scan kernel
-//     reads contiguously).
+//   * partition path: 3-kernel (hist -> hierarchical scan -> scatter;
+//     row-major [nb][ND] hist so every access is thread==digit coalesced)
+//     because the host must see counts to lay out HBM blocks first.
+//   * sort path: onesweep — one kernel per pass with decoupled lookback
+//     (u64 flag|count agent-scope descriptors), AoS records, deferred
+//     walk; selectable hist+scan mode kept for A/B.
 //
-// Limits: n < 2^32 elements per call; buckets = pow2, 2^8..2^12.
+// Limits: n < 2^32 elements per call; buckets = pow2, 2^4..2^12.
 
 #include "common.h"
 
