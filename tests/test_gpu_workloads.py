@@ -88,3 +88,11 @@ def test_terasort_rccl_mode_single_rank(engine):
                   mode="rccl", validate=True)
     r = ts.run_step()
     assert r.records == 1_000_000
+
+
+def test_reduce_by_key_gpu(engine):
+    from sparkrdma_amd.workloads.reduce_by_key import ReduceByKey
+    r = ReduceByKey(engine, rows_per_executor=400_000, num_keys=3000,
+                    partitions_per_executor=32, device="cuda", validate=True)
+    res = r.run_step()
+    assert res.groups == 3000

@@ -59,3 +59,11 @@ def test_groupby_cpu(engine):
     r = g.run_step()
     assert r.groups == 97
     assert r.rows == 20_000
+
+
+def test_reduce_by_key_cpu(engine):
+    from sparkrdma_amd.workloads.reduce_by_key import ReduceByKey
+    r = ReduceByKey(engine, rows_per_executor=30_000, num_keys=500,
+                    partitions_per_executor=16, device="cpu", validate=True)
+    res = r.run_step()
+    assert res.groups == 500
