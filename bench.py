@@ -29,7 +29,8 @@ def main():
                     choices=["framework", "rccl", "shuffleread"],
                     default=os.environ.get("TERASORT_MODE", "framework"))
     ap.add_argument("--workload",
-                    choices=["terasort", "pagerank", "join", "groupby"],
+                    choices=["terasort", "pagerank", "join", "groupby",
+                             "reducebykey"],
                     default="terasort",
                     help="terasort is the headline metric; others cover "
                          "the remaining BASELINE configs")
@@ -99,6 +100,11 @@ def main():
         ts = SortMergeJoin(eng, rows_per_executor=n_rec // 2,
                            partitions_per_executor=ppe, device=device,
                            key_space_bits=40, validate=args.validate)
+    elif args.workload == "reducebykey":
+        from sparkrdma_amd.workloads.reduce_by_key import ReduceByKey
+        ts = ReduceByKey(eng, rows_per_executor=n_rec,
+                         partitions_per_executor=ppe, device=device,
+                         validate=args.validate)
     else:
         from sparkrdma_amd.workloads.groupby import GroupByKey
         ts = GroupByKey(eng, rows_per_executor=min(n_rec, 1_000_000))
@@ -145,6 +151,9 @@ def main():
     elif args.workload == "join":
         per_step_bytes = n_rec * 16 * world  # both tables
         metric = "join_row_gb_per_s"
+    elif args.workload == "reducebykey":
+        per_step_bytes = n_rec * 16 * world
+        metric = "reducebykey_gb_per_s"
     else:
         per_step_bytes = results[0].rows * world * 64
         metric = "groupby_gb_per_s"
