@@ -14,10 +14,11 @@ partitions are packed greedily into ~shuffle_write_block_size blocks, split
 only at partition boundaries, so every partition is one contiguous
 (addr, len) range.
 
-Record model (fixed-width path): a partition segment is SoA —
-``[keys: n×u64][values: n×value_width bytes]`` — GPU-coalescing-friendly
-and n is inferable from the segment length. The bytes path (arbitrary
-pickled records) concatenates pickle frames.
+Record model (fixed-width path): a partition segment is AoS —
+``[key u64 | value bytes] × n`` interleaved records (doubles the GPU
+scatter's per-digit write bursts and lets the reduce side consume fetched
+chunks without splitting); n is inferable from the segment length. The
+bytes path (arbitrary pickled records) concatenates pickle frames.
 """
 
 from __future__ import annotations
