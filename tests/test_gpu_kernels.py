@@ -219,3 +219,17 @@ def test_sort_aos_beyond_2pow30(hs):
     # (commutative mod 2^64; with sortedness + payload equality this
     # pins the output)
     assert int(k.sum().item()) == int(keys.sum().item())
+
+
+def test_aos_sorter_persistent_buffers(hs):
+    """AosSorter: repeated sorts reuse one workspace/ping-pong pair."""
+    from sparkrdma_amd.ops.radix import AosSorter
+    n = 500_000
+    sorter = AosSorter(n)
+    for seed in (1, 2):
+        keys = rand_keys(n, seed=seed)
+        pairs = torch.stack([keys, keys], dim=1).reshape(-1).contiguous()
+        out = sorter.sort_(pairs)
+        torch.cuda.synchronize()
+        got = out[0::2].cpu().numpy().view(np.uint64)
+        assert np.array_equal(got, np.sort(keys.cpu().numpy().view(np.uint64)))
