@@ -18,6 +18,17 @@ import time
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 
+def _fetch_rate(results, ts):
+    """Mean shuffle-read GB/s over steps with a timed fetch phase (steps
+    with fetch_s == 0 are excluded, not averaged in as zeros)."""
+    rb = getattr(ts, "RECORD_BYTES", None)
+    if rb is None:
+        return None
+    rates = [(getattr(r, "records", 0) * rb / r.fetch_s) / 1e9
+             for r in results if getattr(r, "fetch_s", 0) > 0]
+    return round(sum(rates) / len(rates), 2) if rates else None
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -186,12 +197,13 @@ def main():
                 "parallelism": f"shuffle{world}",
                 "wall_clock_s_per_job": round(elapsed / args.steps, 3),
                 "remote_gb_per_step": round(remote_gb / max(1, args.steps), 2),
-                # BASELINE metric names shuffle-read GB/s explicitly
-                "shuffle_read_gb_per_s": round(
-                    sum(getattr(r, "fetch_s", 0) and
-                        (n_rec * 16 / getattr(r, "fetch_s")) / 1e9
-                        for r in results) / max(1, len(results)), 2)
-                if args.workload == "terasort" else None,
+                # BASELINE metric names shuffle-read GB/s explicitly;
+                # mean over steps that actually timed a fetch phase
+                "shuffle_read_gb_per_s": _fetch_rate(results, ts),
+                # honesty label: at N=1 every one-sided read is a local
+                # D2D copy; xGMI rates only appear at N>1
+                "read_locality": ("local_d2d" if world == 1 or remote_gb == 0
+                                  else "xgmi_remote"),
             },
         }))
     eng.shutdown()

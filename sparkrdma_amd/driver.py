@@ -144,6 +144,38 @@ class Driver:
             sid = rpc.unpack_unregister(body)
             self.unregister_shuffle(sid)
             conn.send(rpc.MSG_HANDLE, rpc.pack_handle(sid, 0, 0, "", 0))
+        elif mtype == rpc.MSG_LOOKUP:
+            sid = rpc.unpack_unregister(body)
+            with self._lock:
+                meta = self._shuffles.get(sid)
+            if meta is None:
+                conn.send(rpc.MSG_ERROR, rpc.pack_unregister(sid))
+            else:
+                conn.send(rpc.MSG_HANDLE, rpc.pack_handle(
+                    sid, meta.num_maps, meta.num_partitions,
+                    meta.table.path, 0))
+        elif mtype == rpc.MSG_TABLE_READ:
+            # cross-host hop 1: serve the driver table over RPC (executors
+            # on the driver's host read it one-sidedly via mmap instead)
+            sid = rpc.unpack_unregister(body)
+            with self._lock:
+                meta = self._shuffles.get(sid)
+            if meta is None:
+                conn.send(rpc.MSG_ERROR, rpc.pack_unregister(sid))
+            else:
+                raw = meta.table.read(0, meta.num_maps * MAP_ENTRY_SIZE)
+                conn.send(rpc.MSG_TABLE_DATA, rpc.pack_table_data(sid, raw))
+        elif mtype == rpc.MSG_TABLE_WRITE:
+            sid, map_id, addr, key = rpc.unpack_table_write(body)
+            with self._lock:
+                meta = self._shuffles.get(sid)
+            if meta is None:
+                conn.send(rpc.MSG_ERROR, rpc.pack_unregister(sid))
+            else:
+                import struct as _struct
+                meta.table.write(map_id * MAP_ENTRY_SIZE,
+                                 _struct.pack("<QI", addr, key))
+                conn.send(rpc.MSG_HANDLE, rpc.pack_handle(sid, 0, 0, "", 0))
         elif mtype == rpc.MSG_BYE:
             pass
         elif mtype == rpc.MSG_BARRIER:

@@ -80,6 +80,16 @@ class Engine:
                     f"shuffle id drift: driver={h.shuffle_id} local={sid} "
                     "(register_shuffle must go through Engine)")
         self.barrier()  # driver table exists before anyone writes it
+        if self.rank != 0:
+            # verify the derived id against the authoritative registry so
+            # out-of-band registrations desynchronize LOUDLY on every rank
+            # (only rank 0 asserted)
+            nm, np_, _uri = self.manager.lookup_shuffle(sid)
+            if (nm, np_) != (num_maps, num_partitions):
+                raise RuntimeError(
+                    f"shuffle id drift on rank {self.rank}: id {sid} is "
+                    f"({nm} maps, {np_} parts) at the driver, expected "
+                    f"({num_maps}, {num_partitions})")
         return ShuffleHandle(sid, num_maps, num_partitions,
                              driver_table_path(self.conf.shm_dir,
                                                self.manager.app_id, sid))

@@ -74,6 +74,7 @@ class ShuffleManager:
         self._members: Dict[int, rpc.ExecutorInfo] = {}
         self._announce_evt = threading.Event()
         self._reply_lock = threading.Lock()
+        self._rpc_serial = threading.Lock()   # one in-flight driver RPC
         self._replies: List[tuple] = []
         self._reply_evt = threading.Event()
         self._meta_segment: Optional[HostSegment] = None
@@ -84,6 +85,7 @@ class ShuffleManager:
         self._next_segment_id = FIRST_DATA_SEGMENT_ID
         self._registry: Optional[SegmentRegistry] = None
         self._driver_tables: Dict[int, mmap.mmap] = {}   # shuffle_id -> rw mmap
+        self._remote_tables: set = set()   # shuffle ids served via RPC lane
         self._cached_tables: Dict[int, list] = {}        # shuffle_id -> [(addr,key)]
         self._shuffle_outputs: Dict[int, Dict[int, list]] = {}  # sid -> map_id -> blocks
         self.reader_stats = (ShuffleReaderStats(conf)
@@ -154,18 +156,21 @@ class ShuffleManager:
             return
 
     def _rpc_call(self, mtype: int, body: bytes, timeout: float = 30.0) -> tuple:
-        self._conn.send(mtype, body)
-        deadline = time.monotonic() + timeout
-        while True:
-            if not self._reply_evt.wait(max(0.0, deadline - time.monotonic())):
-                raise TimeoutError(f"driver RPC {mtype} timed out")
-            with self._reply_lock:
-                if self._replies:
-                    reply = self._replies.pop(0)
-                    if not self._replies:
-                        self._reply_evt.clear()
-                    return reply
-                self._reply_evt.clear()
+        # serialize request/response pairs: the reply FIFO has no request
+        # ids, so two concurrent callers could swap replies
+        with self._rpc_serial:
+            self._conn.send(mtype, body)
+            deadline = time.monotonic() + timeout
+            while True:
+                if not self._reply_evt.wait(max(0.0, deadline - time.monotonic())):
+                    raise TimeoutError(f"driver RPC {mtype} timed out")
+                with self._reply_lock:
+                    if self._replies:
+                        reply = self._replies.pop(0)
+                        if not self._replies:
+                            self._reply_evt.clear()
+                        return reply
+                    self._reply_evt.clear()
 
     def _init_segments(self) -> None:
         path = segment_path(self.conf.shm_dir, self.app_id, self.executor_id,
@@ -291,25 +296,66 @@ class ShuffleManager:
                               self._meta_segment.view(addr, nbytes))
         return table, addr
 
-    def _driver_table_mm(self, handle: ShuffleHandle) -> mmap.mmap:
-        mm = self._driver_tables.get(handle.shuffle_id)
+    def _driver_table_mm(self, handle: ShuffleHandle) -> Optional[mmap.mmap]:
+        """mmap of the driver table when it is reachable as a local file
+        (same host as the driver); None => cross-host, use the RPC lane.
+        The path embeds the app's uuid so existence == ours."""
+        sid = handle.shuffle_id
+        if sid in self._remote_tables:
+            return None
+        mm = self._driver_tables.get(sid)
         if mm is None:
-            fd = os.open(handle.driver_table_path, os.O_RDWR)
+            try:
+                fd = os.open(handle.driver_table_path, os.O_RDWR)
+            except FileNotFoundError:
+                self._remote_tables.add(sid)
+                return None
             try:
                 mm = mmap.mmap(fd, max(handle.num_maps * MAP_ENTRY_SIZE, 4096))
             finally:
                 os.close(fd)
-            self._driver_tables[handle.shuffle_id] = mm
+            self._driver_tables[sid] = mm
         return mm
 
     def publish_map_output(self, handle: ShuffleHandle, map_id: int,
                            table_addr: int) -> None:
-        """One-sided 12-byte write into the driver table at map_id*12
-        (reference RdmaShuffleManager.scala:410-412)."""
+        """12-byte write into the driver table at map_id*12 — one-sided
+        mmap store on the driver's host (reference one-sided WRITE,
+        RdmaShuffleManager.scala:410-412), RPC lane across hosts."""
         import struct
-        mm = self._driver_table_mm(handle)
         key = make_key(self.executor_id, META_SEGMENT_ID)
-        struct.pack_into("<QI", mm, map_id * MAP_ENTRY_SIZE, table_addr, key)
+        mm = self._driver_table_mm(handle)
+        if mm is not None:
+            struct.pack_into("<QI", mm, map_id * MAP_ENTRY_SIZE, table_addr, key)
+            return
+        mtype, _ = self._rpc_call(rpc.MSG_TABLE_WRITE, rpc.pack_table_write(
+            handle.shuffle_id, map_id, table_addr, key))
+        if mtype != rpc.MSG_HANDLE:
+            raise RuntimeError(
+                f"driver rejected publish for shuffle {handle.shuffle_id}")
+
+    def _read_driver_table(self, handle: ShuffleHandle) -> bytes:
+        mm = self._driver_table_mm(handle)
+        if mm is not None:
+            return bytes(mm[:handle.num_maps * MAP_ENTRY_SIZE])
+        mtype, body = self._rpc_call(
+            rpc.MSG_TABLE_READ, rpc.pack_unregister(handle.shuffle_id))
+        if mtype != rpc.MSG_TABLE_DATA:
+            raise RuntimeError(f"shuffle {handle.shuffle_id} unknown to driver")
+        sid, raw = rpc.unpack_table_data(body)
+        assert sid == handle.shuffle_id
+        return raw
+
+    def lookup_shuffle(self, shuffle_id: int) -> tuple:
+        """Ask the driver for a shuffle's (num_maps, num_partitions, path)
+        — used by ranks that did not issue the REGISTER to verify their
+        derived handle matches the authoritative registry."""
+        mtype, body = self._rpc_call(rpc.MSG_LOOKUP,
+                                     rpc.pack_unregister(shuffle_id))
+        if mtype != rpc.MSG_HANDLE:
+            raise KeyError(f"shuffle {shuffle_id} not registered with driver")
+        sid, nm, np_, uri, _ = rpc.unpack_handle(body)
+        return nm, np_, uri
 
     def get_map_task_output_table(self, handle: ShuffleHandle) -> list:
         """Hop 1: read the whole driver table one-sidedly, poll until every
@@ -318,12 +364,11 @@ class ShuffleManager:
         cached = self._cached_tables.get(handle.shuffle_id)
         if cached is not None:
             return cached
-        mm = self._driver_table_mm(handle)
         deadline = time.monotonic() + \
             self.conf.partition_location_fetch_timeout_ms / 1000
         delay = 0.0005
         while True:
-            raw = bytes(mm[:handle.num_maps * MAP_ENTRY_SIZE])
+            raw = self._read_driver_table(handle)
             entries = DriverTable.parse(raw)
             if all(key != 0 for _, key in entries):
                 self._cached_tables[handle.shuffle_id] = entries

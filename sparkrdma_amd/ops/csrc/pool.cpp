@@ -53,7 +53,9 @@ class SlabPool {
     slabs_.erase(it);
   }
 
-  const Slab& get(int id) {
+  // By VALUE: a reference into the map would dangle if a concurrent
+  // free() erases/rehashes after the lock drops.
+  Slab get(int id) {
     std::lock_guard<std::mutex> g(mu_);
     auto it = slabs_.find(id);
     if (it == slabs_.end()) throw std::runtime_error("bad slab id");
@@ -63,7 +65,7 @@ class SlabPool {
   uintptr_t base(int id) { return reinterpret_cast<uintptr_t>(get(id).ptr); }
 
   std::string handle_bytes(int id) {
-    const Slab& s = get(id);
+    Slab s = get(id);
     return std::string(reinterpret_cast<const char*>(&s.handle),
                        sizeof(hipIpcMemHandle_t));
   }

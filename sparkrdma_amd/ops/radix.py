@@ -191,7 +191,9 @@ def sort_pairs(keys: torch.Tensor, vals: Optional[torch.Tensor] = None,
 class AosSorter:
     """Persistent-buffer AoS sorter: workspace and ping-pong buffer are
     allocated once, so a sort issues ONLY kernel launches and stream
-    memsets — hipGraph-capturable (see tests/test_gpu_graphs.py)."""
+    memsets. (Intended to make the launch sequence hipGraph-capturable;
+    torch.cuda.graph capture of the extension's launches SIGABRTs on this
+    ROCm/torch build — see ROADMAP.md — so capture is not wired up.)"""
 
     def __init__(self, n: int, device="cuda", start_bit: int = 0,
                  end_bit: int = 64):

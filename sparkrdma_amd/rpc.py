@@ -18,6 +18,9 @@ broadcast to smuggle handles through):
 * HANDLE     driver  -> app       shuffle handle (driver table coords)
 * UNREGISTER app     -> driver    drop shuffle
 * BYE        executor-> driver    clean detach
+* LOOKUP     executor-> driver    verify a shuffle id's (numMaps, numParts)
+* TABLE_READ executor-> driver    driver-table bytes (cross-host hop 1)
+* TABLE_WRITE executor-> driver   12-byte publish (cross-host publish)
 """
 
 from __future__ import annotations
@@ -37,6 +40,10 @@ MSG_BYE = 5
 MSG_ERROR = 6
 MSG_BARRIER = 7
 MSG_BARRIER_OK = 8
+MSG_LOOKUP = 9        # body: shuffle_id i32 -> HANDLE reply (or ERROR)
+MSG_TABLE_READ = 10   # body: shuffle_id i32 -> TABLE_DATA reply
+MSG_TABLE_DATA = 11   # body: shuffle_id i32 + raw table bytes
+MSG_TABLE_WRITE = 12  # body: shuffle_id i32, map_id i32, addr u64, key u32
 
 _HDR = struct.Struct("<III")  # frame_len (incl. header), msg_type, total_body_len
 
@@ -139,6 +146,29 @@ def pack_unregister(shuffle_id: int) -> bytes:
 
 def unpack_unregister(body: bytes) -> int:
     return struct.unpack("<i", body)[0]
+
+
+_TABLE_WRITE = struct.Struct("<iiQI")  # shuffle_id, map_id, addr, key
+
+
+def pack_table_write(shuffle_id: int, map_id: int, addr: int, key: int) -> bytes:
+    """Cross-host publish: the 12-byte (addr, key) driver-table entry
+    carried over RPC instead of a local one-sided write — same layout as
+    the reference's 12 B RDMA WRITE (RdmaShuffleManager.scala:410-412)."""
+    return _TABLE_WRITE.pack(shuffle_id, map_id, addr, key)
+
+
+def unpack_table_write(body: bytes) -> tuple:
+    return _TABLE_WRITE.unpack(body)
+
+
+def pack_table_data(shuffle_id: int, table: bytes) -> bytes:
+    return struct.pack("<i", shuffle_id) + table
+
+
+def unpack_table_data(body: bytes) -> tuple:
+    (sid,) = struct.unpack_from("<i", body, 0)
+    return sid, body[4:]
 
 
 # ---------------------------------------------------------------------------

@@ -96,15 +96,32 @@ class ShuffleWriter:
 
     # -- arbitrary-record path -----------------------------------------
 
-    def write_records(self, records, partitioner) -> None:
-        """records: iterable of (key, value) python objects; partition by
-        pickled-key hash (stable across processes)."""
+    def write_records(self, records, partitioner=None) -> None:
+        """records: iterable of (key, value) python objects. Partition by
+        ``partitioner`` when given (a callable key->pid, or an object with
+        ``partition_ids`` for integer keys); default is pickled-key crc32
+        (stable across processes)."""
         R = self.handle.num_partitions
         if self._byte_records is None:
             self._byte_records = [[] for _ in range(R)]
         import zlib
+        if partitioner is None:
+            def pid_of(k):
+                return zlib.crc32(pickle.dumps(k, protocol=4)) % R
+        elif hasattr(partitioner, "partition_ids"):
+            def pid_of(k):
+                return int(partitioner.partition_ids(
+                    np.array([k], dtype=np.uint64))[0])
+        elif callable(partitioner):
+            def pid_of(k):
+                return int(partitioner(k)) % R
+        else:
+            raise TypeError(f"unsupported partitioner {partitioner!r}")
         for k, v in records:
-            pid = zlib.crc32(pickle.dumps(k, protocol=4)) % R
+            pid = pid_of(k)
+            if not 0 <= pid < R:
+                raise ValueError(f"partitioner returned {pid} for key {k!r}, "
+                                 f"outside [0, {R})")
             self._byte_records[pid].append(pickle.dumps((k, v), protocol=4))
             self.metrics.records_written += 1
 
@@ -174,7 +191,7 @@ class ShuffleWriter:
         batches = self._gpu_batches
         keys = (batches[0][0] if len(batches) == 1
                 else torch.cat([b[0] for b in batches]))
-        vals = (batches[0][1] if len(batches) == 1
+        vals = (batches[0][1] if len(batches) == 1 or batches[0][1] is None
                 else torch.cat([b[1] for b in batches]))
         has_val = vals is not None
         n = keys.numel()

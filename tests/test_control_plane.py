@@ -1,0 +1,123 @@
+"""Control-plane additions of round 2: the cross-host driver-table RPC
+lane (hop 1 / publish when the driver's /dev/shm is unreachable), the
+LOOKUP verification used by Engine.register_shuffle on non-zero ranks,
+and the serialized driver-RPC correlation fix."""
+
+import threading
+
+import numpy as np
+import pytest
+
+from sparkrdma_amd.conf import ShuffleConf
+from sparkrdma_amd.driver import Driver
+from sparkrdma_amd.manager import ShuffleHandle, ShuffleManager
+
+
+@pytest.fixture
+def pair(tmp_path):
+    conf = ShuffleConf(shm_dir=str(tmp_path), max_buffer_allocation_size=1 << 30)
+    driver = Driver(conf)
+    conf.driver_port = driver.port
+    mgr = ShuffleManager(conf, executor_id=0, driver_port=driver.port)
+    yield driver, mgr
+    mgr.stop()
+    driver.stop()
+
+
+def test_lookup_shuffle(pair):
+    driver, mgr = pair
+    h = mgr.register_shuffle(num_maps=2, num_partitions=8)
+    nm, np_, uri = mgr.lookup_shuffle(h.shuffle_id)
+    assert (nm, np_) == (2, 8)
+    assert uri == h.driver_table_path
+    with pytest.raises(KeyError):
+        mgr.lookup_shuffle(h.shuffle_id + 17)
+
+
+def test_table_rpc_lane(pair):
+    """A handle whose table path does not exist locally (cross-host driver)
+    must route publish + hop-1 reads through the driver RPC lane and still
+    converge (ADVICE r01 medium finding)."""
+    driver, mgr = pair
+    h = mgr.register_shuffle(num_maps=2, num_partitions=4)
+    # simulate a remote driver: path that cannot be opened on this host
+    remote_h = ShuffleHandle(h.shuffle_id, h.num_maps, h.num_partitions,
+                             "/nonexistent/dir/driver_table")
+    # publish both map outputs through the RPC lane
+    for map_id in range(2):
+        table, addr = mgr.alloc_table(h.num_partitions)
+        for p in range(h.num_partitions):
+            table.put(p, 0, 0, 0x10001)
+        mgr.publish_map_output(remote_h, map_id, addr)
+    entries = mgr.get_map_task_output_table(remote_h)
+    assert len(entries) == 2
+    assert all(key != 0 for _addr, key in entries)
+    # the local-mmap view of the same shuffle agrees bit-for-bit
+    mgr._cached_tables.clear()
+    local_entries = mgr.get_map_task_output_table(h)
+    assert local_entries == entries
+
+
+def test_rpc_concurrent_correlation(pair):
+    """Two threads issuing driver RPCs concurrently must each get their
+    own reply (the r01 shared-FIFO race is closed by serializing request/
+    response pairs)."""
+    driver, mgr = pair
+    handles = [mgr.register_shuffle(num_maps=1, num_partitions=1 + i)
+               for i in range(4)]
+    errors = []
+
+    def worker(h):
+        try:
+            for _ in range(50):
+                nm, np_, _ = mgr.lookup_shuffle(h.shuffle_id)
+                assert (nm, np_) == (h.num_maps, h.num_partitions)
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    threads = [threading.Thread(target=worker, args=(h,)) for h in handles]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not errors
+
+
+def test_write_records_honors_partitioner(pair):
+    _driver, mgr = pair
+    h = mgr.register_shuffle(num_maps=1, num_partitions=4)
+    w = mgr.get_writer(h, 0)
+    # callable partitioner: everything to partition 2
+    w.write_records([(k, k * 10) for k in range(20)],
+                    partitioner=lambda k: 2)
+    w.stop(True)
+    reader = mgr.get_reader(h, 0, 3)
+    parts = reader.collect_partitions()
+    import pickle
+    got = {p: [] for p in parts}
+    for p, chunks in parts.items():
+        for chunk in chunks:
+            buf = bytes(chunk)
+            off = 0
+            while off < len(buf):
+                obj = pickle.loads(buf[off:])
+                frame = len(pickle.dumps(obj, protocol=4))
+                got[p].append(obj)
+                off += frame
+    assert len(got[2]) == 20
+    assert all(not got[p] for p in (0, 1, 3))
+
+
+def test_write_records_partition_ids_interface(pair):
+    _driver, mgr = pair
+    from sparkrdma_amd.partitioner import RangePartitioner
+    h = mgr.register_shuffle(num_maps=1, num_partitions=4)
+    w = mgr.get_writer(h, 0)
+    part = RangePartitioner.uniform(4)
+    keys = np.array([0, 2 ** 62, 2 ** 63, 2 ** 63 + 2 ** 62], dtype=np.uint64)
+    w.write_records([(int(k), 0) for k in keys], partitioner=part)
+    w.stop(True)
+    reader = mgr.get_reader(h, 0, 3)
+    parts = reader.collect_partitions()
+    sizes = {p: sum(len(c) for c in chunks) for p, chunks in parts.items()}
+    assert all(sizes[p] > 0 for p in range(4))  # one record per quarter
