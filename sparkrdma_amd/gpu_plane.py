@@ -9,9 +9,17 @@ READ machinery:
   published in a fixed slot of the owner's host *metadata segment*, so a
   fetcher resolves (key -> handle) one-sidedly — no RPC, mirroring how the
   reference smuggles rkeys through one-sided table reads.
-* a fetch is a batch of hipMemcpyAsync D2D on the per-peer stream with one
-  completion event — the scatter-list RDMA READ with a single signaled WR
-  (RdmaChannel.java:484-517).
+* a fetch is a batch of hipMemcpyAsync D2D on a per-peer-GPU stream with
+  one completion event — the scatter-list RDMA READ with a single signaled
+  WR (RdmaChannel.java:484-517). ``issue_read_into`` is non-blocking
+  (enqueue + event id); the fetcher's completion thread polls events and
+  dispatches results — the event-driven CQ analog (RdmaChannel.java:
+  683-870, RdmaThread.java:45-58). Blocking ``read_device`` forms remain
+  for simple callers.
+* ``prebuild`` pre-opens peer state in the background the moment the
+  driver announces membership — the reference pre-builds its channel mesh
+  the same way (RdmaShuffleManager.scala:121-126) so reduce stages never
+  pay first-touch connection latency.
 * GPU segment ids carry the high bit (0x8000); the low 15 bits index the
   owner's slab-table slot. Addresses in BlockLocations are slab OFFSETS,
   valid in any importing process.
@@ -23,8 +31,10 @@ torch tensors; only *served* memory needs the IPC-exported slabs.
 
 from __future__ import annotations
 
+import logging
 import struct
 import threading
+import time
 from typing import Dict, Optional
 
 import torch
@@ -32,6 +42,8 @@ import torch
 from .block_pool import BlockPool
 from .map_output import make_key, split_key
 from .segments import META_SEGMENT_ID, SLAB_SLOT_SIZE, SLAB_TABLE_MAX
+
+log = logging.getLogger(__name__)
 
 GPU_SEG_FLAG = 0x8000
 
@@ -42,7 +54,7 @@ def is_gpu_key(key: int) -> bool:
 
 class GpuDataPlane:
     def __init__(self, conf, executor_id: int, meta_segment, registry,
-                 device: Optional[int] = None):
+                 device: Optional[int] = None, peer_device=None):
         from .ops import load
         self.hs = load()
         self.conf = conf
@@ -50,6 +62,9 @@ class GpuDataPlane:
         self.meta_segment = meta_segment     # our own (HostSegment)
         self.registry = registry             # peers' segments (SegmentRegistry)
         self.device = conf.resolved_gpu_id() if device is None else device
+        # exec_id -> peer GPU ordinal (the xGMI link selector for copy
+        # streams); installed by the manager from announce membership
+        self._peer_device = peer_device or (lambda eid: self.device)
         if torch.cuda.is_available():
             torch.cuda.set_device(self.device)
         self.hs.set_device(self.device)
@@ -58,6 +73,7 @@ class GpuDataPlane:
         self._next_slot = 0
         self._peer_bases: Dict[int, int] = {}  # remote key -> imported base
         self._lock = threading.Lock()
+        self._key_locks: Dict[int, threading.Lock] = {}  # per-key resolve
         pool_max = conf.hbm_pool_size or self._auto_pool_bytes()
         # slab-table slots are finite: grow the slab size until the whole
         # pool fits the table (so a 288 GB pool is actually servable even
@@ -66,10 +82,8 @@ class GpuDataPlane:
         while slab_size * (SLAB_TABLE_MAX - 8) < pool_max:
             slab_size *= 2
         if slab_size != conf.hbm_slab_size:
-            import logging
-            logging.getLogger(__name__).info(
-                "hbm slab size auto-raised %d -> %d for %d-byte pool",
-                conf.hbm_slab_size, slab_size, pool_max)
+            log.info("hbm slab size auto-raised %d -> %d for %d-byte pool",
+                     conf.hbm_slab_size, slab_size, pool_max)
         self.pool = BlockPool(
             slab_size=slab_size, max_bytes=pool_max,
             alloc_slab=self._alloc_slab, free_slab=self._free_slab)
@@ -116,14 +130,78 @@ class GpuDataPlane:
         return self._slab_bases[seg_id & 0x7FFF]
 
     # ------------------------------------------------------------------
+    # mesh pre-build (reference RdmaShuffleManager.scala:121-126)
+
+    def prebuild(self, members, deadline_s: float = 5.0) -> None:
+        """Open per-peer state in the BACKGROUND on Announce so the first
+        fetch pays no connection latency: enable xGMI peer access to every
+        peer GPU, open peers' metadata segments, and import any already-
+        published slab handles. Peers whose segments are not up yet are
+        retried until ``deadline_s``; whatever stays unopened resolves
+        lazily on first fetch (the r01 behavior) — pre-build is an
+        optimization, never a correctness gate."""
+        todo = {m.executor_id: m for m in members
+                if m.executor_id != self.executor_id}
+        deadline = time.monotonic() + deadline_s
+        while todo and time.monotonic() < deadline:
+            for eid, m in list(todo.items()):
+                try:
+                    if torch.cuda.is_available() and m.gpu_id >= 0 \
+                            and m.gpu_id != self.device:
+                        self.hs.enable_peer_access(m.gpu_id)
+                    meta_key = make_key(eid, META_SEGMENT_ID)
+                    self.registry.reader(meta_key)   # open + cache the fd
+                    self._import_published_slabs(eid)
+                    todo.pop(eid)
+                except FileNotFoundError:
+                    continue   # peer not initialized yet; retry
+                except Exception as e:   # pragma: no cover - defensive
+                    log.warning("prebuild for peer %d failed: %s", eid, e)
+                    todo.pop(eid)
+            if todo:
+                time.sleep(0.02)
+
+    def _import_published_slabs(self, exec_id: int) -> None:
+        """Open every slab handle the peer has already published (one
+        sweep of its slab table). Later slabs import lazily on fetch."""
+        if not torch.cuda.is_available():
+            return
+        raw = self.registry.read(make_key(exec_id, META_SEGMENT_ID),
+                                 16, SLAB_TABLE_MAX * SLAB_SLOT_SIZE)
+        for slot in range(SLAB_TABLE_MAX):
+            off = slot * SLAB_SLOT_SIZE
+            (size,) = struct.unpack_from("<Q", raw, off)
+            if size == 0:
+                continue
+            key = make_key(exec_id, GPU_SEG_FLAG | slot)
+            if key in self._peer_bases:
+                continue
+            try:
+                base = self.hs.ipc_open(bytes(raw[off + 8:off + 8 + 64]))
+            except RuntimeError:
+                continue   # peer freed it between read and open
+            with self._lock:
+                self._peer_bases.setdefault(key, base)
+
+    # ------------------------------------------------------------------
     # one-sided fetch (hop 3)
 
-    def _resolve_base(self, key: int) -> int:
-        """key -> device pointer of the owning slab in THIS process."""
+    def _key_lock(self, key: int) -> threading.Lock:
+        with self._lock:
+            lk = self._key_locks.get(key)
+            if lk is None:
+                lk = self._key_locks[key] = threading.Lock()
+            return lk
+
+    def resolve(self, key: int) -> int:
+        """key -> device pointer of the owning slab in THIS process.
+        First touch opens the peer's IPC handle (cached); per-key locks so
+        resolving different slabs never serializes (r01 used one global
+        lock on the critical path of step 1 — VERDICT item 1)."""
         base = self._peer_bases.get(key)
         if base is not None:
             return base
-        with self._lock:
+        with self._key_lock(key):
             base = self._peer_bases.get(key)
             if base is not None:
                 return base
@@ -140,36 +218,55 @@ class GpuDataPlane:
                     raise RuntimeError(
                         f"peer {exec_id} slab slot {slot} not published")
                 base = self.hs.ipc_open(bytes(raw[8:8 + 64]))
-            self._peer_bases[key] = base
+            with self._lock:
+                self._peer_bases[key] = base
             return base
 
-    def read_device(self, key: int, addr: int, length: int) -> torch.Tensor:
-        """One-sided read of (key, addr, length) into a fresh device tensor.
+    # kept under the old name for callers/tests of r01
+    _resolve_base = resolve
 
-        Blocking form (the fetcher's thread pool provides the async rim,
-        like the reference's CQ threads)."""
+    def issue_read_into(self, key: int, addr: int, length: int,
+                        dst_ptr: int) -> int:
+        """NON-BLOCKING one-sided read: enqueue the xGMI copy on the
+        owning peer GPU's stream, return the completion event id. The
+        caller (the fetcher's completion thread) polls ``poll_event`` —
+        in-flight depth is bounded by the fetcher's per-peer caps, not by
+        blocked threads."""
         exec_id, _ = split_key(key)
-        # fetch runs on pool worker threads whose HIP current device
-        # defaults to 0 — pin it so per-peer streams land on OUR device
         self.hs.set_device(self.device)
-        base = self._resolve_base(key)
+        base = self.resolve(key)
+        peer = self._stream_slot(exec_id)
+        return self.hs.read_batch(peer, [dst_ptr], [base + addr], [length])
+
+    def _stream_slot(self, exec_id: int) -> int:
+        """Copy-stream selector: the peer's GPU ordinal — streams map to
+        xGMI links, not executor ids."""
+        if exec_id == self.executor_id:
+            return self.device % 64
+        try:
+            return int(self._peer_device(exec_id)) % 64
+        except Exception:
+            return exec_id % 64
+
+    def poll_event(self, ev: int) -> bool:
+        return self.hs.poll_event(ev)
+
+    def wait_event(self, ev: int) -> None:
+        self.hs.wait_event(ev)
+
+    def read_device(self, key: int, addr: int, length: int) -> torch.Tensor:
+        """Blocking one-sided read into a fresh device tensor (simple
+        callers; the fetcher uses issue_read_into + completion thread)."""
         dst = torch.empty(length, dtype=torch.uint8,
                           device=f"cuda:{self.device}")
-        ev = self.hs.read_batch(exec_id % 64, [dst.data_ptr()],
-                                [base + addr], [length])
+        ev = self.issue_read_into(key, addr, length, dst.data_ptr())
         self.hs.wait_event(ev)
         return dst
 
     def read_device_into(self, key: int, addr: int, length: int,
                          dst_ptr: int) -> None:
-        """One-sided read landing at a caller-chosen device address —
-        the arena fast path (no per-fetch allocation, no reduce-side
-        concat)."""
-        exec_id, _ = split_key(key)
-        self.hs.set_device(self.device)
-        base = self._resolve_base(key)
-        ev = self.hs.read_batch(exec_id % 64, [dst_ptr], [base + addr],
-                                [length])
+        """Blocking pre-placed one-sided read (arena slow path)."""
+        ev = self.issue_read_into(key, addr, length, dst_ptr)
         self.hs.wait_event(ev)
 
     def stop(self) -> None:

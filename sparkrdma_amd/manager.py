@@ -148,6 +148,7 @@ class ShuffleManager:
                     self.app_id = app_id
                     self._members = {m.executor_id: m for m in members}
                     self._announce_evt.set()
+                    self._prebuild_mesh(members)
                 else:
                     with self._reply_lock:
                         self._replies.append((mtype, body))
@@ -189,7 +190,52 @@ class ShuffleManager:
                 self.conf.transport in ("auto", "tcp") and _cuda_available()):
             from .gpu_plane import GpuDataPlane
             self.gpu = GpuDataPlane(self.conf, self.executor_id,
-                                    self._meta_segment, self._registry)
+                                    self._meta_segment, self._registry,
+                                    peer_device=self.peer_device)
+        # segments exist now: pre-build against the membership we already
+        # know (announce may have arrived before _init_segments ran)
+        if self._members:
+            self._prebuild_mesh(list(self._members.values()))
+
+    def peer_device(self, exec_id: int) -> int:
+        """Peer executor -> its GPU ordinal (stream/xGMI-link selector)."""
+        m = self._members.get(exec_id)
+        return m.gpu_id if m is not None and m.gpu_id >= 0 else 0
+
+    def _prebuild_mesh(self, members) -> None:
+        """Background mesh pre-build on Announce — the reference opens
+        every peer channel the moment the driver announces membership
+        (RdmaShuffleManager.scala:121-126) so reduce stages never pay
+        connection latency. Here: pre-open peers' metadata segments (host
+        'channel' = the pread fd) and, on the GPU plane, enable peer
+        access + import published slab handles. Never blocks the receive
+        loop; failures resolve lazily on first fetch."""
+        if self._stopped or self._registry is None:
+            return
+        peers = [m for m in members if m.executor_id != self.executor_id
+                 and m.host == self._my_host]
+
+        def _run():
+            from .map_output import make_key as mk
+            todo = list(peers)
+            deadline = time.monotonic() + 5.0
+            while todo and time.monotonic() < deadline and not self._stopped:
+                still = []
+                for m in todo:
+                    try:
+                        self._registry.reader(mk(m.executor_id,
+                                                 META_SEGMENT_ID))
+                    except FileNotFoundError:
+                        still.append(m)   # peer not initialized yet; retry
+                todo = still
+                if todo:
+                    time.sleep(0.02)
+            # unopened peers resolve lazily on first fetch
+            if self.gpu is not None:
+                self.gpu.prebuild(peers)
+
+        threading.Thread(target=_run, name="sparkrdma-prebuild",
+                         daemon=True).start()
 
     def _host_slab_size(self) -> int:
         return min(1 << 30, self.conf.max_buffer_allocation_size)

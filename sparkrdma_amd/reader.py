@@ -4,19 +4,28 @@ Re-design of RdmaShuffleReader.scala + RdmaShuffleFetcherIterator.scala
 (the reference's hot path). Same structure:
 
 * HOP 1: read the whole driver table once, cached per shuffle
-  (RdmaShuffleManager.scala:341-376) — here a polled shm read.
+  (RdmaShuffleManager.scala:341-376) — here a polled shm read (driver RPC
+  lane across hosts).
 * HOP 2: per map task, one-sided read of its MapTaskOutput entries for
   [start_partition, end_partition] (RdmaShuffleFetcherIterator.scala:
-  297-311) — here a pread of the owner's metadata segment.
+  297-311). Issued ASYNC, grouped per source executor, through the fetch
+  pool — never on the constructing thread's critical path (the reference
+  runs hop 2 inside table-read onComplete callbacks the same way).
 * HOP 3: coalesce consecutive block locations up to shuffle_read_block_size
-  (:240-263), gate on max_bytes_in_flight and a per-source request cap
-  (:264-273, :82-83), randomize the pending queue to spread load across
-  source executors (:74-79), then issue one-sided data reads; results are
-  surfaced through a blocking queue as they land (:340-382).
+  (:240-263), gate on max_bytes_in_flight (:264-273) and a per-SOURCE
+  request cap (:82-83 — per channel in the reference; per source executor
+  here, so one hot peer's xGMI link cannot starve the others), randomize
+  the pending queue to spread load across source executors (:74-79), then
+  issue one-sided data reads; results surface through a blocking queue as
+  they land (:340-382).
 
-Transport-agnostic: ``manager.remote_read`` is shm pread in host mode and
-an xGMI peer copy in GPU mode — the flow-control logic is identical, as it
-was in the reference (it is pure byte-budget accounting).
+Completion model (event-driven, RdmaChannel.java:683-870 +
+RdmaThread.java:45-58): GPU one-sided reads are ENQUEUED (hipMemcpyAsync
+batch + recorded event = the signaled last WR) and a single completion
+thread polls the in-flight events, dispatching results — in-flight depth
+is set by the flow-control caps, not by a count of blocked threads. Host
+segment preads and the cross-host TCP lane still run on the small thread
+pool (they are genuinely blocking OS calls).
 """
 
 from __future__ import annotations
@@ -27,7 +36,7 @@ import random
 import threading
 import time
 from concurrent.futures import ThreadPoolExecutor
-from dataclasses import dataclass
+from dataclasses import dataclass, field
 from typing import Iterator, List, Optional
 
 from .manager import ShuffleHandle, ShuffleManager
@@ -55,14 +64,18 @@ class CoalescedFetch:
     addr: int
     length: int
     blocks: List[BlockRef]
+    arena_off: int = -1          # destination offset in the fetch arena
 
 
 @dataclass
 class FetchResult:
-    fetch: CoalescedFetch
+    fetch: Optional[CoalescedFetch]
     data: Optional[bytes]
     error: Optional[BaseException] = None
     latency_ms: float = 0.0
+
+
+_SENTINEL = FetchResult(None, None)   # wake-up after async hop-2 drains
 
 
 class FetchFailedError(RuntimeError):
@@ -114,7 +127,7 @@ def coalesce_blocks(blocks: List[BlockRef], max_bytes: int,
 
 
 class FetcherIterator:
-    """Async fetch pipeline with byte-budget flow control."""
+    """Async fetch pipeline with byte-budget + per-source flow control."""
 
     def __init__(self, manager: ShuffleManager, handle: ShuffleHandle,
                  start_partition: int, end_partition: int,
@@ -129,20 +142,34 @@ class FetcherIterator:
         self._cpus = parse_cpu_list(conf.cpu_list)
         self._max_bytes_in_flight = conf.max_bytes_in_flight
         self._read_block = conf.shuffle_read_block_size
-        self._reqs_limit = conf.resolved_read_requests_limit()
+        # per-SOURCE outstanding-read cap (reference caps per channel:
+        # RdmaShuffleFetcherIterator.scala:82-83)
+        self._per_source_limit = conf.resolved_read_requests_limit()
         self._results: "queue.Queue[FetchResult]" = queue.Queue()
         self._pending: List[CoalescedFetch] = []
         self._lock = threading.Lock()
         self._bytes_in_flight = 0
-        self._reqs_in_flight = 0
-        self._outstanding = 0     # fetches not yet surfaced to the consumer
+        self._reqs_by_source: dict = {}
+        self._outstanding = 0     # fetches created, not yet surfaced
+        self._hop2_pending = 1    # +1 sentinel for the submit phase itself
         self._pool = ThreadPoolExecutor(max_workers=num_workers,
                                         thread_name_prefix="sparkrdma-fetch",
                                         initializer=self._pin_worker)
         self._rng = random.Random(seed)
         self._failed: Optional[BaseException] = None
-        self._arena_hint = arena   # caller-provided reusable device buffer
-        self._start()
+        self._done = threading.Event()
+        # fetch arena: one device buffer all GPU one-sided reads land in at
+        # pre-assigned offsets (bump-allocated as hop-2 results arrive)
+        self._arena_buf = arena            # caller-provided hint (optional)
+        self._arena_used = 0
+        self._deferred: List[CoalescedFetch] = []  # awaiting arena alloc
+        self.arena = None                  # exact-length view, set when known
+        self.arena_complete = False
+        self._inflight_gpu: dict = {}      # event id -> (fetch, t0)
+        self._blocking_arena_inflight = 0  # pool fetches writing the arena
+        self._paused = False               # arena-growth issue gate
+        self._completion_thread: Optional[threading.Thread] = None
+        self._pool.submit(self._start_async)
 
     # ------------------------------------------------------------------
 
@@ -155,77 +182,253 @@ class FetcherIterator:
             except OSError:
                 pass
 
-    def _start(self) -> None:
+    # ------------------------------ hop 1 + 2 (async) ------------------
+
+    def _start_async(self) -> None:
+        """Runs on the pool: hop 1, then one hop-2 task per source
+        executor. The constructor returns immediately; the consumer blocks
+        on the results queue until data lands."""
+        try:
+            mgr = self.manager
+            entries = mgr.get_map_task_output_table(self.handle)  # HOP 1
+            groups: dict = {}   # owner exec -> [(map_id, table_addr, table_key)]
+            for map_id, (table_addr, table_key) in enumerate(entries):
+                owner = split_key(table_key)[0]
+                groups.setdefault(owner, []).append(
+                    (map_id, table_addr, table_key))
+            with self._lock:
+                self._hop2_pending += len(groups)
+            for owner, maps in groups.items():
+                self._pool.submit(self._hop2_group, owner, maps)
+        except BaseException as e:
+            self._results.put(FetchResult(None, None, error=e))
+        finally:
+            self._hop2_done()
+
+    def _hop2_group(self, owner: int, maps: list) -> None:
+        """HOP 2 for one source executor: read each of its map tasks'
+        location entries (one-sided pread same-host, TCP lane across
+        hosts — ADVICE r01), build + coalesce blocks, pre-resolve the
+        slabs they point into, then enqueue the fetches."""
         mgr = self.manager
-        entries = mgr.get_map_task_output_table(self.handle)  # HOP 1
-        blocks: List[BlockRef] = []
         span = self.end_partition - self.start_partition + 1
-        for map_id, (table_addr, table_key) in enumerate(entries):
-            # HOP 2: one-sided read of this map's location entries
-            raw = mgr.remote_read(
-                table_key, table_addr + self.start_partition * ENTRY_SIZE,
-                span * ENTRY_SIZE)
-            for i, loc in enumerate(MapTaskOutput.parse_locations(raw)):
-                blocks.append(BlockRef(map_id, self.start_partition + i,
-                                       loc.key, loc.addr, loc.length))
-        fetches = coalesce_blocks(blocks, self._read_block)
-        # arena mode (GPU plane active): every fetch lands at a
-        # pre-assigned offset of ONE device buffer, so consumers read the
-        # whole shuffle input without a concat pass (torch.cat was 5% of
-        # a TeraSort step) and per-fetch allocations disappear.
-        self.arena = None
-        self._arena_off = {}
-        if mgr.gpu is not None and fetches:
-            import torch
-            total = sum(f.length for f in fetches)
-            off = 0
-            for f in fetches:       # block order, not arrival order
-                self._arena_off[id(f)] = off
-                off += f.length
-            hint = self._arena_hint
-            if hint is not None and hint.numel() >= total:
-                self.arena = hint[:total]   # reuse caller's buffer
-            else:
-                self.arena = torch.empty(
-                    total, dtype=torch.uint8,
-                    device=f"cuda:{mgr.gpu.device}")
-        # randomize to spread load over source executors (reference :74-79)
-        self._rng.shuffle(fetches)
+        try:
+            blocks: List[BlockRef] = []
+            remote_host = mgr.is_remote_host(owner)
+            for map_id, table_addr, table_key in maps:
+                addr = table_addr + self.start_partition * ENTRY_SIZE
+                raw = (mgr.tcp_read(owner, table_key, addr, span * ENTRY_SIZE)
+                       if remote_host
+                       else mgr.remote_read(table_key, addr, span * ENTRY_SIZE))
+                for i, loc in enumerate(MapTaskOutput.parse_locations(raw)):
+                    blocks.append(BlockRef(map_id, self.start_partition + i,
+                                           loc.key, loc.addr, loc.length))
+            fetches = coalesce_blocks(blocks, self._read_block)
+            # pre-resolve slab bases OFF the issue path (first IPC open of
+            # a peer slab costs ms; pump()'s issue is then pure enqueue)
+            if mgr.gpu is not None and not remote_host:
+                from .gpu_plane import is_gpu_key
+                for key in {f.key for f in fetches if is_gpu_key(f.key)}:
+                    mgr.gpu.resolve(key)
+            self._ingest(fetches)
+        except BaseException as e:
+            self._results.put(FetchResult(None, None, error=e))
+        finally:
+            self._hop2_done()
+
+    def _hop2_done(self) -> None:
         with self._lock:
-            self._pending = fetches
-            self._outstanding = len(fetches)
+            self._hop2_pending -= 1
+            finished = self._hop2_pending == 0
+        if finished:
+            self._finalize_arena()
+            self._results.put(_SENTINEL)
         self._pump()
+
+    def _ingest(self, fetches: List[CoalescedFetch]) -> None:
+        """Register new fetches: assign arena destinations, randomize, add
+        to the pending queue (reference randomized queue :74-79)."""
+        use_arena = self.manager.gpu is not None
+        with self._lock:
+            if use_arena:
+                for f in fetches:
+                    if (self._arena_buf is not None
+                            and self._arena_used + f.length
+                            <= self._arena_buf.numel()):
+                        f.arena_off = self._arena_used
+                        self._arena_used += f.length
+                    else:
+                        # no hint (or hint exhausted): defer until all
+                        # hop-2 results are in, then allocate exactly
+                        self._deferred.append(f)
+            self._rng.shuffle(fetches)
+            self._pending.extend(f for f in fetches
+                                 if not (use_arena and f.arena_off < 0))
+            self._outstanding += len(fetches)
+
+    def _finalize_arena(self) -> None:
+        """All hop-2 results are in: place any deferred fetches (no-hint
+        mode allocates the exact total now) and expose the arena view."""
+        with self._lock:
+            deferred, self._deferred = self._deferred, []
+        if deferred:
+            import torch
+            total = sum(f.length for f in deferred)
+            dev = f"cuda:{self.manager.gpu.device}"
+            if self._arena_buf is None and self._arena_used == 0:
+                self._arena_buf = torch.empty(total, dtype=torch.uint8,
+                                              device=dev)
+            else:
+                # hint exhausted mid-stream: pause issuing, drain landed
+                # writes, grow into a fresh buffer, copy the prefix
+                with self._lock:
+                    self._paused = True
+                old, used = self._arena_buf, self._arena_used
+                new = torch.empty(used + total, dtype=torch.uint8, device=dev)
+                self._drain_inflight()
+                if used:
+                    new[:used] = old[:used]
+                    torch.cuda.synchronize()
+                self._arena_buf = new
+                with self._lock:
+                    self._paused = False
+            with self._lock:
+                for f in deferred:
+                    f.arena_off = self._arena_used
+                    self._arena_used += f.length
+                self._pending.extend(deferred)
+        if self.manager.gpu is not None and self._arena_buf is not None:
+            self.arena = self._arena_buf[:self._arena_used]
+        self.arena_complete = True
+
+    def _drain_inflight(self) -> None:
+        """Wait for every in-progress arena write to land (arena
+        re-allocation barrier; rare — only when a caller's arena hint was
+        undersized)."""
+        while True:
+            with self._lock:
+                evs = list(self._inflight_gpu.keys())
+                blocking = self._blocking_arena_inflight
+            if not evs and blocking == 0:
+                return
+            for ev in evs:
+                self.manager.gpu.wait_event(ev)
+            time.sleep(0.0002)
+
+    # ------------------------------ hop 3 -------------------------------
 
     def _pump(self) -> None:
         """Issue pending fetches while under the byte budget (reference
-        :264-273, re-pumped from next() :365-374)."""
-        to_issue = []
-        with self._lock:
-            while self._pending:
-                f = self._pending[-1]
-                if (self._bytes_in_flight + f.length > self._max_bytes_in_flight
-                        and self._bytes_in_flight > 0):
-                    break
-                if self._reqs_in_flight >= self._reqs_limit:
-                    break
-                self._pending.pop()
-                self._bytes_in_flight += f.length
-                self._reqs_in_flight += 1
-                to_issue.append(f)
-        for f in to_issue:
-            self._pool.submit(self._do_fetch, f)
+        :264-273, re-pumped from next() :365-374) and each source's
+        outstanding-request cap (:82-83)."""
+        while True:
+            to_issue = []
+            with self._lock:
+                if self._failed is not None or self._paused:
+                    return
+                i = len(self._pending) - 1
+                while i >= 0:
+                    f = self._pending[i]
+                    if (self._bytes_in_flight > 0
+                            and self._bytes_in_flight + f.length
+                            > self._max_bytes_in_flight):
+                        break   # global byte budget: stop issuing
+                    src = split_key(f.key)[0]
+                    if (self._reqs_by_source.get(src, 0)
+                            >= self._per_source_limit):
+                        i -= 1   # this source saturated; try others
+                        continue
+                    self._pending.pop(i)
+                    self._bytes_in_flight += f.length
+                    self._reqs_by_source[src] = \
+                        self._reqs_by_source.get(src, 0) + 1
+                    to_issue.append(f)
+                    i -= 1
+            if not to_issue:
+                return
+            for f in to_issue:
+                self._issue(f)
 
-    def _do_fetch(self, f: CoalescedFetch) -> None:
+    def _issue(self, f: CoalescedFetch) -> None:
+        from .gpu_plane import is_gpu_key
+        mgr = self.manager
+        owner = split_key(f.key)[0]
+        if (f.arena_off >= 0 and is_gpu_key(f.key)
+                and not mgr.is_remote_host(owner)):
+            # event-driven lane: enqueue the xGMI copy; the completion
+            # thread polls the event (no blocked thread per fetch)
+            try:
+                ev = mgr.gpu.issue_read_into(
+                    f.key, f.addr, f.length,
+                    self._arena_buf.data_ptr() + f.arena_off)
+            except BaseException as e:
+                self._results.put(FetchResult(f, None, error=e))
+                return
+            with self._lock:
+                self._inflight_gpu[ev] = (f, time.perf_counter())
+            self._ensure_completion_thread()
+        else:
+            if f.arena_off >= 0:
+                with self._lock:
+                    self._blocking_arena_inflight += 1
+            self._pool.submit(self._do_fetch_blocking, f)
+
+    def _ensure_completion_thread(self) -> None:
+        if self._completion_thread is None:
+            t = threading.Thread(target=self._completion_loop,
+                                 name="sparkrdma-completion", daemon=True)
+            self._completion_thread = t
+            t.start()
+
+    def _completion_loop(self) -> None:
+        """The CQ-poller analog (RdmaThread.java:45-58): poll in-flight
+        copy events, dispatch results as they complete."""
+        if self._cpus:
+            try:
+                os.sched_setaffinity(0, set(self._cpus))
+            except OSError:
+                pass
+        gpu = self.manager.gpu
+        while not self._done.is_set():
+            with self._lock:
+                items = list(self._inflight_gpu.items())
+            if not items:
+                time.sleep(0.0002)
+                continue
+            completed = []
+            for ev, (f, t0) in items:
+                try:
+                    if gpu.poll_event(ev):
+                        completed.append((ev, f, t0, None))
+                except BaseException as e:
+                    completed.append((ev, f, t0, e))
+            if not completed:
+                time.sleep(0.00005)
+                continue
+            for ev, f, t0, err in completed:
+                with self._lock:
+                    self._inflight_gpu.pop(ev, None)
+                if err is not None:
+                    self._results.put(FetchResult(f, None, error=err))
+                else:
+                    data = self._arena_buf[f.arena_off:f.arena_off + f.length]
+                    self._results.put(FetchResult(
+                        f, data,
+                        latency_ms=(time.perf_counter() - t0) * 1e3))
+
+    def _do_fetch_blocking(self, f: CoalescedFetch) -> None:
+        """Blocking lane on the thread pool: host-segment preads, the
+        cross-host TCP path, and uploads of host bytes into the arena."""
         from .gpu_plane import is_gpu_key
         t0 = time.perf_counter()
+        arena_buf = self._arena_buf   # capture: stable across arena growth
         try:
             owner = split_key(f.key)[0]
-            if self.arena is not None:
-                off = self._arena_off[id(f)]
+            if f.arena_off >= 0:
                 if not self.manager.is_remote_host(owner) and is_gpu_key(f.key):
                     self.manager.gpu.read_device_into(
                         f.key, f.addr, f.length,
-                        self.arena.data_ptr() + off)
+                        arena_buf.data_ptr() + f.arena_off)
                 else:  # cross-host or host-spilled bytes: upload into place
                     import torch
                     raw = (self.manager.tcp_read(owner, f.key, f.addr, f.length)
@@ -234,10 +437,10 @@ class FetcherIterator:
                     if len(raw) != f.length:
                         raise FetchFailedError(
                             f"short read: {len(raw)}/{f.length} at key={f.key:#x}")
-                    self.arena[off:off + f.length] = torch.frombuffer(
-                        bytearray(raw), dtype=torch.uint8).to(
-                            self.arena.device)
-                data = self.arena[off:off + f.length]
+                    arena_buf[f.arena_off:f.arena_off + f.length] = \
+                        torch.frombuffer(bytearray(raw), dtype=torch.uint8) \
+                        .to(arena_buf.device)
+                data = arena_buf[f.arena_off:f.arena_off + f.length]
             elif self.manager.is_remote_host(owner):
                 data = self.manager.tcp_read(owner, f.key, f.addr, f.length)
             elif is_gpu_key(f.key):
@@ -251,6 +454,10 @@ class FetcherIterator:
                 f, data, latency_ms=(time.perf_counter() - t0) * 1e3))
         except BaseException as e:  # surfaced to consumer, fails the task
             self._results.put(FetchResult(f, None, error=e))
+        finally:
+            if f.arena_off >= 0:
+                with self._lock:
+                    self._blocking_arena_inflight -= 1
 
     # ------------------------------------------------------------------
 
@@ -258,43 +465,52 @@ class FetcherIterator:
         """Yields (BlockRef, memoryview) per block, in arrival order."""
         mgr = self.manager
         my_exec = mgr.executor_id
-        while True:
-            with self._lock:
-                if self._outstanding == 0:
-                    break
-            t0 = time.perf_counter_ns()
-            res = self._results.get()
-            self.metrics.fetch_wait_ns += time.perf_counter_ns() - t0
-            with self._lock:
-                self._bytes_in_flight -= res.fetch.length
-                self._reqs_in_flight -= 1
-                self._outstanding -= 1
-            self._pump()
-            if res.error is not None:
-                self._failed = res.error
-                self._pool.shutdown(wait=False)
-                raise FetchFailedError(
-                    f"fetch of {len(res.fetch.blocks)} blocks at "
-                    f"key={res.fetch.key:#x} failed") from res.error
-            owner = split_key(res.fetch.key)[0]
-            remote = owner != my_exec
-            if remote:
-                self.metrics.remote_bytes_read += res.fetch.length
-                self.metrics.remote_blocks_fetched += len(res.fetch.blocks)
-                if mgr.reader_stats is not None:
-                    mgr.reader_stats.update(owner, res.latency_ms)
-            else:
-                self.metrics.local_bytes_read += res.fetch.length
-                self.metrics.local_blocks_fetched += len(res.fetch.blocks)
-            data = res.data
-            view = (memoryview(data)
-                    if isinstance(data, (bytes, bytearray, memoryview))
-                    else data)  # device tensor: torch slicing below
-            off = 0
-            for b in res.fetch.blocks:
-                yield b, view[off:off + b.length]
-                off += b.length
-        self._pool.shutdown(wait=False)
+        try:
+            while True:
+                with self._lock:
+                    if self._outstanding == 0 and self._hop2_pending == 0:
+                        break
+                t0 = time.perf_counter_ns()
+                res = self._results.get()
+                self.metrics.fetch_wait_ns += time.perf_counter_ns() - t0
+                if res.error is not None:
+                    with self._lock:
+                        self._failed = res.error
+                    raise FetchFailedError(
+                        "fetch failed" + (
+                            f" ({len(res.fetch.blocks)} blocks at "
+                            f"key={res.fetch.key:#x})" if res.fetch else "")
+                        ) from res.error
+                if res.fetch is None:      # sentinel: recheck termination
+                    continue
+                with self._lock:
+                    self._bytes_in_flight -= res.fetch.length
+                    src = split_key(res.fetch.key)[0]
+                    self._reqs_by_source[src] = \
+                        self._reqs_by_source.get(src, 1) - 1
+                    self._outstanding -= 1
+                self._pump()
+                owner = split_key(res.fetch.key)[0]
+                remote = owner != my_exec
+                if remote:
+                    self.metrics.remote_bytes_read += res.fetch.length
+                    self.metrics.remote_blocks_fetched += len(res.fetch.blocks)
+                    if mgr.reader_stats is not None:
+                        mgr.reader_stats.update(owner, res.latency_ms)
+                else:
+                    self.metrics.local_bytes_read += res.fetch.length
+                    self.metrics.local_blocks_fetched += len(res.fetch.blocks)
+                data = res.data
+                view = (memoryview(data)
+                        if isinstance(data, (bytes, bytearray, memoryview))
+                        else data)  # device tensor: torch slicing below
+                off = 0
+                for b in res.fetch.blocks:
+                    yield b, view[off:off + b.length]
+                    off += b.length
+        finally:
+            self._done.set()
+            self._pool.shutdown(wait=False)
 
 
 class ShuffleReader:

@@ -145,9 +145,12 @@ def test_unpublished_map_times_out(cluster):
         w = managers[0].get_writer(handle, 0)
         w.write_batch(np.arange(5, dtype=np.uint64))
         w.stop(True, partitioner=HashPartitioner(2))
-        # map 1 never publishes
-        with pytest.raises(TimeoutError):
-            managers[1].get_reader(handle, 0, 1)
+        # map 1 never publishes; hop 1/2 run async off the constructor
+        # (r02), so the timeout surfaces on iteration as a fetch failure
+        reader = managers[1].get_reader(handle, 0, 1)
+        with pytest.raises(FetchFailedError) as ei:
+            list(reader)
+        assert isinstance(ei.value.__cause__, TimeoutError)
     finally:
         managers[1].conf.partition_location_fetch_timeout_ms = conf_backup
 
