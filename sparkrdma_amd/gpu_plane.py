@@ -133,13 +133,13 @@ class GpuDataPlane:
     # mesh pre-build (reference RdmaShuffleManager.scala:121-126)
 
     def prebuild(self, members, deadline_s: float = 5.0) -> None:
-        """Open per-peer state in the BACKGROUND on Announce so the first
-        fetch pays no connection latency: enable xGMI peer access to every
-        peer GPU, open peers' metadata segments, and import any already-
-        published slab handles. Peers whose segments are not up yet are
-        retried until ``deadline_s``; whatever stays unopened resolves
-        lazily on first fetch (the r01 behavior) — pre-build is an
-        optimization, never a correctness gate."""
+        """One-shot per-peer setup on Announce: enable xGMI peer access to
+        every peer GPU and open peers' metadata segments. Peers whose
+        segments are not up yet are retried until ``deadline_s``; whatever
+        stays unopened resolves lazily on first fetch — pre-build is an
+        optimization, never a correctness gate. Slab handles themselves
+        are imported continuously by ``importer_tick`` (slabs appear as
+        writers commit, long after Announce)."""
         todo = {m.executor_id: m for m in members
                 if m.executor_id != self.executor_id}
         deadline = time.monotonic() + deadline_s
@@ -160,6 +160,19 @@ class GpuDataPlane:
                     todo.pop(eid)
             if todo:
                 time.sleep(0.02)
+
+    def importer_tick(self, members) -> None:
+        """One sweep of every same-host peer's slab table, importing any
+        newly published handles — keeps hop-3 issue latency independent of
+        first-touch (the reference needs no analog: ibverbs rkeys are
+        usable without a per-MR open; hipIpc handles are not)."""
+        for m in members:
+            if m.executor_id == self.executor_id:
+                continue
+            try:
+                self._import_published_slabs(m.executor_id)
+            except FileNotFoundError:
+                pass   # peer not initialized yet
 
     def _import_published_slabs(self, exec_id: int) -> None:
         """Open every slab handle the peer has already published (one

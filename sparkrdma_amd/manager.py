@@ -196,6 +196,24 @@ class ShuffleManager:
         # know (announce may have arrived before _init_segments ran)
         if self._members:
             self._prebuild_mesh(list(self._members.values()))
+        if self.gpu is not None:
+            threading.Thread(target=self._slab_importer_loop,
+                             name="sparkrdma-slab-importer",
+                             daemon=True).start()
+
+    def _slab_importer_loop(self) -> None:
+        """Continuously import peers' newly published slab handles so the
+        fetch hot path never does a first-touch hipIpcOpen (the channel
+        pre-build discipline of RdmaShuffleManager.scala:121-126 extended
+        to served memory, which on ROCm needs an explicit import)."""
+        while not self._stopped:
+            members = [m for m in self._members.values()
+                       if m.host == self._my_host]
+            try:
+                self.gpu.importer_tick(members)
+            except Exception:   # pragma: no cover - defensive
+                pass
+            time.sleep(0.05)
 
     def peer_device(self, exec_id: int) -> int:
         """Peer executor -> its GPU ordinal (stream/xGMI-link selector)."""
