@@ -47,6 +47,10 @@ def main():
                          "the remaining BASELINE configs")
     ap.add_argument("--partitions-per-executor", type=int, default=0,
                     help="0 = auto (pow2, ~128 per GPU)")
+    ap.add_argument("--record-bytes", type=int,
+                    default=int(os.environ.get("TERASORT_RECORD_BYTES", 16)),
+                    help="terasort record width: 16 (u64 key + u64 payload) "
+                         "or 100 (canonical 10B key + 90B value, GPU only)")
     ap.add_argument("--validate", action="store_true")
     ap.add_argument("--cpu", action="store_true",
                     help="force CPU path (plumbing debug)")
@@ -72,7 +76,10 @@ def main():
     from sparkrdma_amd.engine import Engine
     from sparkrdma_amd.workloads.terasort import TeraSort
 
-    n_rec = int(args.gb_per_gpu * (1 << 30) / TeraSort.RECORD_BYTES)
+    rec_bytes = args.record_bytes if args.workload == "terasort" else 16
+    if rec_bytes != 16 and (not use_cuda or args.mode != "framework"):
+        rec_bytes = 16   # wide records: GPU framework mode only
+    n_rec = int(args.gb_per_gpu * (1 << 30) / rec_bytes)
     ppe = args.partitions_per_executor
     if ppe == 0:
         # keep R = 256 total: 8-bit partition pass AND a 7-pass reduce sort
@@ -98,7 +105,8 @@ def main():
         mode = args.mode if (use_cuda or args.mode != "rccl") \
             else "framework"
         ts = TeraSort(eng, n_rec, partitions_per_executor=ppe,
-                      device=device, mode=mode, validate=args.validate)
+                      device=device, mode=mode, validate=args.validate,
+                      record_bytes=rec_bytes)
     elif args.workload == "pagerank":
         from sparkrdma_amd.workloads.pagerank import PageRank
         # 19 GB edge list analog: 16 B records
@@ -151,10 +159,10 @@ def main():
 
     ms_per_step = elapsed / args.steps * 1e3
     if args.workload == "terasort" and args.mode == "shuffleread":
-        per_step_bytes = n_rec * TeraSort.RECORD_BYTES * world
+        per_step_bytes = n_rec * ts.RECORD_BYTES * world
         metric = "shuffle_read_gb_per_s"
     elif args.workload == "terasort":
-        per_step_bytes = n_rec * TeraSort.RECORD_BYTES * world
+        per_step_bytes = n_rec * ts.RECORD_BYTES * world
         metric = "terasort_sorted_gb_per_s"
     elif args.workload == "pagerank":
         per_step_bytes = results[0].iterations * n_rec * 16 * world
@@ -185,12 +193,14 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "u64key+u64payload",
+            "dtype": ("10Bkey+90Bval" if rec_bytes == 100 else
+                      f"{rec_bytes}B-records" if rec_bytes != 16 else
+                      "u64key+u64payload"),
             "data": "synthetic",
             "config": {
                 "model": args.workload,
                 "global_batch": n_rec * world,
-                "seq_len": TeraSort.RECORD_BYTES,
+                "seq_len": ts.RECORD_BYTES,
                 "dataset_gb": round(total_bytes / (1 << 30), 1),
                 "partitions": world * ppe,
                 "mode": getattr(ts, "mode", args.mode),

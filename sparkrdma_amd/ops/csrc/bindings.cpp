@@ -49,14 +49,29 @@ PYBIND11_MODULE(_hipshuffle, m) {
   m.def("radix_scan_ws_bytes", &hs::radix_scan_ws_bytes);
   m.def("radix_hist", &hs::radix_hist, py::arg("keys"), py::arg("n"),
         py::arg("shift"), py::arg("nbits"), py::arg("hist"),
-        py::arg("stream") = 0, py::arg("hash_mix") = 0);
+        py::arg("stream") = 0, py::arg("func") = 0, py::arg("in_stride") = 1,
+        py::arg("nparts") = 0);
   m.def("radix_scan", &hs::radix_scan, py::arg("hist"), py::arg("n"),
         py::arg("nbits"), py::arg("totals"), py::arg("scan_ws"),
         py::arg("stream") = 0);
   m.def("radix_scatter", &hs::radix_scatter, py::arg("keys"), py::arg("vals"),
         py::arg("n"), py::arg("shift"), py::arg("nbits"), py::arg("hist"),
         py::arg("key_dst"), py::arg("val_dst"), py::arg("stream") = 0,
-        py::arg("hash_mix") = 0, py::arg("aos_out") = 0);
+        py::arg("func") = 0, py::arg("aos_out") = 0, py::arg("in_stride") = 1,
+        py::arg("nparts") = 0);
+  m.def("extract_pairs", &hs::extract_pairs, py::arg("recs"), py::arg("n"),
+        py::arg("rec_bytes"), py::arg("key_bytes"), py::arg("pairs"),
+        py::arg("stream") = 0);
+  m.def("gather_records", &hs::gather_records, py::arg("recs"),
+        py::arg("pairs"), py::arg("n"), py::arg("rec_bytes"),
+        py::arg("dst_mode"), py::arg("out_base"), py::arg("dst_addr") = 0,
+        py::arg("dstart") = 0, py::arg("shift") = 0, py::arg("mask") = 0,
+        py::arg("func") = 0, py::arg("nparts") = 0, py::arg("stream") = 0);
+  m.def("onesweep_sort_aos_word_u64", &hs::onesweep_sort_aos_word_u64,
+        py::arg("pairs"), py::arg("tmp_pairs"), py::arg("n"),
+        py::arg("start_bit"), py::arg("end_bit"), py::arg("ws"),
+        py::arg("stream") = 0, py::arg("sort_word") = 0,
+        py::call_guard<py::gil_scoped_release>());
   m.def("sort_workspace_bytes", &hs::sort_workspace_bytes);
   m.def("sort_pairs_u64", &hs::sort_pairs_u64, py::arg("keys"),
         py::arg("vals"), py::arg("tmp_keys"), py::arg("tmp_vals"),

@@ -28,13 +28,21 @@ void memcpy_d2h(uintptr_t dst, uintptr_t src, size_t n);
 size_t radix_hist_bytes(uint32_t n, int nbits);
 size_t radix_scan_ws_bytes(uint32_t n, int nbits);
 void radix_hist(uintptr_t keys, uint32_t n, int shift, int nbits,
-                uintptr_t hist, uintptr_t stream, int hash_mix = 0);
+                uintptr_t hist, uintptr_t stream, int func = 0,
+                int in_stride = 1, uint32_t nparts = 0);
 void radix_scan(uintptr_t hist, uint32_t n, int nbits, uintptr_t totals,
                 uintptr_t scan_ws, uintptr_t stream);
 void radix_scatter(uintptr_t keys, uintptr_t vals, uint32_t n, int shift,
                    int nbits, uintptr_t hist, uintptr_t key_dst,
-                   uintptr_t val_dst, uintptr_t stream, int hash_mix = 0,
-                   int aos_out = 0);
+                   uintptr_t val_dst, uintptr_t stream, int func = 0,
+                   int aos_out = 0, int in_stride = 1, uint32_t nparts = 0);
+void extract_pairs(uintptr_t recs, uint64_t n, uint32_t rec_bytes,
+                   uint32_t key_bytes, uintptr_t pairs, uintptr_t stream);
+void gather_records(uintptr_t recs, uintptr_t pairs, uint64_t n,
+                    uint32_t rec_bytes, int dst_mode, uint64_t out_base,
+                    uintptr_t dst_addr, uintptr_t dstart, int shift,
+                    uint32_t mask, int func, uint32_t nparts,
+                    uintptr_t stream);
 size_t sort_workspace_bytes(uint32_t n);
 int sort_pairs_u64(uintptr_t keys, uintptr_t vals, uintptr_t tmp_keys,
                    uintptr_t tmp_vals, uint32_t n, int start_bit, int end_bit,
@@ -61,6 +69,10 @@ int onesweep_sort_aos7_u64(uintptr_t pairs, uintptr_t tmp_pairs, uint32_t n,
 int onesweep_sort_aos_u64(uintptr_t pairs, uintptr_t tmp_pairs, uint32_t n,
                           int start_bit, int end_bit, uintptr_t ws,
                           uintptr_t stream);
+int onesweep_sort_aos_word_u64(uintptr_t pairs, uintptr_t tmp_pairs,
+                               uint32_t n, int start_bit, int end_bit,
+                               uintptr_t ws, uintptr_t stream,
+                               int sort_word);
 int onesweep_sort_pairs_u64(uintptr_t keys, uintptr_t vals,
                             uintptr_t tmp_keys, uintptr_t tmp_vals,
                             uint32_t n, int start_bit, int end_bit,

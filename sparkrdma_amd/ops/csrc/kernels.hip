@@ -91,6 +91,32 @@ __device__ __forceinline__ uint64_t hash_mix64(uint64_t k) {
   return k;
 }
 
+// Partition functions (the `func` runtime param of hist/scatter). 0/1 are
+// the r01 bit-extraction modes; 2/3 lift the pow2-only restriction
+// (VERDICT r01 missing item 7): any R <= 2^12 via multiply-high range
+// partitioning (pid = floor(key * R / 2^64), exactly the CPU
+// RangePartitioner with ceil bounds) or hash-mod. The digit still fits
+// the pow2 LDS counter layout (ND = next pow2 >= R).
+enum PartFunc : int {
+  kPartBits = 0,      // (key >> shift) & mask
+  kPartHashBits = 1,  // (mix(key) >> shift) & mask   (== mod for pow2 R)
+  kPartRange = 2,     // mulhi(key, R)                 (any R)
+  kPartHashMod = 3,   // mix(key) % R                  (any R)
+};
+
+__device__ __forceinline__ uint32_t digit_of(uint64_t k, int shift,
+                                             uint32_t mask, int func,
+                                             uint32_t nparts) {
+  switch (func) {
+    default:
+    case kPartBits: return (uint32_t)(k >> shift) & mask;
+    case kPartHashBits: return (uint32_t)(hash_mix64(k) >> shift) & mask;
+    case kPartRange:
+      return (uint32_t)(((unsigned __int128)k * nparts) >> 64);
+    case kPartHashMod: return (uint32_t)(hash_mix64(k) % nparts);
+  }
+}
+
 // Match from PRE-COMPUTED per-bit ballots (register votes) — lets rank
 // logic reuse one iteration's ballots against another iteration's digits.
 template <int PBITS>
@@ -123,8 +149,8 @@ __device__ __forceinline__ uint64_t match_lanes(uint32_t digit,
 template <int NBITS>
 __global__ __launch_bounds__(BLOCK) void radix_hist_kernel(
     const uint64_t* __restrict__ keys, uint32_t n, int shift,
-    uint32_t* __restrict__ hist, uint32_t nb, int hash_mix,
-    int in_stride = 1) {
+    uint32_t* __restrict__ hist, uint32_t nb, int func,
+    int in_stride = 1, uint32_t nparts = 0) {
   constexpr int ND = 1 << NBITS;
   extern __shared__ char smem_raw[];
   uint32_t* counters = reinterpret_cast<uint32_t*>(smem_raw);  // [NW][ND]
@@ -151,9 +177,7 @@ __global__ __launch_bounds__(BLOCK) void radix_hist_kernel(
   for (int i = 0; i < ITEMS; ++i) {
     uint64_t e = chunk + (uint64_t)i * kWave + lane;
     bool valid = e < n;
-    uint64_t k = key_reg[i];
-    if (hash_mix) k = hash_mix64(k);
-    uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
+    uint32_t d = digit_of(key_reg[i], shift, ND - 1, func, nparts);
     uint64_t vm = __ballot(valid);
     if (valid) {
       uint64_t match = match_lanes<NBITS>(d, vm);
@@ -261,7 +285,7 @@ __global__ __launch_bounds__(BLOCK) void radix_scatter_kernel(
     const uint64_t* __restrict__ keys, const uint64_t* __restrict__ vals,
     uint32_t n, int shift, const uint32_t* __restrict__ hist, uint32_t nb,
     const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst,
-    int hash_mix, int aos_out) {
+    int func, int aos_out, int in_stride = 1, uint32_t nparts = 0) {
   constexpr int ND = 1 << NBITS;
   extern __shared__ char smem_raw[];
   // layout: exchange u64[TILE] | counters u32[NW][ND] | start u32[ND]
@@ -295,14 +319,13 @@ __global__ __launch_bounds__(BLOCK) void radix_scatter_kernel(
 #pragma unroll
   for (int i = 0; i < ITEMS; ++i) {
     uint64_t e = chunk + (uint64_t)i * kWave + lane;
-    key_reg[i] = e < n ? keys[e] : 0;
+    key_reg[i] = e < n ? keys[e * in_stride] : 0;
   }
 #pragma unroll
   for (int i = 0; i < ITEMS; ++i) {
     uint64_t e = chunk + (uint64_t)i * kWave + lane;
     bool valid = e < n;
-    uint64_t k = key_reg[i];
-    uint32_t d = (uint32_t)((hash_mix ? hash_mix64(k) : k) >> shift) & (ND - 1);
+    uint32_t d = digit_of(key_reg[i], shift, ND - 1, func, nparts);
     dig_reg[i] = d;
     uint64_t vm = __ballot(valid);
     uint32_t r = 0;
@@ -353,7 +376,7 @@ __global__ __launch_bounds__(BLOCK) void radix_scatter_kernel(
     uint32_t j = i * BLOCK + tid;
     if (j < tile_n) {
       uint64_t k = exch[j];
-      uint32_t d = (uint32_t)((hash_mix ? hash_mix64(k) : k) >> shift) & (ND - 1);
+      uint32_t d = digit_of(k, shift, ND - 1, func, nparts);
       uint32_t off = (pref[d] + (j - start[d])) << aos_out;
       out_d[i] = d;
       out_off[i] = off;
@@ -365,7 +388,7 @@ __global__ __launch_bounds__(BLOCK) void radix_scatter_kernel(
 #pragma unroll 1
     for (int i = 0; i < ITEMS; ++i) {
       uint64_t e = chunk + (uint64_t)i * kWave + lane;
-      if (e < n) exch[rank_reg[i]] = vals[e];
+      if (e < n) exch[rank_reg[i]] = vals[e * in_stride];
     }
     __syncthreads();
 #pragma unroll 1
@@ -430,7 +453,8 @@ __device__ __forceinline__ void lookback_walk(
 template <int MAX_PASSES, int PBITS>
 __global__ __launch_bounds__(BLOCK) void onesweep_hist_all_kernel(
     const uint64_t* __restrict__ keys, uint32_t n, int start_bit, int passes,
-    uint32_t* __restrict__ totals /* [passes][1<<PBITS] */, int in_stride) {
+    uint32_t* __restrict__ totals /* [passes][1<<PBITS] */, int in_stride,
+    int word_off = 0) {
   constexpr int PD = 1 << PBITS;
   __shared__ uint32_t cnt[NW][MAX_PASSES * PD];
   const int tid = threadIdx.x;
@@ -440,7 +464,7 @@ __global__ __launch_bounds__(BLOCK) void onesweep_hist_all_kernel(
   __syncthreads();
   const uint64_t stride = (uint64_t)gridDim.x * BLOCK;
   for (uint64_t e = (uint64_t)blockIdx.x * BLOCK + tid; e < n; e += stride) {
-    uint64_t k = keys[e * in_stride];
+    uint64_t k = keys[e * in_stride + word_off];
     for (int p = 0; p < passes; ++p) {
       int sb = start_bit + p * PBITS;
       if (sb > 64 - PBITS) sb = 64 - PBITS;  // last pass re-covers top bits
@@ -468,7 +492,7 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
     uint32_t* __restrict__ ticket,
     const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst,
     int stage = 3, uint64_t* __restrict__ timing = nullptr,
-    const uint32_t* __restrict__ hist_pref = nullptr) {
+    const uint32_t* __restrict__ hist_pref = nullptr, int sort_word = 0) {
   constexpr int ND = 1 << PBITS;
   extern __shared__ char smem_raw[];
   constexpr int TILE_T = BS * IT;
@@ -532,8 +556,12 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
     uint64_t e0 = chunk + (uint64_t)i0 * kWave + lane;
     uint64_t e1 = chunk + (uint64_t)i1 * kWave + lane;
     bool val0 = e0 < n, val1 = e1 < n;
-    uint32_t d0 = (uint32_t)(key_reg[i0] >> shift) & (ND - 1);
-    uint32_t d1 = (uint32_t)(key_reg[i1] >> shift) & (ND - 1);
+    // sort_word selects WHICH u64 of the AoS pair orders this pass
+    // (the 80-bit-key path sorts 16 aux bits before the 64 prefix bits)
+    uint64_t sk0 = (AOS && sort_word) ? val_reg[i0] : key_reg[i0];
+    uint64_t sk1 = (AOS && sort_word) ? val_reg[i1] : key_reg[i1];
+    uint32_t d0 = (uint32_t)(sk0 >> shift) & (ND - 1);
+    uint32_t d1 = (uint32_t)(sk1 >> shift) & (ND - 1);
     uint64_t vm0 = __ballot(val0), vm1 = __ballot(val1);
     uint64_t v0[PBITS], v1[PBITS];
 #pragma unroll
@@ -628,7 +656,8 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
       uint32_t j = i * BS + tid;
       if (j < tile_n) {
         u64x2 kv = exch2[j];
-        uint32_t d = (uint32_t)((uint64_t)kv.x >> shift) & (ND - 1);
+        uint32_t d = (uint32_t)((uint64_t)(sort_word ? kv.y : kv.x)
+                                >> shift) & (ND - 1);
         uint32_t off = (pref[d] + (j - start[d])) & off_mask;
         reinterpret_cast<u64x2*>(key_dst[d])[off] = kv;
       }
@@ -705,6 +734,123 @@ __global__ void onesweep_digit_bases_kernel(
     key_dst[tid] = out_keys + (uint64_t)arr[tid] * rec_bytes;
     val_dst[tid] = out_vals ? out_vals + (uint64_t)arr[tid] * rec_bytes : 0;
   }
+}
+
+// ---------------------------------------------------------------------------
+// Wide-record machinery (VERDICT r01 missing item 3: the reference serves
+// WHATEVER bytes the writer produced — RdmaMappedFile.java:113-157 —
+// while the r01 fast path handled only 16-byte records; canonical
+// TeraSort is 10 B key + 90 B value = 100 B records).
+//
+// Design: records stay put while 16-byte (sortkey, aux) PAIRS flow
+// through the existing radix machinery, then ONE gather pass moves each
+// W-byte record to its final position:
+//   * extract_pairs: record -> (key-prefix u64, aux = keylo16<<48 | idx)
+//   * partition/sort the pairs (existing kernels; `key_word` selects
+//     which u64 of the pair the digit comes from, so the 80-bit key
+//     sorts as 16 aux bits then 64 prefix bits, LSD-stable)
+//   * gather_records: walk the grouped/sorted pairs linearly, copy record
+//     idx to its slot — writes are linear (W-byte runs), reads are
+//     W-byte contiguous chunks; both sides move each record exactly once.
+// Traffic: ~2.2x the record bytes + ~64 B/pass for the 16 B pairs —
+// pair passes cost 16% of what full-record passes would at W = 100.
+
+// Key layout inside a record: bytes [0,8) u64 LE prefix, bytes [8,10)
+// u16 LE low bits (key_bytes == 8 or 10); 80-bit order = (prefix, lo).
+__global__ __launch_bounds__(BLOCK) void extract_pairs_kernel(
+    const uint8_t* __restrict__ recs, uint64_t n, uint32_t rec_bytes,
+    uint32_t key_bytes, uint64_t* __restrict__ pairs) {
+  const uint64_t stride = (uint64_t)gridDim.x * BLOCK;
+  for (uint64_t e = (uint64_t)blockIdx.x * BLOCK + threadIdx.x; e < n;
+       e += stride) {
+    const uint8_t* p = recs + e * rec_bytes;
+    // 4-byte-aligned loads (rec_bytes % 4 == 0 enforced host-side)
+    uint32_t lo = *reinterpret_cast<const uint32_t*>(p);
+    uint32_t hi = *reinterpret_cast<const uint32_t*>(p + 4);
+    uint64_t prefix = ((uint64_t)hi << 32) | lo;
+    uint64_t aux = e;
+    if (key_bytes > 8) {
+      uint16_t klo = (uint16_t)(*reinterpret_cast<const uint32_t*>(p + 8));
+      aux |= (uint64_t)klo << 48;
+    }
+    reinterpret_cast<ulonglong2*>(pairs)[e] =
+        ulonglong2{prefix, aux};
+  }
+}
+
+// Move records to their final slots following the grouped/sorted pairs.
+// Slot j's record index comes from pairs[2j+1]; its destination is
+//   dst_mode 0: out_base + j*rec_bytes                  (sorted output)
+//   dst_mode 1: dst_addr[d] + (j - dstart[d])*rec_bytes (partition path,
+//               d = digit_of(pairs[2j]) — per-digit HBM block positions)
+constexpr uint64_t AUX_IDX_MASK = (1ull << 48) - 1;
+
+__global__ __launch_bounds__(BLOCK) void gather_records_kernel(
+    const uint32_t* __restrict__ recs, const uint64_t* __restrict__ pairs,
+    uint64_t n, uint32_t w4 /* rec_bytes/4 */, uint32_t recs_per_block,
+    int dst_mode, uint64_t out_base,
+    const uint64_t* __restrict__ dst_addr,
+    const uint32_t* __restrict__ dstart, int shift, uint32_t mask,
+    int func, uint32_t nparts) {
+  const uint64_t r0 = (uint64_t)blockIdx.x * recs_per_block;
+  if (r0 >= n) return;
+  const uint32_t nr = (uint32_t)min((uint64_t)recs_per_block, n - r0);
+  const uint32_t total = nr * w4;
+  for (uint32_t w = threadIdx.x; w < total; w += BLOCK) {
+    uint32_t r = w / w4, o = w - r * w4;
+    uint64_t j = r0 + r;
+    ulonglong2 pr = reinterpret_cast<const ulonglong2*>(pairs)[j];
+    uint64_t idx = pr.y & AUX_IDX_MASK;
+    uint32_t* dst;
+    if (dst_mode == 0) {
+      dst = reinterpret_cast<uint32_t*>(out_base) + j * w4;
+    } else {
+      uint32_t d = digit_of(pr.x, shift, mask, func, nparts);
+      dst = reinterpret_cast<uint32_t*>(dst_addr[d]) +
+            (uint64_t)(uint32_t)(j - dstart[d]) * w4;
+    }
+    dst[o] = recs[idx * w4 + o];
+  }
+}
+
+void extract_pairs(uintptr_t recs, uint64_t n, uint32_t rec_bytes,
+                   uint32_t key_bytes, uintptr_t pairs, uintptr_t stream) {
+  if (rec_bytes % 4 || rec_bytes < 8)
+    throw std::runtime_error("rec_bytes must be a multiple of 4, >= 8");
+  if (key_bytes != 8 && key_bytes != 10)
+    throw std::runtime_error("key_bytes must be 8 or 10");
+  if (n >> 48)
+    throw std::runtime_error("record count exceeds 2^48");
+  auto s = reinterpret_cast<hipStream_t>(stream);
+  uint32_t grid = (uint32_t)min((n + BLOCK - 1) / BLOCK, (uint64_t)4096);
+  hipLaunchKernelGGL(extract_pairs_kernel, dim3(grid ? grid : 1),
+                     dim3(BLOCK), 0, s,
+                     reinterpret_cast<const uint8_t*>(recs), n, rec_bytes,
+                     key_bytes, reinterpret_cast<uint64_t*>(pairs));
+  HIP_CHECK(hipGetLastError());
+}
+
+void gather_records(uintptr_t recs, uintptr_t pairs, uint64_t n,
+                    uint32_t rec_bytes, int dst_mode, uint64_t out_base,
+                    uintptr_t dst_addr, uintptr_t dstart, int shift,
+                    uint32_t mask, int func, uint32_t nparts,
+                    uintptr_t stream) {
+  if (rec_bytes % 4) throw std::runtime_error("rec_bytes % 4 != 0");
+  auto s = reinterpret_cast<hipStream_t>(stream);
+  uint32_t w4 = rec_bytes / 4;
+  uint32_t rpb = 16384 / w4;           // ~16k words per block
+  if (rpb < 1) rpb = 1;
+  uint64_t grid = (n + rpb - 1) / rpb;
+  if (grid > 0x7FFFFFFF) throw std::runtime_error("grid too large");
+  hipLaunchKernelGGL(gather_records_kernel, dim3((uint32_t)(grid ? grid : 1)),
+                     dim3(BLOCK), 0, s,
+                     reinterpret_cast<const uint32_t*>(recs),
+                     reinterpret_cast<const uint64_t*>(pairs), n, w4, rpb,
+                     dst_mode, out_base,
+                     reinterpret_cast<const uint64_t*>(dst_addr),
+                     reinterpret_cast<const uint32_t*>(dstart), shift, mask,
+                     func, nparts);
+  HIP_CHECK(hipGetLastError());
 }
 
 // ---------------------------------------------------------------------------
@@ -846,12 +992,12 @@ size_t radix_hist_bytes(uint32_t n, int nbits) {
 
 template <int NBITS>
 static void hist_launch(const uint64_t* keys, uint32_t n, int shift,
-                        uint32_t* hist, hipStream_t s, int hash_mix = 0,
-                        int in_stride = 1) {
+                        uint32_t* hist, hipStream_t s, int func = 0,
+                        int in_stride = 1, uint32_t nparts = 0) {
   uint32_t nb = num_tiles(n);
   size_t lds = (size_t)NW * (1 << NBITS) * 4;
   hipLaunchKernelGGL(radix_hist_kernel<NBITS>, dim3(nb), dim3(BLOCK), lds, s,
-                     keys, n, shift, hist, nb, hash_mix, in_stride);
+                     keys, n, shift, hist, nb, func, in_stride, nparts);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -859,7 +1005,8 @@ template <int NBITS, bool HAS_VAL>
 static void scatter_launch(const uint64_t* keys, const uint64_t* vals,
                            uint32_t n, int shift, const uint32_t* hist,
                            const uint64_t* key_dst, const uint64_t* val_dst,
-                           hipStream_t s, int hash_mix = 0, int aos_out = 0) {
+                           hipStream_t s, int func = 0, int aos_out = 0,
+                           int in_stride = 1, uint32_t nparts = 0) {
   constexpr int ND = 1 << NBITS;
   uint32_t nb = num_tiles(n);
   size_t lds = (size_t)TILE * 8 + (size_t)NW * ND * 4 + (size_t)ND * 4 * 2 +
@@ -873,7 +1020,8 @@ static void scatter_launch(const uint64_t* keys, const uint64_t* vals,
     attr_set[NBITS] = true;
   }
   hipLaunchKernelGGL(kfn, dim3(nb), dim3(BLOCK), lds, s, keys, vals, n, shift,
-                     hist, nb, key_dst, val_dst, hash_mix, aos_out);
+                     hist, nb, key_dst, val_dst, func, aos_out, in_stride,
+                     nparts);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -892,10 +1040,12 @@ static void scatter_launch(const uint64_t* keys, const uint64_t* vals,
   }
 
 void radix_hist(uintptr_t keys, uint32_t n, int shift, int nbits,
-                uintptr_t hist, uintptr_t stream, int hash_mix) {
+                uintptr_t hist, uintptr_t stream, int func, int in_stride,
+                uint32_t nparts) {
   auto s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_NBITS(nbits, hist_launch, reinterpret_cast<const uint64_t*>(keys),
-                 n, shift, reinterpret_cast<uint32_t*>(hist), s, hash_mix);
+                 n, shift, reinterpret_cast<uint32_t*>(hist), s, func,
+                 in_stride, nparts);
 }
 
 size_t radix_scan_ws_bytes(uint32_t n, int nbits) {
@@ -934,27 +1084,28 @@ void radix_scan(uintptr_t hist, uint32_t n, int nbits, uintptr_t totals,
 template <int NBITS>
 static void scatter_hv(const uint64_t* keys, const uint64_t* vals, uint32_t n,
                        int shift, const uint32_t* hist, const uint64_t* kd,
-                       const uint64_t* vd, hipStream_t s, int hash_mix = 0,
-                       int aos_out = 0) {
+                       const uint64_t* vd, hipStream_t s, int func = 0,
+                       int aos_out = 0, int in_stride = 1,
+                       uint32_t nparts = 0) {
   if (vals)
     scatter_launch<NBITS, true>(keys, vals, n, shift, hist, kd, vd, s,
-                                hash_mix, aos_out);
+                                func, aos_out, in_stride, nparts);
   else
     scatter_launch<NBITS, false>(keys, nullptr, n, shift, hist, kd, nullptr,
-                                 s, hash_mix, aos_out);
+                                 s, func, aos_out, in_stride, nparts);
 }
 
 void radix_scatter(uintptr_t keys, uintptr_t vals, uint32_t n, int shift,
                    int nbits, uintptr_t hist, uintptr_t key_dst,
-                   uintptr_t val_dst, uintptr_t stream, int hash_mix,
-                   int aos_out) {
+                   uintptr_t val_dst, uintptr_t stream, int func,
+                   int aos_out, int in_stride, uint32_t nparts) {
   auto s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_NBITS(nbits, scatter_hv, reinterpret_cast<const uint64_t*>(keys),
                  reinterpret_cast<const uint64_t*>(vals), n, shift,
                  reinterpret_cast<const uint32_t*>(hist),
                  reinterpret_cast<const uint64_t*>(key_dst),
-                 reinterpret_cast<const uint64_t*>(val_dst), s, hash_mix,
-                 aos_out);
+                 reinterpret_cast<const uint64_t*>(val_dst), s, func,
+                 aos_out, in_stride, nparts);
 }
 
 // Full LSD sort of (keys[, vals]) by bits [start_bit, end_bit).
@@ -1009,7 +1160,8 @@ template <int PBITS>
 static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
                               uintptr_t tmp_keys, uintptr_t tmp_vals,
                               uint32_t n, int start_bit, int end_bit,
-                              uintptr_t ws, hipStream_t s, int aos) {
+                              uintptr_t ws, hipStream_t s, int aos,
+                              int sort_word = 0) {
   constexpr int PD = 1 << PBITS;
   const int aos_tile = g_aos_tile;
   uint32_t nb = aos ? os_num_tiles_t(n, aos_tile) : os_num_tiles(n);
@@ -1019,13 +1171,15 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
   uint64_t* val_dst = key_dst + PD;
   uint32_t* ticket = reinterpret_cast<uint32_t*>(val_dst + PD);
   uint64_t* desc = reinterpret_cast<uint64_t*>(ticket + 4);
-  if (!(g_sort_mode == 1 && aos && g_aos_tile == 4096)) {
+  const bool want_hist = g_sort_mode == 1 && aos && g_aos_tile == 4096 &&
+                         sort_word == 0;
+  if (!want_hist) {
     HIP_CHECK(hipMemsetAsync(totals, 0, (size_t)passes * PD * 4, s));
     uint32_t hist_grid = nb < 1024 ? (nb ? nb : 1) : 1024;
     hipLaunchKernelGGL((onesweep_hist_all_kernel<10, PBITS>), dim3(hist_grid),
                        dim3(BLOCK), 0, s,
                        reinterpret_cast<const uint64_t*>(keys), n, start_bit,
-                       passes, totals, aos ? 2 : 1);
+                       passes, totals, aos ? 2 : 1, sort_word);
     HIP_CHECK(hipGetLastError());
   }
   size_t lds_soa = (size_t)BLOCK * OS_ITEMS * 8 + (size_t)NW * PD * 4 +
@@ -1051,7 +1205,7 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
     attr_set = true;
   }
   // hist mode: reuse the desc area as [hist u32[nb*PD] | scan partials]
-  const bool use_hist = g_sort_mode == 1 && aos && aos_tile == 4096;
+  const bool use_hist = want_hist;
   uint32_t* hist32 = reinterpret_cast<uint32_t*>(desc);
   uint32_t* hist_scan_ws = hist32 + (size_t)nb * PD;
   uintptr_t src_k = keys, src_v = vals, dst_k = tmp_keys, dst_v = tmp_vals;
@@ -1079,19 +1233,19 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          sb, desc, ticket, key_dst, val_dst, g_pass_stage,
-                         g_timing_buf, pass_pref);
+                         g_timing_buf, pass_pref, sort_word);
     } else if (aos && aos_tile == 2048) {
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 4, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          sb, desc, ticket, key_dst, val_dst, g_pass_stage,
-                         g_timing_buf, pass_pref);
+                         g_timing_buf, pass_pref, sort_word);
     } else if (aos) {
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 8, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          sb, desc, ticket, key_dst, val_dst, g_pass_stage,
-                         g_timing_buf, pass_pref);
+                         g_timing_buf, pass_pref, sort_word);
     } else if (vals) {
       hipLaunchKernelGGL(
           (onesweep_pass_kernel<true, OS_ITEMS, false, BLOCK, PBITS>),
@@ -1128,6 +1282,18 @@ int onesweep_sort_aos_u64(uintptr_t pairs, uintptr_t tmp_pairs, uint32_t n,
                           uintptr_t stream) {
   return onesweep_sort_tmpl<8>(pairs, 0, tmp_pairs, 0, n, start_bit, end_bit,
                                ws, reinterpret_cast<hipStream_t>(stream), 1);
+}
+
+// sort_word = 1: pairs order by their SECOND u64's bits (the aux word of
+// the wide-record path — 80-bit keys sort as 16 aux bits then 64 prefix
+// bits across two calls, LSD-stable).
+int onesweep_sort_aos_word_u64(uintptr_t pairs, uintptr_t tmp_pairs,
+                               uint32_t n, int start_bit, int end_bit,
+                               uintptr_t ws, uintptr_t stream,
+                               int sort_word) {
+  return onesweep_sort_tmpl<8>(pairs, 0, tmp_pairs, 0, n, start_bit, end_bit,
+                               ws, reinterpret_cast<hipStream_t>(stream), 1,
+                               sort_word);
 }
 
 int onesweep_sort_aos7_u64(uintptr_t pairs, uintptr_t tmp_pairs, uint32_t n,

@@ -146,6 +146,77 @@ def sort_pairs_aos(pairs: torch.Tensor, start_bit: int = 0,
     return pairs if res == 0 else tmp
 
 
+def extract_pairs(recs: torch.Tensor, rec_bytes: int, key_bytes: int = 8,
+                  pairs: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """records -> interleaved (key-prefix u64, aux u64) pairs, where
+    aux = key_lo16 << 48 | record_index (kernels.hip extract_pairs)."""
+    m = load()
+    n = recs.numel() // rec_bytes
+    if pairs is None:
+        pairs = torch.empty(2 * n, dtype=torch.int64, device=recs.device)
+    else:
+        assert pairs.numel() >= 2 * n
+        pairs = pairs[:2 * n]
+    m.extract_pairs(recs.data_ptr(), n, rec_bytes, key_bytes,
+                    pairs.data_ptr(), _stream())
+    return pairs
+
+
+def sort_records(recs: torch.Tensor, rec_bytes: int, key_bytes: int = 8,
+                 end_bit: int = 64, out: Optional[torch.Tensor] = None,
+                 pairs: Optional[torch.Tensor] = None,
+                 tmp: Optional[torch.Tensor] = None,
+                 ws: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Sort W-byte AoS records by their 80-bit key: u64 LE prefix at
+    offset 0 (bits [0, end_bit) significant — callers whose partitions
+    share top prefix bits pass end_bit = 64 - shared) then, for
+    key_bytes == 10, a u16 LE low word at offset 8.
+
+    Only 16-byte (prefix, aux) pairs flow through the radix passes; one
+    gather pass then moves each record once — pair passes cost 16% of
+    full-record passes at W = 100 (kernels.hip 'wide-record machinery').
+    Returns the sorted records (a fresh tensor, or ``out``).
+    """
+    m = load()
+    n = recs.numel() // rec_bytes
+    if n == 0:
+        return recs
+    dev = recs.device
+    prs = extract_pairs(recs, rec_bytes, key_bytes, pairs)
+    if tmp is None:
+        tmp = torch.empty_like(prs)
+    else:
+        assert tmp.numel() >= 2 * n
+        tmp = tmp[:2 * n]
+    passes = -(-end_bit // 8) + (2 if key_bytes > 8 else 0)
+    need_ws = m.onesweep_workspace_bytes(n, passes)
+    if ws is None:
+        ws = torch.empty(need_ws, dtype=torch.uint8, device=dev)
+    else:
+        assert ws.numel() >= need_ws
+    cur, other = prs, tmp
+    if key_bytes > 8:
+        # LSD over the 80-bit key: 16 aux bits first, then the prefix
+        r = m.onesweep_sort_aos_word_u64(cur.data_ptr(), other.data_ptr(),
+                                         n, 48, 64, ws.data_ptr(),
+                                         _stream(), 1)
+        if r == 1:
+            cur, other = other, cur
+    r = m.onesweep_sort_aos_word_u64(cur.data_ptr(), other.data_ptr(), n,
+                                     0, min(end_bit, 64), ws.data_ptr(),
+                                     _stream(), 0)
+    if r == 1:
+        cur, other = other, cur
+    if out is None:
+        out = torch.empty_like(recs)
+    else:
+        assert out.numel() >= recs.numel()
+        out = out[:recs.numel()]
+    m.gather_records(recs.data_ptr(), cur.data_ptr(), n, rec_bytes, 0,
+                     out.data_ptr(), 0, 0, 0, 0, 0, 0, _stream())
+    return out
+
+
 def sort_pairs(keys: torch.Tensor, vals: Optional[torch.Tensor] = None,
                start_bit: int = 0, end_bit: int = 64,
                onesweep: Optional[bool] = None

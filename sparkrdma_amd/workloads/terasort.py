@@ -45,17 +45,27 @@ class TeraSortResult:
 
 
 class TeraSort:
+    """record_bytes = 16: (u64 key, u64 payload) AoS — the r01 fast shape.
+    record_bytes = 100 (canonical): 10 B key (u64 LE prefix + u16 LE low)
+    + 90 B payload, the reference benchmark's true record
+    (README.md:7-17); GPU-only (wide-record kernels)."""
+
     RECORD_BYTES = 16
 
     def __init__(self, engine: Engine, records_per_executor: int,
                  partitions_per_executor: int = 64,
                  device: str = "cpu", mode: str = "framework",
-                 validate: bool = False, seed: int = 0):
+                 validate: bool = False, seed: int = 0,
+                 record_bytes: int = 16):
         self.engine = engine
         self.n = records_per_executor
         self.device = device
         self.mode = mode
         self.validate = validate
+        self.RECORD_BYTES = record_bytes
+        self.wide = record_bytes != 16
+        if self.wide and (device != "cuda" or mode != "framework"):
+            raise ValueError("wide records need device=cuda, mode=framework")
         W = engine.world_size
         # R must be pow2 for the GPU top-bits partitioner
         R = W * partitions_per_executor
@@ -66,7 +76,19 @@ class TeraSort:
         self.part = RangePartitioner.uniform(R)
         self.low_bits = 64 - (R - 1).bit_length()
         rank = engine.rank
-        if device == "cuda":
+        if self.wide:
+            import torch
+            g = torch.Generator(device="cuda").manual_seed(seed * 1000 + rank)
+            Wb = record_bytes
+            # record: [key u64 LE | keylo u16 | payload]; payload's first
+            # 8 bytes mirror the prefix for integrity validation
+            recs = torch.randint(-128, 128, (self.n, Wb), dtype=torch.int8,
+                                 device="cuda", generator=g)
+            prefix = recs[:, :8]
+            recs[:, 10:18] = prefix   # integrity mirror
+            self.recs = recs.view(torch.uint8).reshape(-1)
+            self.keys = self.vals = None
+        elif device == "cuda":
             import torch
             g = torch.Generator(device="cuda").manual_seed(seed * 1000 + rank)
             self.keys = torch.randint(-2**63, 2**63 - 1, (self.n,),
@@ -129,7 +151,9 @@ class TeraSort:
         t0 = time.perf_counter()
         handle = eng.register_shuffle(eng.world_size, self.R)
         w = eng.manager.get_writer(handle, rank)
-        if self.device == "cuda":
+        if self.wide:
+            w.write_device_records(self.recs, self.RECORD_BYTES, 10)
+        elif self.device == "cuda":
             w.write_device_batch(self.keys, self.vals)
         else:
             w.write_batch(self.keys, self.vals)
@@ -160,7 +184,20 @@ class TeraSort:
             parts = reader.collect_partitions()
             ts_ = time.perf_counter()
             arena = getattr(reader.fetcher, "arena", None)
-            if self.device == "cuda" and arena is not None:
+            if self.wide and arena is not None:
+                import torch
+                from ..ops.radix import sort_records
+                nrec = arena.numel() // self.RECORD_BYTES
+                out = sort_records(
+                    arena, self.RECORD_BYTES, key_bytes=10,
+                    end_bit=64 - chunk_shared_bits,
+                    out=self._rec_out(arena.numel()),
+                    pairs=self._sort_tmp(2 * nrec, "_pairs_cache"),
+                    tmp=self._sort_tmp(2 * nrec, "_tmp_cache"),
+                    ws=self._sort_ws())
+                torch.cuda.synchronize()
+                outs.append(out)
+            elif self.device == "cuda" and arena is not None:
                 # fetches landed pre-placed in one device buffer: sort it
                 # directly (no concat pass); tmp/ws persist across steps
                 # so a 40 GB step performs no large allocations at all
@@ -201,20 +238,29 @@ class TeraSort:
                                  for _ in range(H)]
         return self._arena_cache
 
-    def _sort_tmp(self, n_i64: int):
+    def _sort_tmp(self, n_i64: int, attr: str = "_tmp_cache"):
         import torch
-        cur = getattr(self, "_tmp_cache", None)
+        cur = getattr(self, attr, None)
         if cur is None or cur.numel() < n_i64:
-            self._tmp_cache = torch.empty(int(n_i64 * 1.05) + 1024,
-                                          dtype=torch.int64, device="cuda")
-        return self._tmp_cache
+            setattr(self, attr, torch.empty(int(n_i64 * 1.05) + 1024,
+                                            dtype=torch.int64, device="cuda"))
+        return getattr(self, attr)
+
+    def _rec_out(self, nbytes: int):
+        import torch
+        cur = getattr(self, "_rec_out_cache", None)
+        if cur is None or cur.numel() < nbytes:
+            self._rec_out_cache = torch.empty(int(nbytes * 1.05) + 4096,
+                                              dtype=torch.uint8,
+                                              device="cuda")
+        return self._rec_out_cache
 
     def _sort_ws(self):
         from ..ops import load
         import torch
         cur = getattr(self, "_ws_cache", None)
         need = load().onesweep_workspace_bytes(
-            int(self.n * 1.25) + 4096, 8)
+            int(self.n * 1.25) + 4096, 10)
         if cur is None or cur.numel() < need:
             self._ws_cache = torch.empty(need, dtype=torch.uint8,
                                          device="cuda")
@@ -257,6 +303,8 @@ class TeraSort:
         return k[order], v[order]
 
     def _validate(self, sorted_out, lo: int, span: Optional[int] = None) -> None:
+        if self.wide:
+            return self._validate_wide(sorted_out, lo, span)
         k, v = sorted_out
         if k is None:
             return
@@ -271,6 +319,25 @@ class TeraSort:
         hi = lo + (span or self.ppe) - 1
         assert np.all((pids >= lo) & (pids <= hi)), "foreign keys in range"
         log.info("validated %d records in partitions [%d, %d]", len(ku), lo, hi)
+
+    def _validate_wide(self, out, lo: int, span: Optional[int] = None) -> None:
+        """Wide records: 80-bit key order (prefix, then u16 low bits),
+        payload integrity via the mirrored prefix, partition membership."""
+        W = self.RECORD_BYTES
+        arr = out.cpu().numpy().reshape(-1, W)
+        prefix = arr[:, :8].copy().view("<u8").ravel()
+        lo16 = arr[:, 8:10].copy().view("<u2").ravel()
+        mirror = arr[:, 10:18].copy().view("<u8").ravel()
+        assert np.all(prefix[1:] >= prefix[:-1]), "prefixes not sorted"
+        ties = prefix[1:] == prefix[:-1]
+        assert np.all(lo16[1:][ties] >= lo16[:-1][ties]), \
+            "low key bits not sorted within prefix ties"
+        assert np.array_equal(mirror, prefix), "payload corrupted"
+        pids = self.part.partition_ids(prefix)
+        hi = lo + (span or self.ppe) - 1
+        assert np.all((pids >= lo) & (pids <= hi)), "foreign keys in range"
+        log.info("validated %d wide records in partitions [%d, %d]",
+                 len(prefix), lo, hi)
 
     # ------------------------------------------------------------------
 

@@ -94,6 +94,28 @@ class ShuffleWriter:
         self._gpu_batches = getattr(self, "_gpu_batches", [])
         self._gpu_batches.append((keys, values))
 
+    def write_device_records(self, records, record_bytes: int,
+                             key_bytes: int = 8) -> None:
+        """Wide-record GPU path (canonical TeraSort: 100-byte records =
+        10 B key + 90 B value). ``records`` is a flat uint8 CUDA tensor of
+        n*record_bytes; key layout: u64 LE prefix at offset 0 (+ u16 LE
+        low bits at offset 8 when key_bytes == 10). The reference's
+        contract is 'serve whatever bytes the writer produced'
+        (RdmaMappedFile.java:113-157) — this is the GPU fast path for it
+        (r01 handled only 16-byte records)."""
+        if record_bytes % 4 or record_bytes < 8:
+            raise ValueError("record_bytes must be a multiple of 4, >= 8")
+        if key_bytes not in (8, 10):
+            raise ValueError("key_bytes must be 8 or 10")
+        self._gpu_records = getattr(self, "_gpu_records", [])
+        if records.numel() % record_bytes:
+            raise ValueError("records length not a record multiple")
+        prev = getattr(self, "_record_shape", None)
+        if prev is not None and prev != (record_bytes, key_bytes):
+            raise ValueError("record shape changed between batches")
+        self._record_shape = (record_bytes, key_bytes)
+        self._gpu_records.append(records)
+
     # -- arbitrary-record path -----------------------------------------
 
     def write_records(self, records, partitioner=None) -> None:
@@ -134,7 +156,9 @@ class ShuffleWriter:
         if not success:
             return
         t0 = time.perf_counter_ns()
-        if getattr(self, "_gpu_batches", None):
+        if getattr(self, "_gpu_records", None):
+            self._commit_gpu_records(partitioner)
+        elif getattr(self, "_gpu_batches", None):
             self._commit_gpu(partitioner)
         else:
             if self._byte_records is not None:
@@ -167,6 +191,28 @@ class ShuffleWriter:
                     vals_sorted[starts[p]:ends[p]] if vals_sorted is not None else None)
                 for p in range(R)]
 
+    def _gpu_part_params(self, partitioner):
+        """(func, shift, nparts, nbits_eff) for the kernel-side partition
+        function — any R <= 4096 (pow2 via bit extraction, arbitrary via
+        mulhi-range / hash-mod; partitioner.gpu_params())."""
+        R = self.handle.num_partitions
+        nbits = max((R - 1).bit_length(), 1)
+        if nbits > 12:
+            raise ValueError(
+                f"GPU partitioner supports R <= 4096 in one pass, got {R}; "
+                "use the CPU write path for larger partition counts")
+        params = None
+        if hasattr(partitioner, "gpu_params"):
+            params = partitioner.gpu_params()
+        elif hasattr(partitioner, "gpu_shift"):
+            params = (0, partitioner.gpu_shift, 0)
+        if params is None:
+            raise ValueError(
+                f"partitioner {partitioner!r} has no GPU dispatch "
+                "(gpu_params() returned None) — use the CPU write path")
+        func, shift, nparts = params
+        return func, shift, nparts, max(nbits, 4)
+
     def _commit_gpu(self, partitioner) -> None:
         """Map-side GPU write: one radix pass whose scatter writes each
         partition's [keys|vals] segment STRAIGHT into its final position in
@@ -179,15 +225,7 @@ class ShuffleWriter:
             raise RuntimeError("GPU writer requires the GPU data plane")
         hs = ops_load()
         R = self.handle.num_partitions
-        nbits = max((R - 1).bit_length(), 1)
-        if (1 << nbits) != R or nbits > 12:
-            raise ValueError(
-                f"GPU partitioner requires pow2 partitions <= 4096, got {R}")
-        shift = getattr(partitioner, "gpu_shift", 64 - nbits)
-        hash_mix = int(getattr(partitioner, "gpu_hash", False))
-        nbits_eff = max(nbits, 4)  # kernel instantiations start at 4 bits
-        if hash_mix and nbits_eff != nbits:
-            raise ValueError("GPU hash partitioning requires >= 16 partitions")
+        func, shift, nparts, nbits_eff = self._gpu_part_params(partitioner)
         batches = self._gpu_batches
         keys = (batches[0][0] if len(batches) == 1
                 else torch.cat([b[0] for b in batches]))
@@ -204,7 +242,7 @@ class ShuffleWriter:
                               dtype=torch.int32, device=dev)
         totals = torch.empty(nd, dtype=torch.int32, device=dev)
         hs.radix_hist(keys.data_ptr(), n, shift, nbits_eff, hist.data_ptr(),
-                      stream, hash_mix)
+                      stream, func, 1, nparts)
         hs.radix_scan(hist.data_ptr(), n, nbits_eff, totals.data_ptr(),
                       scan_ws.data_ptr(), stream)
         counts = totals.cpu().numpy().astype(np.int64)[:R]  # syncs the stream
@@ -279,8 +317,138 @@ class ShuffleWriter:
         hs.radix_scatter(keys.data_ptr(),
                          vals.data_ptr() if has_val else 0,
                          n, shift, nbits_eff, hist.data_ptr(),
-                         kd.data_ptr(), vd.data_ptr(), stream, hash_mix,
-                         1 if has_val else 0)
+                         kd.data_ptr(), vd.data_ptr(), stream, func,
+                         1 if has_val else 0, 1, nparts)
+        torch.cuda.synchronize()
+        if spill_groups:
+            soff = 0
+            for parts, total in spill_groups:
+                hblk = mgr.pool.get(total)     # host pool
+                hseg = mgr.data_segment(hblk.segment_id)
+                hkey = make_key(mgr.executor_id, hblk.segment_id)
+                cpu_bytes = spill_stage[soff:soff + total].cpu().numpy()
+                hseg.write(hblk.offset, cpu_bytes.tobytes())
+                off = hblk.offset
+                for p in parts:
+                    nb = int(seg_bytes[p])
+                    table.put(p, off, nb, hkey)
+                    off += nb
+                    self.metrics.bytes_written += nb
+                soff += total
+                blocks.append(hblk)
+        self.metrics.records_written += n
+        mgr.keep_alive(self.handle, self.map_id, blocks)
+        mgr.publish_map_output(self.handle, self.map_id, table_addr)
+
+    def _commit_gpu_records(self, partitioner) -> None:
+        """Wide-record map-side GPU write. Records stay put while 16-byte
+        (key-prefix, aux) pairs run the radix machinery; ONE gather pass
+        then moves each W-byte record straight to its final HBM position
+        (ops/csrc/kernels.hip 'wide-record machinery'). Replaces the
+        reference's CPU writer + mmap + register for arbitrary-width
+        records (RdmaMappedFile.java:113-189)."""
+        import torch
+        from .ops import load as ops_load
+        mgr = self.manager
+        if mgr.gpu is None:
+            raise RuntimeError("GPU writer requires the GPU data plane")
+        hs = ops_load()
+        R = self.handle.num_partitions
+        func, shift, nparts, nbits_eff = self._gpu_part_params(partitioner)
+        W, key_bytes = self._record_shape
+        batches = self._gpu_records
+        recs = (batches[0] if len(batches) == 1 else torch.cat(batches))
+        n = recs.numel() // W
+        dev = recs.device
+        stream = torch.cuda.current_stream().cuda_stream
+        nd = 1 << nbits_eff
+        pairs = torch.empty(2 * n, dtype=torch.int64, device=dev)
+        hs.extract_pairs(recs.data_ptr(), n, W, key_bytes, pairs.data_ptr(),
+                         stream)
+        hist = torch.empty(hs.radix_hist_bytes(n, nbits_eff) // 4,
+                           dtype=torch.int32, device=dev)
+        scan_ws = torch.empty(hs.radix_scan_ws_bytes(n, nbits_eff) // 4,
+                              dtype=torch.int32, device=dev)
+        totals = torch.empty(nd, dtype=torch.int32, device=dev)
+        hs.radix_hist(pairs.data_ptr(), n, shift, nbits_eff, hist.data_ptr(),
+                      stream, func, 2, nparts)
+        hs.radix_scan(hist.data_ptr(), n, nbits_eff, totals.data_ptr(),
+                      scan_ws.data_ptr(), stream)
+        counts_nd = totals.cpu().numpy().astype(np.int64)  # syncs the stream
+        counts = counts_nd[:R]
+        seg_bytes = counts * W
+
+        # greedy chunking of partitions into HBM blocks (same policy as
+        # the 16-byte path / RdmaMappedFile.java:113-157)
+        table, table_addr = mgr.alloc_table(R)
+        write_block = mgr.conf.shuffle_write_block_size
+        pool = mgr.gpu.pool
+        blocks = []
+        rec_dst = np.zeros(nd, dtype=np.int64)    # byte base per digit
+        group, group_bytes = [], 0
+        flushes = []
+        for p in range(R):
+            if group and group_bytes + seg_bytes[p] > write_block:
+                flushes.append(group)
+                group, group_bytes = [], 0
+            group.append(p)
+            group_bytes += int(seg_bytes[p])
+        if group:
+            flushes.append(group)
+        meta_key = make_key(mgr.executor_id, 1)
+        spill_groups = []   # (parts, total) that did not fit the HBM pool
+        for parts in flushes:
+            total = int(sum(seg_bytes[p] for p in parts))
+            if total == 0:
+                for p in parts:
+                    table.put(p, 0, 0, meta_key)
+                continue
+            try:
+                blk = pool.get(total)
+            except MemoryError:
+                spill_groups.append((parts, total))
+                continue
+            base = mgr.gpu.local_base(blk.segment_id)
+            key = make_key(mgr.executor_id, blk.segment_id)
+            off = blk.offset
+            for p in parts:
+                nb = int(seg_bytes[p])
+                table.put(p, off, nb, key)
+                if nb:
+                    rec_dst[p] = base + off
+                off += nb
+                self.metrics.bytes_written += nb
+            blocks.append(blk)
+        spill_stage = None
+        if spill_groups:
+            stage_total = sum(t for _, t in spill_groups)
+            spill_stage = torch.empty(stage_total, dtype=torch.uint8,
+                                      device=dev)
+            soff = 0
+            for parts, total in spill_groups:
+                for p in parts:
+                    nb = int(seg_bytes[p])
+                    if nb:
+                        rec_dst[p] = spill_stage.data_ptr() + soff
+                    soff += nb
+
+        # group the PAIRS by digit (contiguous per-digit runs)
+        starts = np.zeros(nd, dtype=np.int64)
+        np.cumsum(counts_nd[:-1], out=starts[1:])
+        pairs_out = torch.empty_like(pairs)
+        bases_t = torch.from_numpy(starts).to(dev)
+        kd = pairs_out.data_ptr() + bases_t * 16
+        vd = kd + 8
+        hs.radix_scatter(pairs.data_ptr(), pairs.data_ptr() + 8, n, shift,
+                         nbits_eff, hist.data_ptr(), kd.data_ptr(),
+                         vd.data_ptr(), stream, func, 1, 2, nparts)
+        # one gather moves every record to its final slot
+        dstart_t = torch.from_numpy(starts.astype(np.uint32)
+                                    .view(np.int32)).to(dev)
+        dst_addr_t = torch.from_numpy(rec_dst).to(dev)
+        hs.gather_records(recs.data_ptr(), pairs_out.data_ptr(), n, W, 1, 0,
+                          dst_addr_t.data_ptr(), dstart_t.data_ptr(), shift,
+                          nd - 1, func, nparts, stream)
         torch.cuda.synchronize()
         if spill_groups:
             soff = 0
