@@ -514,7 +514,14 @@ class FetcherIterator:
 
 
 class ShuffleReader:
-    """Public reader: iterate raw blocks or aggregate per partition."""
+    """Public reader: iterate raw blocks, or consume via the GENERIC
+    aggregation/ordering hookup — the role the reference's reader plays
+    once (deserialize -> aggregate -> ExternalSorter ordering,
+    RdmaShuffleReader.scala:61-114) instead of every workload
+    re-implementing it."""
+
+    #: named reductions for read_aos(aggregator=...)
+    AGGREGATORS = ("sum", "sum_f64", "count", "min", "max")
 
     def __init__(self, manager: ShuffleManager, handle: ShuffleHandle,
                  start_partition: int, end_partition: int, arena=None):
@@ -535,6 +542,143 @@ class ShuffleReader:
         for ref, data in self.fetcher:
             out[ref.partition].append(data)
         return out
+
+    # ---------------- generic consumption (fixed-width AoS records) ----
+
+    def _drain_aos_pairs(self):
+        """CPU lane: fetch everything, return (keys u64, values u64)
+        numpy arrays (the GPU lane drains into the fetch arena instead)."""
+        from .writer import unpack_partition_segment
+        import numpy as np
+        ks, vs = [], []
+        for _ref, data in self.fetcher:
+            k, v = unpack_partition_segment(data, 8)
+            ks.append(np.array(k))
+            vs.append(np.ascontiguousarray(v).reshape(-1, 8)
+                      .view(np.uint64).reshape(-1))
+        if not ks:
+            return None, None
+        return np.concatenate(ks), np.concatenate(vs)
+
+    def read_aos(self, aggregator: Optional[str] = None,
+                 ordering: bool = False, end_bit: int = 64,
+                 sort_tmp=None, sort_ws=None):
+        """Consume the whole partition range of AoS (u64 key, u64 value)
+        records, applying an optional keyed reduction and/or key ordering
+        ON DEVICE. Returns (keys, values):
+
+        * aggregator=None, ordering=True  -> records sorted by key bits
+          [0, end_bit) (callers whose partitions share top bits pass a
+          smaller end_bit and save radix passes)
+        * aggregator in AGGREGATORS -> one row per distinct key, keys
+          ascending; "sum_f64" treats the value payload as float64
+        * both None/False -> unordered concatenated records
+        """
+        if aggregator is not None and aggregator not in self.AGGREGATORS:
+            raise ValueError(f"unknown aggregator {aggregator!r}; "
+                             f"expected one of {self.AGGREGATORS}")
+        mgr = self.manager
+        if mgr.gpu is not None:
+            import torch
+            for _ in self.fetcher:
+                pass
+            arena = self.fetcher.arena
+            pairs = arena.view(torch.int64)
+            if pairs.numel() == 0:
+                empty = torch.empty(0, dtype=torch.int64,
+                                    device=f"cuda:{mgr.gpu.device}")
+                return empty, empty.clone()
+            if aggregator is not None or ordering:
+                from .ops.radix import sort_pairs_aos
+                pairs = sort_pairs_aos(pairs, 0, end_bit, tmp=sort_tmp,
+                                       ws=sort_ws)
+            k = pairs[0::2]
+            v = pairs[1::2]
+            if aggregator is None:
+                return k, v
+            k = k.contiguous()
+            v = v.contiguous()
+            uk, inverse, cnt = torch.unique_consecutive(
+                k, return_inverse=True, return_counts=True)
+            if aggregator == "count":
+                return uk, cnt
+            if aggregator == "sum":
+                out = torch.zeros(uk.numel(), dtype=torch.int64,
+                                  device=k.device)
+                out.index_add_(0, inverse, v)
+                return uk, out
+            if aggregator == "sum_f64":
+                out = torch.zeros(uk.numel(), dtype=torch.float64,
+                                  device=k.device)
+                out.index_add_(0, inverse, v.view(torch.float64))
+                return uk, out
+            red = "amin" if aggregator == "min" else "amax"
+            out = torch.empty(uk.numel(), dtype=torch.int64, device=k.device)
+            out.scatter_reduce_(0, inverse, v, reduce=red,
+                                include_self=False)
+            return uk, out
+        # ---- CPU path: numpy mirror of the device semantics ----
+        import numpy as np
+        k, v = self._drain_aos_pairs()
+        if k is None:
+            return np.array([], dtype=np.uint64), np.array([], dtype=np.uint64)
+        if aggregator is not None or ordering:
+            mask = np.uint64((1 << end_bit) - 1) if end_bit < 64 \
+                else np.uint64(2 ** 64 - 1)
+            order = np.argsort(k & mask, kind="stable")
+            k, v = k[order], v[order]
+        if aggregator is None:
+            return k, v
+        uk, start = np.unique(k, return_index=True)
+        if aggregator == "count":
+            return uk, np.diff(np.append(start, len(k)))
+        if aggregator == "sum":
+            return uk, np.add.reduceat(v, start)
+        if aggregator == "sum_f64":
+            return uk, np.add.reduceat(v.view(np.float64), start)
+        fn = np.minimum if aggregator == "min" else np.maximum
+        return uk, fn.reduceat(v, start)
+
+    def dense_sum(self, key_lo: int, span: int, dtype: str = "f64"):
+        """Keyed sum into a DENSE [key_lo, key_lo+span) vector without a
+        sort — the iterative-workload (PageRank) reduction: every fetched
+        chunk is index-added as it arrives, overlapping with in-flight
+        fetches. dtype 'f64' treats payloads as float64, 'i64' as int64."""
+        mgr = self.manager
+        if mgr.gpu is not None:
+            import torch
+            tdtype = torch.float64 if dtype == "f64" else torch.int64
+            sums = torch.zeros(span, dtype=tdtype,
+                               device=f"cuda:{mgr.gpu.device}")
+            from .utils import as_device_i64
+            for _ref, data in self.fetcher:
+                t = as_device_i64(data)
+                idx = t[0::2] - key_lo
+                val = t[1::2].contiguous()
+                sums.index_add_(0, idx, val.view(tdtype)
+                                if dtype == "f64" else val)
+            return sums
+        import numpy as np
+        from .writer import unpack_partition_segment
+        ndtype = np.float64 if dtype == "f64" else np.int64
+        sums = np.zeros(span, dtype=ndtype)
+        for _ref, data in self.fetcher:
+            k, v = unpack_partition_segment(data, 8)
+            idx = (np.asarray(k) - key_lo).astype(np.int64)
+            np.add.at(sums, idx, np.ascontiguousarray(v).reshape(-1, 8)
+                      .view(ndtype).reshape(-1))
+        return sums
+
+    def read_records(self):
+        """Pickled-record lane: yields (key, value) python objects —
+        the deserialize role of RdmaShuffleReader.scala:61-77."""
+        import io
+        import pickle
+        for _ref, data in self.fetcher:
+            bio = io.BytesIO(bytes(data))
+            end = len(bio.getvalue())
+            while bio.tell() < end:
+                yield pickle.load(bio)
 
     @property
     def metrics(self) -> TaskMetrics:

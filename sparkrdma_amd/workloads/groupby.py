@@ -9,8 +9,6 @@ reference's production validation workload shape.
 
 from __future__ import annotations
 
-import io
-import pickle
 import time
 from dataclasses import dataclass
 
@@ -49,10 +47,8 @@ class GroupByKey:
         lo, hi = eng.rank * per, (eng.rank + 1) * per - 1
         reader = eng.manager.get_reader(handle, lo, hi)
         groups = {}
-        for ref, data in reader:
-            bio = io.BytesIO(bytes(data))
-            while bio.tell() < len(bio.getvalue()):
-                k, v = pickle.load(bio)
-                groups.setdefault(k, []).append(v)
+        # reader-side generic deserialize (read_records) + group-by
+        for k, v in reader.read_records():
+            groups.setdefault(k, []).append(v)
         eng.unregister_shuffle(handle)
         return GroupByResult(time.perf_counter() - t0, self.n, len(groups))

@@ -21,7 +21,6 @@ import numpy as np
 
 from ..engine import Engine
 from ..partitioner import RangePartitioner
-from ..writer import unpack_partition_segment
 
 DAMPING = 0.85
 
@@ -117,28 +116,18 @@ class PageRank:
                 arena = self._arena_cache
         reader = eng.manager.get_reader(handle, lo, hi, arena=arena)
         span = self.own_hi - self.own_lo
+        # the reader's generic dense keyed-sum (chunks index-add as they
+        # arrive, overlapping in-flight fetches) — shared with any
+        # iterative workload instead of a per-workload loop
+        sums = reader.dense_sum(self.own_lo, span, dtype="f64")
+        self.ranks = (1.0 - DAMPING) / self.V + DAMPING * sums
         if self.device == "cuda":
             import torch
-            sums = torch.zeros(span, dtype=torch.float64, device="cuda")
-            from ..utils import as_device_i64
-            for ref, data in reader:
-                t = as_device_i64(data)  # AoS (dst, contrib) records
-                idx = t[0::2] - self.own_lo
-                sums.index_add_(0, idx, t[1::2].contiguous().view(torch.float64))
-            self.ranks = (1.0 - DAMPING) / self.V + DAMPING * sums
             torch.cuda.synchronize()
             if dbg:
                 import sys
                 print(f"[pr-iter] write+fetch+agg={time.perf_counter()-t0:.3f}s",
                       file=sys.stderr)
-        else:
-            sums = np.zeros(span, dtype=np.float64)
-            for ref, data in reader:
-                k, v = unpack_partition_segment(data, 8)
-                idx = (np.asarray(k) - self.own_lo).astype(np.int64)
-                np.add.at(sums, idx, np.ascontiguousarray(v).reshape(-1, 8)
-                          .view(np.float64).reshape(-1))
-            self.ranks = (1.0 - DAMPING) / self.V + DAMPING * sums
         eng.unregister_shuffle(handle)
         return reader.metrics.remote_bytes_read + reader.metrics.local_bytes_read
 

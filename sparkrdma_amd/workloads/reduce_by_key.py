@@ -18,7 +18,6 @@ import numpy as np
 
 from ..engine import Engine
 from ..partitioner import HashPartitioner
-from ..writer import unpack_partition_segment
 
 
 @dataclass
@@ -71,12 +70,14 @@ class ReduceByKey:
         eng.barrier()
         lo, hi = eng.rank * self.ppe, (eng.rank + 1) * self.ppe - 1
         reader = eng.manager.get_reader(handle, lo, hi)
-        parts = reader.collect_partitions()
+        # the reader's GENERIC aggregate hookup (RdmaShuffleReader.scala:
+        # 61-114 role): sort by key + segmented sum, device-side on GPU
+        uk, sums = reader.read_aos(aggregator="sum")
         if self.device == "cuda":
-            uk, sums = self._aggregate_gpu(parts, reader)
+            import torch
+            torch.cuda.synchronize()
             groups = int(uk.numel())
         else:
-            uk, sums = self._aggregate_cpu(parts)
             groups = len(uk)
         if self.validate:
             self._validate(uk, sums)
@@ -84,47 +85,6 @@ class ReduceByKey:
         return ReduceByKeyResult(
             time.perf_counter() - t0, self.n, groups,
             reader.metrics.remote_bytes_read + reader.metrics.local_bytes_read)
-
-    def _aggregate_gpu(self, parts, reader):
-        import torch
-        from ..ops.radix import sort_pairs_aos
-        from ..utils import as_device_i64
-        arena = getattr(reader.fetcher, "arena", None)
-        if arena is not None:
-            pairs = arena.view(torch.int64)
-        else:
-            ts = [as_device_i64(c) for chunks in parts.values()
-                  for c in chunks]
-            if not ts:
-                return torch.empty(0, dtype=torch.int64, device="cuda"), None
-            pairs = torch.cat(ts) if len(ts) > 1 else ts[0].contiguous()
-        # sort by full key, then sum each equal-key run (segmented reduce)
-        pairs = sort_pairs_aos(pairs, 0, 64)
-        k = pairs[0::2].contiguous()
-        v = pairs[1::2].contiguous()
-        uk, inverse = torch.unique_consecutive(k, return_inverse=True)
-        sums = torch.zeros(uk.numel(), dtype=torch.int64, device="cuda")
-        sums.index_add_(0, inverse, v)
-        torch.cuda.synchronize()
-        return uk, sums
-
-    def _aggregate_cpu(self, parts):
-        ks, vs = [], []
-        for chunks in parts.values():
-            for c in chunks:
-                k, v = unpack_partition_segment(c, 8)
-                ks.append(np.array(k))
-                vs.append(np.ascontiguousarray(v).reshape(-1, 8)
-                          .view(np.uint64).reshape(-1))
-        if not ks:
-            return np.array([], dtype=np.uint64), np.array([], dtype=np.uint64)
-        k = np.concatenate(ks)
-        v = np.concatenate(vs)
-        order = np.argsort(k, kind="stable")
-        k, v = k[order], v[order]
-        uk, start = np.unique(k, return_index=True)
-        sums = np.add.reduceat(v, start)
-        return uk, sums
 
     def _validate(self, uk, sums) -> None:
         """Single-rank oracle: this rank's groups must equal the dict-based
