@@ -95,6 +95,10 @@ class ShuffleConf:
     shm_dir: str = "/dev/shm"
     read_requests_limit: int = 0      # 0 = send_queue_depth // cores
     executor_cores: int = 1
+    # cross-host TCP lane (the one place a codec pays — PARITY.md): the
+    # intra-node xGMI/shm paths never compress
+    tcp_compress: bool = False        # zlib-1 per chunk
+    tcp_chunk_size: int = 4 << 20     # streaming chunk (pinned D2H unit)
 
     def __post_init__(self) -> None:
         self._validate_range("recv_queue_depth", self.recv_queue_depth, 16, 1 << 20)
@@ -149,6 +153,8 @@ class ShuffleConf:
         "shmDir": ("shm_dir", str),
         "rdmaReadRequestsLimit": ("read_requests_limit", int),
         "executorCores": ("executor_cores", int),
+        "tcpCompress": ("tcp_compress", None),
+        "tcpChunkSize": ("tcp_chunk_size", parse_bytes),
     }
 
     @classmethod

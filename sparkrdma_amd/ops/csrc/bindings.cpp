@@ -39,6 +39,8 @@ PYBIND11_MODULE(_hipshuffle, m) {
         py::call_guard<py::gil_scoped_release>());
   m.def("wait_event", &hs::wait_event,
         py::call_guard<py::gil_scoped_release>());
+  m.def("host_alloc_pinned", &hs::host_alloc_pinned);
+  m.def("host_free_pinned", &hs::host_free_pinned);
   m.def("memcpy_h2d", &hs::memcpy_h2d,
         py::call_guard<py::gil_scoped_release>());
   m.def("memcpy_d2h", &hs::memcpy_d2h,

@@ -21,6 +21,8 @@ uint64_t read_batch_ids(int peer, const std::vector<uintptr_t>& dsts,
                         const std::vector<size_t>& sizes);
 bool poll_event(uint64_t id);
 void wait_event(uint64_t id);
+uintptr_t host_alloc_pinned(size_t n);
+void host_free_pinned(uintptr_t p);
 void memcpy_h2d(uintptr_t dst, uintptr_t src, size_t n);
 void memcpy_d2h(uintptr_t dst, uintptr_t src, size_t n);
 

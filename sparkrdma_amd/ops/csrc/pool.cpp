@@ -144,6 +144,8 @@ class CopyEngine {
   }
 
   // Issue a batch of copies; returns an event id to poll.
+  // hipMemcpyDefault: the pointers select the direction, so the same
+  // engine serves xGMI D2D reads AND pinned-host staging (TCP lane).
   uint64_t read_batch(int peer, const std::vector<uintptr_t>& dsts,
                       const std::vector<uintptr_t>& srcs,
                       const std::vector<size_t>& sizes) {
@@ -151,7 +153,7 @@ class CopyEngine {
     for (size_t i = 0; i < dsts.size(); ++i) {
       HIP_CHECK(hipMemcpyAsync(reinterpret_cast<void*>(dsts[i]),
                                reinterpret_cast<void*>(srcs[i]), sizes[i],
-                               hipMemcpyDeviceToDevice, s));
+                               hipMemcpyDefault, s));
     }
     hipEvent_t ev = acquire_event();
     HIP_CHECK(hipEventRecord(ev, s));
@@ -233,6 +235,17 @@ uint64_t read_batch_ids(int peer, const std::vector<uintptr_t>& dsts,
 }
 bool poll_event(uint64_t id) { return copy_engine().poll(id); }
 void wait_event(uint64_t id) { return copy_engine().wait(id); }
+
+// pinned host memory for the TCP lane's double-buffered D2H staging
+uintptr_t host_alloc_pinned(size_t n) {
+  void* p = nullptr;
+  HIP_CHECK(hipHostMalloc(&p, n, hipHostMallocDefault));
+  return reinterpret_cast<uintptr_t>(p);
+}
+
+void host_free_pinned(uintptr_t p) {
+  HIP_CHECK(hipHostFree(reinterpret_cast<void*>(p)));
+}
 
 // host<->device staging helpers (bytes path / spill)
 void memcpy_h2d(uintptr_t dst, uintptr_t src, size_t n) {
