@@ -89,6 +89,10 @@ class DataServer:
         pinned = None
         try:
             sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+            try:
+                sock.setsockopt(socket.SOL_SOCKET, socket.SO_SNDBUF, 8 << 20)
+            except OSError:
+                pass
             while not self._stopped.is_set():
                 hdr = self._recv_exact(sock, _REQ.size)
                 if hdr is None:
@@ -158,7 +162,9 @@ class DataServer:
     def _stream_hbm(self, sock, seg_id: int, addr: int, length: int,
                     pinned):
         """Device blocks: async D2H into pinned buffer B while buffer A's
-        bytes are on the socket — the copy/send pipeline."""
+        bytes are on the socket — the copy/send pipeline. Sends are
+        ZERO-COPY views over the pinned buffers (safe: the next reuse of
+        a buffer waits on its event only after the send returned)."""
         import ctypes
         gpu = self.manager.gpu
         hs = gpu.hs
@@ -171,22 +177,22 @@ class DataServer:
         slot = 63   # dedicated staging stream slot in the copy engine
         offs = list(range(0, length, self._chunk))
         evs = [None, None]
+
+        def send_slot(idx, nbytes):
+            hs.wait_event(evs[idx])
+            view = (ctypes.c_char * nbytes).from_address(pinned.ptrs[idx])
+            self._send_chunk(sock, view)
+
         for i, off in enumerate(offs):
             c = min(self._chunk, length - off)
             evs[i % 2] = hs.read_batch(slot, [pinned.ptrs[i % 2]],
                                        [base + addr + off], [c])
             if i > 0:
-                prev_off = offs[i - 1]
-                pc = min(self._chunk, length - prev_off)
-                hs.wait_event(evs[(i - 1) % 2])
-                raw = ctypes.string_at(pinned.ptrs[(i - 1) % 2], pc)
-                self._send_chunk(sock, raw)
+                pc = min(self._chunk, length - offs[i - 1])
+                send_slot((i - 1) % 2, pc)
         if offs:
             last = len(offs) - 1
-            c = min(self._chunk, length - offs[last])
-            hs.wait_event(evs[last % 2])
-            raw = ctypes.string_at(pinned.ptrs[last % 2], c)
-            self._send_chunk(sock, raw)
+            send_slot(last % 2, min(self._chunk, length - offs[last]))
         return pinned
 
     @staticmethod
@@ -233,6 +239,10 @@ class DataClient:
         try:
             sock = socket.create_connection(ep, timeout=30)
             sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+            try:
+                sock.setsockopt(socket.SOL_SOCKET, socket.SO_RCVBUF, 8 << 20)
+            except OSError:
+                pass
             return sock
         except BaseException:
             with self._cv:
@@ -266,26 +276,40 @@ class DataClient:
                 self._release(ep, sock, broken=False)
                 raise IOError(f"remote read failed (status {status})")
             out = bytearray(status)
+            mv = memoryview(out)
             off = 0
             while off < status:
                 chdr = DataServer._recv_exact(sock, _CHUNK.size)
                 if chdr is None:
                     raise ConnectionError("short chunk header")
                 raw_len, wire_len = _CHUNK.unpack(chdr)
-                wire = DataServer._recv_exact(sock, wire_len)
-                if wire is None:
-                    raise ConnectionError("short chunk payload")
-                if (flags & FLAG_DEFLATE) and wire_len != raw_len:
-                    wire = zlib.decompress(wire)
-                    if len(wire) != raw_len:
+                if wire_len == raw_len:
+                    # raw chunk: receive STRAIGHT into place (no copy)
+                    self._recv_into(sock, mv[off:off + raw_len])
+                else:
+                    wire = DataServer._recv_exact(sock, wire_len)
+                    if wire is None:
+                        raise ConnectionError("short chunk payload")
+                    dec = zlib.decompress(wire)
+                    if len(dec) != raw_len:
                         raise IOError("chunk decompress length mismatch")
-                out[off:off + raw_len] = wire
+                    mv[off:off + raw_len] = dec
                 off += raw_len
             self._release(ep, sock, broken=False)
             return bytes(out)
         except (OSError, ConnectionError):
             self._release(ep, sock, broken=True)
             raise
+
+    @staticmethod
+    def _recv_into(sock, view) -> None:
+        got = 0
+        n = len(view)
+        while got < n:
+            r = sock.recv_into(view[got:])
+            if r == 0:
+                raise ConnectionError("short chunk payload")
+            got += r
 
     def close(self) -> None:
         with self._cv:

@@ -65,7 +65,7 @@ def test_hash_mix_partition_matches_cpu():
     from sparkrdma_amd.partitioner import HashPartitioner
     n, R = 300_000, 64
     hp = HashPartitioner(R)
-    assert getattr(hp, "gpu_hash", False)
+    assert hp.gpu_params() == (1, 0, 0)   # hash-bits fast path for pow2 R
     rng = np.random.default_rng(9)
     k = rng.integers(0, 2 ** 63, n, dtype=np.uint64)
     keys = torch.from_numpy(k.view(np.int64)).cuda()
