@@ -173,3 +173,59 @@ def test_wide_writer_nonpow2_partitions(tmp_path):
     finally:
         mgr.stop()
         driver.stop()
+
+
+def _wide_ipc_worker(rank, world, driver_port, shm_dir, q):
+    try:
+        import os
+        import sys
+        sys.path.insert(0, os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))))
+        from sparkrdma_amd.conf import ShuffleConf
+        from sparkrdma_amd.engine import Engine
+        from sparkrdma_amd.workloads.terasort import TeraSort
+
+        conf = ShuffleConf(transport="ipc", hbm_pool_size=2 << 30,
+                           shm_dir=shm_dir, gpu_id=0)
+        eng = Engine(conf, rank=rank, world_size=world,
+                     driver_port=driver_port)
+        ts = TeraSort(eng, records_per_executor=300_000,
+                      partitions_per_executor=64, device="cuda",
+                      validate=True, record_bytes=100)
+        res = ts.run_step()
+        q.put((rank, res.records, res.remote_bytes))
+        eng.barrier()
+        eng.shutdown()
+    except BaseException as e:
+        import traceback
+        q.put((rank, f"ERROR: {e}\n{traceback.format_exc()}", 0))
+        raise
+
+
+def test_wide_cross_process_shuffle(tmp_path):
+    """Canonical 100-byte records across a process boundary (two
+    executors on one GPU, hipIpc one-sided fetch) — the shape the
+    multi-GPU driver bench runs."""
+    import multiprocessing as mp
+    import socket
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    procs = [ctx.Process(target=_wide_ipc_worker,
+                         args=(r, 2, port, str(tmp_path), q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    remote_total = 0
+    for _ in range(2):
+        rank, records, remote = q.get(timeout=300)
+        assert not isinstance(records, str), f"rank {rank}: {records}"
+        assert records == 300_000
+        remote_total += remote
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert remote_total > 0
