@@ -36,26 +36,43 @@ def main():
     sort_pairs_aos(pp, 0, 64, tmp=tmp, ws=ws)
     torch.cuda.synchronize()
 
-    m.set_timing_buf(timing.data_ptr())
-    pp = pairs.clone()
-    t0 = time.perf_counter()
-    sort_pairs_aos(pp, 0, 64, tmp=tmp, ws=ws)
-    torch.cuda.synchronize()
-    wall = time.perf_counter() - t0
-    m.set_timing_buf(0)
-
     nblocks = (n + 4095) // 4096
     passes = 8
-    t = timing.cpu().numpy() / REALTIME_MHZ / (nblocks * passes)  # us/blk
     names = ["phaseA(load+rank+scan+pub)", "exchange",
              "lookback walk (+barrier)", "writeout"]
-    print(f"n={n} blocks/pass={nblocks} passes={passes} "
-          f"wall={wall*1e3:.2f} ms ({wall/passes*1e3:.3f} ms/pass)")
-    for nm, v in zip(names, t):
-        print(f"  {nm:<28} {v:7.2f} us/block")
-    print(f"  {'total in-kernel':<28} {t.sum():7.2f} us/block "
-          f"(resident ~512 blocks => "
-          f"{t.sum()*nblocks/512/1e3:.2f} ms/pass serial-equivalent)")
+    ref = None
+    for lb_mode, label in ((0, "[nb][ND] (r01)"), (1, "transposed [ND][nb]")):
+        m.set_lookback_mode(lb_mode)
+        pp = pairs.clone()
+        sort_pairs_aos(pp, 0, 64, tmp=tmp, ws=ws)  # warm this mode
+        torch.cuda.synchronize()
+        timing.zero_()
+        m.set_timing_buf(timing.data_ptr())
+        pp = pairs.clone()
+        t0 = time.perf_counter()
+        out = sort_pairs_aos(pp, 0, 64, tmp=tmp, ws=ws)
+        torch.cuda.synchronize()
+        wall = time.perf_counter() - t0
+        m.set_timing_buf(0)
+        keys_sorted = out[0::2]
+        if ref is None:
+            ref = keys_sorted.clone()
+        else:
+            assert torch.equal(ref, keys_sorted), "modes disagree!"
+        # timed run again without the timing instrumentation
+        pp = pairs.clone()
+        t0 = time.perf_counter()
+        sort_pairs_aos(pp, 0, 64, tmp=tmp, ws=ws)
+        torch.cuda.synchronize()
+        wall_clean = time.perf_counter() - t0
+        t = timing.cpu().numpy() / REALTIME_MHZ / (nblocks * passes)
+        print(f"lb_mode={lb_mode} {label}: wall={wall_clean*1e3:.2f} ms "
+              f"({wall_clean/passes*1e3:.3f} ms/pass; instrumented "
+              f"{wall/passes*1e3:.3f})")
+        for nm, v in zip(names, t):
+            print(f"  {nm:<28} {v:7.2f} us/block")
+        print(f"  {'total in-kernel':<28} {t.sum():7.2f} us/block")
+    m.set_lookback_mode(0)
 
 
 if __name__ == "__main__":

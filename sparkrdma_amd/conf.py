@@ -75,7 +75,7 @@ class ShuffleConf:
     partition_location_fetch_timeout_ms: int = 120_000
     fetch_time_bucket_size_ms: int = 300
     fetch_time_num_buckets: int = 5
-    collect_odp_stats: bool = False
+    collect_odp_stats: bool = True   # reference default (no-op on HBM: no ODP)
 
     # --- control plane (reference :134-142) ---
     driver_host: str = "127.0.0.1"

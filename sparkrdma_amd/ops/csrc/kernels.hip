@@ -447,6 +447,35 @@ __device__ __forceinline__ void lookback_walk(
   pref[tid] = (uint32_t)run;
 }
 
+// TRANSPOSED-descriptor walk: desc laid out [ND][nb] so digit `tid`'s
+// predecessors are CONTIGUOUS backwards — 16 descriptors share one
+// 128-B line, so a completed prefix region costs nb/16 line fills
+// instead of nb (the r01 [nb][ND] layout made every probe its own
+// line). Publish is a scattered 8-B store per digit (one line each) —
+// worth it when walks span many predecessors (512-block resident
+// windows at 64M records).
+template <int ND>
+__device__ __forceinline__ void lookback_walk_t(
+    uint64_t* __restrict__ desc, uint32_t nb, uint32_t b, int tid,
+    uint64_t my_total, uint32_t* __restrict__ pref) {
+  uint64_t* row = desc + (uint64_t)tid * nb;
+  uint64_t run = 0;
+  for (int64_t j = (int64_t)b - 1; j >= 0;) {
+    uint64_t v = __hip_atomic_load(&row[j], __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+    if ((v & FLAG_MASK) == 0) {
+      __builtin_amdgcn_s_sleep(2);
+      continue;
+    }
+    run += v & VAL_MASK;
+    if ((v & FLAG_MASK) == FLAG_INC) break;
+    --j;
+  }
+  __hip_atomic_store(&row[b], FLAG_INC | (run + my_total),
+                     __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  pref[tid] = (uint32_t)run;
+}
+
 // Global digit totals of EVERY pass in one read (digit counts are
 // order-independent, so pass k's totals can be computed from pass 0's
 // input).
@@ -492,7 +521,8 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
     uint32_t* __restrict__ ticket,
     const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst,
     int stage = 3, uint64_t* __restrict__ timing = nullptr,
-    const uint32_t* __restrict__ hist_pref = nullptr, int sort_word = 0) {
+    const uint32_t* __restrict__ hist_pref = nullptr, int sort_word = 0,
+    int lb_mode = 0 /* 0: [nb][ND] descriptors, 1: transposed [ND][nb] */) {
   constexpr int ND = 1 << PBITS;
   extern __shared__ char smem_raw[];
   constexpr int TILE_T = BS * IT;
@@ -617,7 +647,9 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
     if (tid < ND) pref[tid] = hist_pref[(uint64_t)b * ND + tid];
   } else if (tid < ND) {
     my_total = start[tid];
-    uint64_t* slot = &desc[(uint64_t)b * ND + tid];
+    uint64_t* slot = lb_mode
+        ? &desc[(uint64_t)tid * gridDim.x + b]
+        : &desc[(uint64_t)b * ND + tid];
     if (stage == 0 || b == 0) {  // stage 0: ablation, WRONG results
       __hip_atomic_store(slot, FLAG_INC | my_total, __ATOMIC_RELAXED,
                          __HIP_MEMORY_SCOPE_AGENT);
@@ -645,8 +677,12 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
       }
     }
     uint64_t t2 = timing ? __builtin_amdgcn_s_memrealtime() : 0;
-    if (!hist_pref && stage != 0 && b != 0 && tid < ND)
-      lookback_walk<ND>(desc, b, tid, my_total, pref);
+    if (!hist_pref && stage != 0 && b != 0 && tid < ND) {
+      if (lb_mode)
+        lookback_walk_t<ND>(desc, gridDim.x, b, tid, my_total, pref);
+      else
+        lookback_walk<ND>(desc, b, tid, my_total, pref);
+    }
     if (stage < 2 && !timing) return;  // ablation: rank/publish/lookback
     __syncthreads();
     uint64_t t3 = timing ? __builtin_amdgcn_s_memrealtime() : 0;
@@ -671,8 +707,12 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
     }
     return;
   }
-  if (!hist_pref && b != 0 && tid < ND)
-    lookback_walk<ND>(desc, b, tid, my_total, pref);
+  if (!hist_pref && b != 0 && tid < ND) {
+    if (lb_mode)
+      lookback_walk_t<ND>(desc, gridDim.x, b, tid, my_total, pref);
+    else
+      lookback_walk<ND>(desc, b, tid, my_total, pref);
+  }
 
   // SoA path: key exchange + write-out, then val exchange + write-out
 #pragma unroll
@@ -1136,6 +1176,10 @@ void set_sort_mode(int m) { g_sort_mode = m; }
 // optional phase-timing accumulator: u64[4] = {phaseA, exch, lookback, writeout}
 static uint64_t* g_timing_buf = nullptr;
 void set_timing_buf(uintptr_t p) { g_timing_buf = reinterpret_cast<uint64_t*>(p); }
+// descriptor layout: 0 = [nb][ND] (r01), 1 = transposed [ND][nb] (walk
+// reads contiguous backwards — 16 descriptors/line)
+static int g_lb_mode = 0;
+void set_lookback_mode(int m) { g_lb_mode = m; }
 
 static inline uint32_t os_num_tiles_t(uint32_t n, int tile) {
   return (uint32_t)(((uint64_t)n + tile - 1) / tile);
@@ -1233,32 +1277,32 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          sb, desc, ticket, key_dst, val_dst, g_pass_stage,
-                         g_timing_buf, pass_pref, sort_word);
+                         g_timing_buf, pass_pref, sort_word, g_lb_mode);
     } else if (aos && aos_tile == 2048) {
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 4, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          sb, desc, ticket, key_dst, val_dst, g_pass_stage,
-                         g_timing_buf, pass_pref, sort_word);
+                         g_timing_buf, pass_pref, sort_word, g_lb_mode);
     } else if (aos) {
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 8, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          sb, desc, ticket, key_dst, val_dst, g_pass_stage,
-                         g_timing_buf, pass_pref, sort_word);
+                         g_timing_buf, pass_pref, sort_word, g_lb_mode);
     } else if (vals) {
       hipLaunchKernelGGL(
           (onesweep_pass_kernel<true, OS_ITEMS, false, BLOCK, PBITS>),
           dim3(nb), dim3(BLOCK), lds, s,
           reinterpret_cast<const uint64_t*>(src_k),
           reinterpret_cast<const uint64_t*>(src_v), n, sb, desc, ticket,
-          key_dst, val_dst);
+          key_dst, val_dst, 3, nullptr, nullptr, 0, g_lb_mode);
     } else {
       hipLaunchKernelGGL(
           (onesweep_pass_kernel<false, OS_ITEMS, false, BLOCK, PBITS>),
           dim3(nb), dim3(BLOCK), lds, s,
           reinterpret_cast<const uint64_t*>(src_k), nullptr, n, sb, desc,
-          ticket, key_dst, val_dst);
+          ticket, key_dst, val_dst, 3, nullptr, nullptr, 0, g_lb_mode);
     }
     HIP_CHECK(hipGetLastError());
     std::swap(src_k, dst_k);
