@@ -71,7 +71,58 @@ def run(chunk_mb=4, size_mb=512, reps=5, compress=False):
         driver.stop()
 
 
+def run_parallel(nthreads=4, chunk_mb=1, size_mb=256, reps=4):
+    """Aggregate lane bandwidth: the fetcher fans concurrent reads over
+    the client's pooled connections (DataClient.MAX_CONNS_PER_PEER)."""
+    import threading
+    conf = ShuffleConf(transport="tcp", tcp_chunk_size=chunk_mb << 20)
+    driver = Driver(conf)
+    m0 = ShuffleManager(conf, executor_id=0, driver_port=driver.port)
+    m1 = ShuffleManager(conf, executor_id=1, driver_port=driver.port)
+    try:
+        size = size_mb << 20
+        blks = [m1.pool.get(size) for _ in range(nthreads)]
+        keys = []
+        for b in blks:
+            m1.data_segment(b.segment_id).write(b.offset, b"\xcd" * size)
+            keys.append((make_key(1, b.segment_id), b.offset))
+        errs = []
+
+        def worker(key, off):
+            try:
+                for _ in range(reps):
+                    d = m0._data_client.read("127.0.0.1",
+                                             m1._data_server.port,
+                                             key, off, size)
+                    assert len(d) == size
+            except Exception as e:   # pragma: no cover
+                errs.append(e)
+
+        # warm connections
+        for key, off in keys:
+            m0._data_client.read("127.0.0.1", m1._data_server.port, key,
+                                 off, 1 << 20)
+        t0 = time.perf_counter()
+        ts = [threading.Thread(target=worker, args=k)
+              for k in keys]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join()
+        dt = time.perf_counter() - t0
+        assert not errs, errs
+        print({"parallel_streams": nthreads, "chunk_mb": chunk_mb,
+               "aggregate_gb_per_s": round(size * reps * nthreads / dt / 1e9,
+                                           2)})
+    finally:
+        m0.stop()
+        m1.stop()
+        driver.stop()
+
+
 if __name__ == "__main__":
     for chunk in (1, 4, 16):
         run(chunk_mb=chunk)
     run(chunk_mb=4, compress=True, size_mb=128)
+    run_parallel(nthreads=4)
+    run_parallel(nthreads=8)
