@@ -48,9 +48,10 @@ def main():
     ap.add_argument("--partitions-per-executor", type=int, default=0,
                     help="0 = auto (pow2, ~128 per GPU)")
     ap.add_argument("--record-bytes", type=int,
-                    default=int(os.environ.get("TERASORT_RECORD_BYTES", 16)),
-                    help="terasort record width: 16 (u64 key + u64 payload) "
-                         "or 100 (canonical 10B key + 90B value, GPU only)")
+                    default=int(os.environ.get("TERASORT_RECORD_BYTES", 100)),
+                    help="terasort record width: 100 (canonical 10B key + "
+                         "90B value — the flagship config; GPU only) or 16 "
+                         "(u64 key + u64 payload; the CPU/rccl-mode shape)")
     ap.add_argument("--validate", action="store_true")
     ap.add_argument("--cpu", action="store_true",
                     help="force CPU path (plumbing debug)")
