@@ -797,9 +797,13 @@ __global__ void onesweep_digit_bases_kernel(
 
 // Key layout inside a record: bytes [0,8) u64 LE prefix, bytes [8,10)
 // u16 LE low bits (key_bytes == 8 or 10); 80-bit order = (prefix, lo).
+// pid_func >= 0: pair.x carries the PARTITION ID of the record instead
+// of the raw prefix — the two-level >4096-partition path then radixes
+// the pid itself (coarse bits then fine bits).
 __global__ __launch_bounds__(BLOCK) void extract_pairs_kernel(
     const uint8_t* __restrict__ recs, uint64_t n, uint32_t rec_bytes,
-    uint32_t key_bytes, uint64_t* __restrict__ pairs) {
+    uint32_t key_bytes, uint64_t* __restrict__ pairs, int pid_func,
+    int pid_shift, uint32_t pid_mask, uint32_t pid_nparts) {
   const uint64_t stride = (uint64_t)gridDim.x * BLOCK;
   for (uint64_t e = (uint64_t)blockIdx.x * BLOCK + threadIdx.x; e < n;
        e += stride) {
@@ -813,8 +817,11 @@ __global__ __launch_bounds__(BLOCK) void extract_pairs_kernel(
       uint16_t klo = (uint16_t)(*reinterpret_cast<const uint32_t*>(p + 8));
       aux |= (uint64_t)klo << 48;
     }
-    reinterpret_cast<ulonglong2*>(pairs)[e] =
-        ulonglong2{prefix, aux};
+    uint64_t x = pid_func < 0
+        ? prefix
+        : (uint64_t)digit_of(prefix, pid_shift, pid_mask, pid_func,
+                             pid_nparts);
+    reinterpret_cast<ulonglong2*>(pairs)[e] = ulonglong2{x, aux};
   }
 }
 
@@ -854,7 +861,9 @@ __global__ __launch_bounds__(BLOCK) void gather_records_kernel(
 }
 
 void extract_pairs(uintptr_t recs, uint64_t n, uint32_t rec_bytes,
-                   uint32_t key_bytes, uintptr_t pairs, uintptr_t stream) {
+                   uint32_t key_bytes, uintptr_t pairs, uintptr_t stream,
+                   int pid_func, int pid_shift, uint32_t pid_mask,
+                   uint32_t pid_nparts) {
   if (rec_bytes % 4 || rec_bytes < 8)
     throw std::runtime_error("rec_bytes must be a multiple of 4, >= 8");
   if (key_bytes != 8 && key_bytes != 10)
@@ -866,7 +875,8 @@ void extract_pairs(uintptr_t recs, uint64_t n, uint32_t rec_bytes,
   hipLaunchKernelGGL(extract_pairs_kernel, dim3(grid ? grid : 1),
                      dim3(BLOCK), 0, s,
                      reinterpret_cast<const uint8_t*>(recs), n, rec_bytes,
-                     key_bytes, reinterpret_cast<uint64_t*>(pairs));
+                     key_bytes, reinterpret_cast<uint64_t*>(pairs),
+                     pid_func, pid_shift, pid_mask, pid_nparts);
   HIP_CHECK(hipGetLastError());
 }
 

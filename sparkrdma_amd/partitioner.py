@@ -11,8 +11,9 @@ kernel-side partition function (kernels.hip ``PartFunc``):
   1 hash+bits  (mix(key) >> shift) & (R-1)   pow2 R >= 16
   2 range      mulhi(key, R)                 ANY R  (uniform u64 range)
   3 hash mod   mix(key) % R                  ANY R
-Funcs 2/3 lift the r01 pow2-only restriction; R <= 4096 in one pass
-(larger R: CPU path — the LDS counter layout caps one-pass digits at 2^12).
+Funcs 2/3 lift the r01 pow2-only restriction. The LDS counter layout
+caps ONE kernel pass at 2^12 digits; the writer runs a two-level
+coarse/fine pid radix for 4096 < R <= 2^24 (writer._commit_gpu_records).
 """
 
 from __future__ import annotations
@@ -21,7 +22,7 @@ from typing import Optional, Tuple
 
 import numpy as np
 
-GPU_MAX_PARTITIONS = 1 << 12
+GPU_MAX_PARTITIONS = 1 << 24
 
 
 class HashPartitioner:
@@ -42,9 +43,8 @@ class HashPartitioner:
         nbits = (R - 1).bit_length()
         if (1 << nbits) == R and 4 <= nbits <= 12:
             return (1, 0, 0)        # hash then & (R-1) == % R for pow2
-        if R <= GPU_MAX_PARTITIONS:
-            return (3, 0, R)        # hash then % R (any R)
-        return None
+        return (3, 0, R)            # hash then % R (any R; the writer
+                                    # two-levels when R > 4096)
 
     def partition_ids(self, keys: np.ndarray) -> np.ndarray:
         k = keys.astype(np.uint64, copy=False)
@@ -89,8 +89,9 @@ class RangePartitioner:
         R = self.num_partitions
         if hasattr(self, "gpu_shift"):
             return (0, self.gpu_shift, 0)
-        if self._uniform_full_range and R <= GPU_MAX_PARTITIONS:
-            return (2, 0, R)        # mulhi(key, R) — any R
+        if self._uniform_full_range:
+            return (2, 0, R)        # mulhi(key, R) — any R; the writer
+                                    # two-levels when R > 4096
         return None                 # custom bounds: CPU path only
 
     def partition_ids(self, keys: np.ndarray) -> np.ndarray:

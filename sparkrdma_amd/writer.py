@@ -211,7 +211,7 @@ class ShuffleWriter:
                 f"partitioner {partitioner!r} has no GPU dispatch "
                 "(gpu_params() returned None) — use the CPU write path")
         func, shift, nparts = params
-        return func, shift, nparts, max(nbits, 4)
+        return func, shift, nparts, min(max(nbits, 4), 12)
 
     def _commit_gpu(self, partitioner) -> None:
         """Map-side GPU write: one radix pass whose scatter writes each
@@ -225,12 +225,28 @@ class ShuffleWriter:
             raise RuntimeError("GPU writer requires the GPU data plane")
         hs = ops_load()
         R = self.handle.num_partitions
-        func, shift, nparts, nbits_eff = self._gpu_part_params(partitioner)
         batches = self._gpu_batches
         keys = (batches[0][0] if len(batches) == 1
                 else torch.cat([b[0] for b in batches]))
         vals = (batches[0][1] if len(batches) == 1 or batches[0][1] is None
                 else torch.cat([b[1] for b in batches]))
+        if (R - 1).bit_length() > 12:
+            # > 4096 partitions: route through the wide-record two-level
+            # pid radix (records = the interleaved 16-B AoS pairs, or the
+            # bare keys when value-less)
+            if vals is None:
+                recs = keys.contiguous().view(torch.uint8).reshape(-1)
+                self._record_shape = (8, 8)
+            else:
+                pr = torch.empty(2 * keys.numel(), dtype=torch.int64,
+                                 device=keys.device)
+                pr[0::2] = keys
+                pr[1::2] = vals
+                recs = pr.view(torch.uint8).reshape(-1)
+                self._record_shape = (16, 8)
+            self._gpu_records = [recs]
+            return self._commit_gpu_records(partitioner)
+        func, shift, nparts, nbits_eff = self._gpu_part_params(partitioner)
         has_val = vals is not None
         n = keys.numel()
         dev = keys.device
@@ -340,6 +356,89 @@ class ShuffleWriter:
         mgr.keep_alive(self.handle, self.map_id, blocks)
         mgr.publish_map_output(self.handle, self.map_id, table_addr)
 
+    def _group_pairs_2level(self, hs, recs, n, W, key_bytes, R,
+                            func, shift, nparts, dev, stream):
+        """> 4096 partitions: the single radix pass is capped at 2^12 LDS
+        digits, so the PARTITION ID itself is radixed in two levels —
+        extract (pid, aux) pairs, scatter by pid>>12 (coarse), then per
+        coarse bucket by pid&4095 (fine). Returns (full R counts, pairs,
+        grouped pairs, gather digit params: pid is pair.x so the gather
+        recomputes it as plain bits)."""
+        import torch
+        nbits = (R - 1).bit_length()
+        if nbits > 24:
+            raise ValueError(f"GPU partitioner supports R <= 2^24, got {R}")
+        mask_full = (1 << nbits) - 1
+        G = (R + 4095) >> 12
+        nbits_c = max((G - 1).bit_length(), 4)
+        ndc = 1 << nbits_c
+        pairs = torch.empty(2 * n, dtype=torch.int64, device=dev)
+        hs.extract_pairs(recs.data_ptr(), n, W, key_bytes, pairs.data_ptr(),
+                         stream, func, shift, mask_full, nparts)
+        hist = torch.empty(hs.radix_hist_bytes(n, 12) // 4,
+                           dtype=torch.int32, device=dev)
+        scan_ws = torch.empty(hs.radix_scan_ws_bytes(n, 12) // 4,
+                              dtype=torch.int32, device=dev)
+        totals_c = torch.empty(ndc, dtype=torch.int32, device=dev)
+        # coarse: group by pid >> 12
+        hs.radix_hist(pairs.data_ptr(), n, 12, nbits_c, hist.data_ptr(),
+                      stream, 0, 2, 0)
+        hs.radix_scan(hist.data_ptr(), n, nbits_c, totals_c.data_ptr(),
+                      scan_ws.data_ptr(), stream)
+        counts_c = totals_c.cpu().numpy().astype(np.int64)[:G]   # sync
+        starts_c = np.zeros(G, dtype=np.int64)
+        np.cumsum(counts_c[:-1], out=starts_c[1:])
+        pairs_c = torch.empty_like(pairs)
+        bases_c = torch.from_numpy(
+            np.pad(starts_c, (0, ndc - G))).to(dev)
+        kd = pairs_c.data_ptr() + bases_c * 16
+        vd = kd + 8
+        hs.radix_scatter(pairs.data_ptr(), pairs.data_ptr() + 8, n, 12,
+                         nbits_c, hist.data_ptr(), kd.data_ptr(),
+                         vd.data_ptr(), stream, 0, 1, 2, 0)
+        # fine: per coarse bucket, group its subrange by pid & 4095.
+        # All fine hists run first (stream-ordered, per-bucket totals
+        # persist), ONE sync, then all fine scatters.
+        totals_f = torch.empty(G * 4096, dtype=torch.int32, device=dev)
+        bhists = []
+        for b in range(G):
+            cnt = int(counts_c[b])
+            if cnt == 0:
+                totals_f[b * 4096:(b + 1) * 4096] = 0
+                bhists.append(None)
+                continue
+            bh = torch.empty(hs.radix_hist_bytes(cnt, 12) // 4,
+                             dtype=torch.int32, device=dev)
+            bws = torch.empty(hs.radix_scan_ws_bytes(cnt, 12) // 4,
+                              dtype=torch.int32, device=dev)
+            sub = pairs_c.data_ptr() + starts_c[b] * 16
+            hs.radix_hist(sub, cnt, 0, 12, bh.data_ptr(), stream, 0, 2, 0)
+            hs.radix_scan(bh.data_ptr(), cnt, 12,
+                          totals_f.data_ptr() + b * 4096 * 4,
+                          bws.data_ptr(), stream)
+            bhists.append((bh, bws, sub, cnt))
+        counts_full = totals_f.cpu().numpy().astype(np.int64)[:R]   # sync
+        # pids b*4096+j with j beyond R's tail never occur; length R slice
+        starts_full = np.zeros(R, dtype=np.int64)
+        np.cumsum(counts_full[:-1], out=starts_full[1:])
+        pairs_f = pairs   # reuse: original extraction no longer needed
+        base_addr = pairs_f.data_ptr() + torch.from_numpy(
+            starts_full).to(dev) * 16
+        for b in range(G):
+            if bhists[b] is None:
+                continue
+            bh, bws, sub, cnt = bhists[b]
+            lo = b * 4096
+            hi = min(lo + 4096, R)
+            kd_b = torch.zeros(4096, dtype=torch.int64, device=dev)
+            kd_b[:hi - lo] = base_addr[lo:hi]
+            vd_b = kd_b + 8
+            hs.radix_scatter(sub, sub + 8, cnt, 0, 12, bh.data_ptr(),
+                             kd_b.data_ptr(), vd_b.data_ptr(), stream,
+                             0, 1, 2, 0)
+        # gather digit = pair.x (the pid) verbatim: func 0, shift 0
+        return counts_full, pairs_c, pairs_f, 0, 0, mask_full, 0
+
     def _commit_gpu_records(self, partitioner) -> None:
         """Wide-record map-side GPU write. Records stay put while 16-byte
         (key-prefix, aux) pairs run the radix machinery; ONE gather pass
@@ -361,20 +460,29 @@ class ShuffleWriter:
         n = recs.numel() // W
         dev = recs.device
         stream = torch.cuda.current_stream().cuda_stream
-        nd = 1 << nbits_eff
-        pairs = torch.empty(2 * n, dtype=torch.int64, device=dev)
-        hs.extract_pairs(recs.data_ptr(), n, W, key_bytes, pairs.data_ptr(),
-                         stream)
-        hist = torch.empty(hs.radix_hist_bytes(n, nbits_eff) // 4,
-                           dtype=torch.int32, device=dev)
-        scan_ws = torch.empty(hs.radix_scan_ws_bytes(n, nbits_eff) // 4,
-                              dtype=torch.int32, device=dev)
-        totals = torch.empty(nd, dtype=torch.int32, device=dev)
-        hs.radix_hist(pairs.data_ptr(), n, shift, nbits_eff, hist.data_ptr(),
-                      stream, func, 2, nparts)
-        hs.radix_scan(hist.data_ptr(), n, nbits_eff, totals.data_ptr(),
-                      scan_ws.data_ptr(), stream)
-        counts_nd = totals.cpu().numpy().astype(np.int64)  # syncs the stream
+        two_level = (R - 1).bit_length() > 12
+        if two_level:
+            counts_nd, pairs, pairs_out, gfunc, gshift, gmask, gnparts = \
+                self._group_pairs_2level(hs, recs, n, W, key_bytes, R,
+                                         func, shift, nparts, dev, stream)
+            nd = R
+        else:
+            nd = 1 << nbits_eff
+            gfunc, gshift, gmask, gnparts = func, shift, nd - 1, nparts
+            pairs = torch.empty(2 * n, dtype=torch.int64, device=dev)
+            hs.extract_pairs(recs.data_ptr(), n, W, key_bytes,
+                             pairs.data_ptr(), stream)
+            hist = torch.empty(hs.radix_hist_bytes(n, nbits_eff) // 4,
+                               dtype=torch.int32, device=dev)
+            scan_ws = torch.empty(hs.radix_scan_ws_bytes(n, nbits_eff) // 4,
+                                  dtype=torch.int32, device=dev)
+            totals = torch.empty(nd, dtype=torch.int32, device=dev)
+            hs.radix_hist(pairs.data_ptr(), n, shift, nbits_eff,
+                          hist.data_ptr(), stream, func, 2, nparts)
+            hs.radix_scan(hist.data_ptr(), n, nbits_eff, totals.data_ptr(),
+                          scan_ws.data_ptr(), stream)
+            counts_nd = totals.cpu().numpy().astype(np.int64)  # syncs
+            pairs_out = None   # grouped below, after the layout
         counts = counts_nd[:R]
         seg_bytes = counts * W
 
@@ -435,20 +543,22 @@ class ShuffleWriter:
         # group the PAIRS by digit (contiguous per-digit runs)
         starts = np.zeros(nd, dtype=np.int64)
         np.cumsum(counts_nd[:-1], out=starts[1:])
-        pairs_out = torch.empty_like(pairs)
-        bases_t = torch.from_numpy(starts).to(dev)
-        kd = pairs_out.data_ptr() + bases_t * 16
-        vd = kd + 8
-        hs.radix_scatter(pairs.data_ptr(), pairs.data_ptr() + 8, n, shift,
-                         nbits_eff, hist.data_ptr(), kd.data_ptr(),
-                         vd.data_ptr(), stream, func, 1, 2, nparts)
+        if pairs_out is None:   # single-pass grouping (two-level grouped
+            pairs_out = torch.empty_like(pairs)    # during count discovery)
+            bases_t = torch.from_numpy(starts).to(dev)
+            kd = pairs_out.data_ptr() + bases_t * 16
+            vd = kd + 8
+            hs.radix_scatter(pairs.data_ptr(), pairs.data_ptr() + 8, n,
+                             shift, nbits_eff, hist.data_ptr(),
+                             kd.data_ptr(), vd.data_ptr(), stream, func, 1,
+                             2, nparts)
         # one gather moves every record to its final slot
         dstart_t = torch.from_numpy(starts.astype(np.uint32)
                                     .view(np.int32)).to(dev)
         dst_addr_t = torch.from_numpy(rec_dst).to(dev)
         hs.gather_records(recs.data_ptr(), pairs_out.data_ptr(), n, W, 1, 0,
-                          dst_addr_t.data_ptr(), dstart_t.data_ptr(), shift,
-                          nd - 1, func, nparts, stream)
+                          dst_addr_t.data_ptr(), dstart_t.data_ptr(), gshift,
+                          gmask, gfunc, gnparts, stream)
         torch.cuda.synchronize()
         if spill_groups:
             soff = 0
