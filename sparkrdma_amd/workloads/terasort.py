@@ -179,7 +179,7 @@ class TeraSort:
         outs = []
         remote = 0
         chunk_shared_bits = (self.R // per - 1).bit_length()
-        for reader in readers:
+        for (chunk_lo, _chunk_hi), reader in zip(spans, readers):
             tf = time.perf_counter()
             parts = reader.collect_partitions()
             ts_ = time.perf_counter()
@@ -214,9 +214,10 @@ class TeraSort:
             t_fetch_total += ts_ - tf
             t_sort_total += time.perf_counter() - ts_
             remote += reader.metrics.remote_bytes_read
-        if self.validate:
-            for (a, _b), out in zip(spans, outs):
-                self._validate(out, a, per)
+            if self.validate:
+                # validate THIS chunk now: the sort tmp/out caches are
+                # shared across chunks, so chunk h+1 overwrites h's view
+                self._validate(outs[-1], chunk_lo, per)
         eng.unregister_shuffle(handle)
         dt = time.perf_counter() - t0
         return TeraSortResult(
