@@ -193,13 +193,15 @@ class ShuffleWriter:
 
     def _gpu_part_params(self, partitioner):
         """(func, shift, nparts, nbits_eff) for the kernel-side partition
-        function — any R <= 4096 (pow2 via bit extraction, arbitrary via
-        mulhi-range / hash-mod; partitioner.gpu_params())."""
+        function (partitioner.gpu_params()): pow2 R via bit extraction,
+        arbitrary R via mulhi-range / hash-mod. One kernel pass covers
+        nbits <= 12; larger R runs the two-level pid radix
+        (_group_pairs_2level), capped at 2^24."""
         R = self.handle.num_partitions
         nbits = max((R - 1).bit_length(), 1)
-        if nbits > 12:
+        if nbits > 24:
             raise ValueError(
-                f"GPU partitioner supports R <= 4096 in one pass, got {R}; "
+                f"GPU partitioner supports R <= 2^24, got {R}; "
                 "use the CPU write path for larger partition counts")
         params = None
         if hasattr(partitioner, "gpu_params"):
