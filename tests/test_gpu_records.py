@@ -273,7 +273,7 @@ def test_two_level_partition_counts(R, tmp_path):
                 assert np.all(pids == ref.partition)
                 assert len(chunk) == want_counts[ref.partition]
                 seen += len(chunk)
-        assert seen == sum(want_counts[lo:lo + 257]
+        assert seen == sum(int(want_counts[lo:lo + 257].sum())
                            for lo in (0, R // 2, R - 257))
     finally:
         mgr.stop()
