@@ -78,8 +78,8 @@ def main():
     from sparkrdma_amd.workloads.terasort import TeraSort
 
     rec_bytes = args.record_bytes if args.workload == "terasort" else 16
-    if rec_bytes != 16 and (not use_cuda or args.mode != "framework"):
-        rec_bytes = 16   # wide records: GPU framework mode only
+    if rec_bytes != 16 and not use_cuda:
+        rec_bytes = 16   # wide records are GPU-only
     n_rec = int(args.gb_per_gpu * (1 << 30) / rec_bytes)
     ppe = args.partitions_per_executor
     if ppe == 0:

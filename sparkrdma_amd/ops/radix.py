@@ -217,6 +217,52 @@ def sort_records(recs: torch.Tensor, rec_bytes: int, key_bytes: int = 8,
     return out
 
 
+def partition_records(recs: torch.Tensor, rec_bytes: int,
+                      key_bytes: int = 8, nbits: int = 8,
+                      shift: Optional[int] = None, func: int = 0,
+                      nparts: int = 0):
+    """Group W-byte records into 2^nbits contiguous buckets (digit from
+    the key prefix) — the stage-mode (RCCL alltoall) send-buffer builder
+    for wide records. Returns (counts int64 numpy, grouped records)."""
+    m = load()
+    n = recs.numel() // rec_bytes
+    if shift is None:
+        shift = 64 - nbits
+    nbits_eff = max(nbits, 4)
+    nd = 1 << nbits_eff
+    dev = recs.device
+    s = _stream()
+    pairs = extract_pairs(recs, rec_bytes, key_bytes)
+    hist = torch.empty(m.radix_hist_bytes(n, nbits_eff) // 4,
+                       dtype=torch.int32, device=dev)
+    scan_ws = torch.empty(m.radix_scan_ws_bytes(n, nbits_eff) // 4,
+                          dtype=torch.int32, device=dev)
+    totals = torch.empty(nd, dtype=torch.int32, device=dev)
+    m.radix_hist(pairs.data_ptr(), n, shift, nbits_eff, hist.data_ptr(), s,
+                 func, 2, nparts)
+    m.radix_scan(hist.data_ptr(), n, nbits_eff, totals.data_ptr(),
+                 scan_ws.data_ptr(), s)
+    import numpy as np
+    counts = totals.cpu().numpy().astype(np.int64)  # syncs the stream
+    starts = np.zeros(nd, dtype=np.int64)
+    np.cumsum(counts[:-1], out=starts[1:])
+    pairs_out = torch.empty_like(pairs)
+    bases_t = torch.from_numpy(starts).to(dev)
+    kd = pairs_out.data_ptr() + bases_t * 16
+    vd = kd + 8
+    m.radix_scatter(pairs.data_ptr(), pairs.data_ptr() + 8, n, shift,
+                    nbits_eff, hist.data_ptr(), kd.data_ptr(),
+                    vd.data_ptr(), s, func, 1, 2, nparts)
+    out = torch.empty_like(recs)
+    dstart_t = torch.from_numpy(starts.astype(np.uint32)
+                                .view(np.int32)).to(dev)
+    dst_addr_t = out.data_ptr() + bases_t * rec_bytes
+    m.gather_records(recs.data_ptr(), pairs_out.data_ptr(), n, rec_bytes,
+                     1, 0, dst_addr_t.data_ptr(), dstart_t.data_ptr(),
+                     shift, nd - 1, func, nparts, s)
+    return counts[:1 << nbits], out
+
+
 def sort_pairs(keys: torch.Tensor, vals: Optional[torch.Tensor] = None,
                start_bit: int = 0, end_bit: int = 64,
                onesweep: Optional[bool] = None

@@ -64,8 +64,8 @@ class TeraSort:
         self.validate = validate
         self.RECORD_BYTES = record_bytes
         self.wide = record_bytes != 16
-        if self.wide and (device != "cuda" or mode != "framework"):
-            raise ValueError("wide records need device=cuda, mode=framework")
+        if self.wide and device != "cuda":
+            raise ValueError("wide records need device=cuda")
         W = engine.world_size
         # R must be pow2 for the GPU top-bits partitioner
         R = W * partitions_per_executor
@@ -120,7 +120,9 @@ class TeraSort:
         if getattr(self, "_sr_handle", None) is None:
             handle = eng.register_shuffle(eng.world_size, self.R)
             w = eng.manager.get_writer(handle, rank)
-            if self.device == "cuda":
+            if self.wide:
+                w.write_device_records(self.recs, self.RECORD_BYTES, 10)
+            elif self.device == "cuda":
                 w.write_device_batch(self.keys, self.vals)
             else:
                 w.write_batch(self.keys, self.vals)
@@ -345,7 +347,9 @@ class TeraSort:
     def _step_rccl(self) -> TeraSortResult:
         """Stage-mode shuffle: ONE RCCL all_to_all_single over xGMI of
         AoS records (SURVEY §7.1 'collective-form option') — the partition
-        pass scatters straight into the send buffer."""
+        pass scatters straight into the send buffer. Wide records ship as
+        uint8 with byte splits (partition_records builds the grouped send
+        buffer; the reduce side is the same pair-indirection sort)."""
         import torch
         import torch.distributed as dist
         from ..ops.radix import partition_aos, sort_pairs_aos
@@ -353,6 +357,8 @@ class TeraSort:
         W = eng.world_size
         t0 = time.perf_counter()
         wbits = (W - 1).bit_length()
+        if self.wide:
+            return self._step_rccl_wide(W, wbits, t0)
         if W == 1:
             pairs = torch.empty(2 * self.n, dtype=torch.int64, device="cuda")
             pairs[0::2] = self.keys
@@ -384,3 +390,37 @@ class TeraSort:
         return TeraSortResult(dt, self.n, self.n * 16,
                               t_fetch - t0, 0, t_sort - t_fetch,
                               int(sum(out_l)) * 8)
+
+    def _step_rccl_wide(self, W: int, wbits: int, t0: float) -> TeraSortResult:
+        import torch
+        import torch.distributed as dist
+        from ..ops.radix import partition_records, sort_records
+        Wb = self.RECORD_BYTES
+        if W == 1:
+            out = sort_records(self.recs, Wb, key_bytes=10)
+            torch.cuda.synchronize()
+            dt = time.perf_counter() - t0
+            if self.validate:
+                self._validate_wide(out, 0, self.R)
+            return TeraSortResult(dt, self.n, self.n * Wb, 0, 0, dt, 0)
+        counts, grouped = partition_records(self.recs, Wb, key_bytes=10,
+                                            nbits=wbits, shift=64 - wbits)
+        in_t = torch.from_numpy(counts * Wb).cuda()        # bytes per rank
+        out_t = torch.empty_like(in_t)
+        dist.all_to_all_single(out_t, in_t)
+        in_l = (counts * Wb).tolist()
+        out_l = out_t.cpu().tolist()
+        recv = torch.empty(int(sum(out_l)), dtype=torch.uint8,
+                           device="cuda")
+        dist.all_to_all_single(recv, grouped, out_l, in_l)
+        t_fetch = time.perf_counter()
+        out = sort_records(recv, Wb, key_bytes=10, end_bit=64 - wbits)
+        torch.cuda.synchronize()
+        t_sort = time.perf_counter()
+        if self.validate:
+            # rank r holds pids with top wbits == r == [r*ppe, (r+1)*ppe)
+            self._validate_wide(out, self.engine.rank * self.ppe, self.ppe)
+        dt = time.perf_counter() - t0
+        return TeraSortResult(dt, self.n, self.n * Wb,
+                              t_fetch - t0, 0, t_sort - t_fetch,
+                              int(sum(out_l)))
