@@ -149,11 +149,25 @@ class DataServer:
             sock.sendall(raw)
 
     def _stream_host(self, sock, key: int, addr: int, length: int) -> None:
-        reg = self.manager._registry
+        reader = self.manager._registry.reader(key)
+        if not (self._flags & FLAG_DEFLATE):
+            # zero-copy lane: kernel-side sendfile from the shm segment,
+            # no userspace byte handling at all
+            import os
+            off = 0
+            while off < length:
+                c = min(self._chunk, length - off)
+                sock.sendall(_CHUNK.pack(c, c))
+                sent = 0
+                while sent < c:
+                    sent += os.sendfile(sock.fileno(), reader.fd,
+                                        addr + off + sent, c - sent)
+                off += c
+            return
         off = 0
         while off < length:
             c = min(self._chunk, length - off)
-            raw = reg.read(key, addr + off, c)
+            raw = reader.read(addr + off, c)
             if len(raw) != c:
                 raise IOError(f"short segment read {len(raw)}/{c}")
             self._send_chunk(sock, raw)
