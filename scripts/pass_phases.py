@@ -41,8 +41,15 @@ def main():
     names = ["phaseA(load+rank+scan+pub)", "exchange",
              "lookback walk (+barrier)", "writeout"]
     ref = None
-    for lb_mode, label in ((0, "[nb][ND] (r01)"), (1, "transposed [ND][nb]")):
+    configs = ((0, 0, "[nb][ND] (r01)"),
+               (1, 0, "transposed [ND][nb]"),
+               (0, 1, "split exchange (3 blocks/CU)"))
+    import os as _os
+    if _os.environ.get("PHASES_SKIP_TRANSPOSED"):
+        configs = (configs[0], configs[2])
+    for lb_mode, split, label in configs:
         m.set_lookback_mode(lb_mode)
+        m.set_split_exchange(split)
         pp = pairs.clone()
         sort_pairs_aos(pp, 0, 64, tmp=tmp, ws=ws)  # warm this mode
         torch.cuda.synchronize()
@@ -66,13 +73,14 @@ def main():
         torch.cuda.synchronize()
         wall_clean = time.perf_counter() - t0
         t = timing.cpu().numpy() / REALTIME_MHZ / (nblocks * passes)
-        print(f"lb_mode={lb_mode} {label}: wall={wall_clean*1e3:.2f} ms "
+        print(f"{label}: wall={wall_clean*1e3:.2f} ms "
               f"({wall_clean/passes*1e3:.3f} ms/pass; instrumented "
               f"{wall/passes*1e3:.3f})")
         for nm, v in zip(names, t):
             print(f"  {nm:<28} {v:7.2f} us/block")
         print(f"  {'total in-kernel':<28} {t.sum():7.2f} us/block")
     m.set_lookback_mode(0)
+    m.set_split_exchange(0)
 
 
 if __name__ == "__main__":

@@ -522,16 +522,20 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
     const uint64_t* __restrict__ key_dst, const uint64_t* __restrict__ val_dst,
     int stage = 3, uint64_t* __restrict__ timing = nullptr,
     const uint32_t* __restrict__ hist_pref = nullptr, int sort_word = 0,
-    int lb_mode = 0 /* 0: [nb][ND] descriptors, 1: transposed [ND][nb] */) {
+    int lb_mode = 0 /* 0: [nb][ND] descriptors, 1: transposed [ND][nb] */,
+    int split_exch = 0 /* AoS: exchange+writeout in 2 half-tile rounds so
+                          the LDS buffer halves -> 3 blocks/CU resident
+                          hide more of the lookback wait */) {
   constexpr int ND = 1 << PBITS;
   extern __shared__ char smem_raw[];
   constexpr int TILE_T = BS * IT;
   constexpr int NWT = BS / kWave;
   using u64x2 = __attribute__((ext_vector_type(2))) unsigned long long;
-  uint64_t* exch = reinterpret_cast<uint64_t*>(smem_raw);  // [TILE_T] or
-  u64x2* exch2 = reinterpret_cast<u64x2*>(smem_raw);       // [TILE_T] pairs
+  const int ecap = (AOS && split_exch) ? TILE_T / 2 : TILE_T;
+  uint64_t* exch = reinterpret_cast<uint64_t*>(smem_raw);  // [ecap] or
+  u64x2* exch2 = reinterpret_cast<u64x2*>(smem_raw);       // [ecap] pairs
   uint32_t* counters = reinterpret_cast<uint32_t*>(
-      exch + (AOS ? 2 * TILE_T : TILE_T));
+      exch + (AOS ? 2 * ecap : TILE_T));
   uint32_t* start = counters + NWT * ND;
   uint32_t* pref = start + ND;
   uint32_t* sums = pref + ND;  // [BS]
@@ -664,38 +668,49 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
 
   uint64_t t1 = timing ? __builtin_amdgcn_s_memrealtime() : 0;
   if (AOS) {
-    if (stage >= 2) {
-      // pair exchange first: the deferred lookback's wait overlaps it
+    uint64_t t2 = t1, t3 = t1;
+    const uint32_t off_mask = stage < 3 ? 1023u : 0xFFFFFFFFu;
+    const int rounds = split_exch ? 2 : 1;
+    for (int round = 0; round < rounds; ++round) {
+      if (round) __syncthreads();     // previous half's writeout done
+      if (stage >= 2) {
+        // pair exchange first: the deferred lookback's wait overlaps it
 #pragma unroll
-      for (int i = 0; i < IT; ++i) {
-        uint64_t e = chunk + (uint64_t)i * kWave + lane;
-        if (e < n) {
-          uint32_t d = digrank[i] >> 16;
-          uint32_t j = start[d] + my[d] + (digrank[i] & 0xFFFF);
-          exch2[j] = u64x2{key_reg[i], val_reg[i]};
+        for (int i = 0; i < IT; ++i) {
+          uint64_t e = chunk + (uint64_t)i * kWave + lane;
+          if (e < n) {
+            uint32_t d = digrank[i] >> 16;
+            uint32_t j = start[d] + my[d] + (digrank[i] & 0xFFFF);
+            uint32_t jl = j - (uint32_t)(round * ecap);
+            if (jl < (uint32_t)ecap)
+              exch2[jl] = u64x2{key_reg[i], val_reg[i]};
+          }
         }
       }
-    }
-    uint64_t t2 = timing ? __builtin_amdgcn_s_memrealtime() : 0;
-    if (!hist_pref && stage != 0 && b != 0 && tid < ND) {
-      if (lb_mode)
-        lookback_walk_t<ND>(desc, gridDim.x, b, tid, my_total, pref);
-      else
-        lookback_walk<ND>(desc, b, tid, my_total, pref);
-    }
-    if (stage < 2 && !timing) return;  // ablation: rank/publish/lookback
-    __syncthreads();
-    uint64_t t3 = timing ? __builtin_amdgcn_s_memrealtime() : 0;
-    const uint32_t off_mask = stage < 3 ? 1023u : 0xFFFFFFFFu;
+      if (round == 0) {
+        t2 = timing ? __builtin_amdgcn_s_memrealtime() : 0;
+        if (!hist_pref && stage != 0 && b != 0 && tid < ND) {
+          if (lb_mode)
+            lookback_walk_t<ND>(desc, gridDim.x, b, tid, my_total, pref);
+          else
+            lookback_walk<ND>(desc, b, tid, my_total, pref);
+        }
+        if (stage < 2 && !timing) return;  // ablation
+      }
+      __syncthreads();
+      if (round == 0) t3 = timing ? __builtin_amdgcn_s_memrealtime() : 0;
 #pragma unroll
-    for (int i = 0; i < IT; ++i) {
-      uint32_t j = i * BS + tid;
-      if (j < tile_n) {
-        u64x2 kv = exch2[j];
-        uint32_t d = (uint32_t)((uint64_t)(sort_word ? kv.y : kv.x)
-                                >> shift) & (ND - 1);
-        uint32_t off = (pref[d] + (j - start[d])) & off_mask;
-        reinterpret_cast<u64x2*>(key_dst[d])[off] = kv;
+      for (int i = 0; i < IT; ++i) {
+        uint32_t jl = (uint32_t)i * BS + tid;
+        if (jl >= (uint32_t)ecap) break;
+        uint32_t j = (uint32_t)round * ecap + jl;
+        if (j < tile_n) {
+          u64x2 kv = exch2[jl];
+          uint32_t d = (uint32_t)((uint64_t)(sort_word ? kv.y : kv.x)
+                                  >> shift) & (ND - 1);
+          uint32_t off = (pref[d] + (j - start[d])) & off_mask;
+          reinterpret_cast<u64x2*>(key_dst[d])[off] = kv;
+        }
       }
     }
     if (timing && tid == 0) {
@@ -703,7 +718,7 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
       atomicAdd(&timing[0], t1 - t0);  // phase A (+scans+publish)
       atomicAdd(&timing[1], t2 - t1);  // exchange
       atomicAdd(&timing[2], t3 - t2);  // lookback walk + barrier
-      atomicAdd(&timing[3], t4 - t3);  // writeout
+      atomicAdd(&timing[3], t4 - t3);  // writeout (+2nd round when split)
     }
     return;
   }
@@ -1190,6 +1205,10 @@ void set_timing_buf(uintptr_t p) { g_timing_buf = reinterpret_cast<uint64_t*>(p)
 // reads contiguous backwards — 16 descriptors/line)
 static int g_lb_mode = 0;
 void set_lookback_mode(int m) { g_lb_mode = m; }
+// AoS split exchange: halve the LDS exchange buffer (2 rounds) so 3
+// blocks/CU stay resident and hide more of the lookback wait
+static int g_split_exch = 0;
+void set_split_exchange(int m) { g_split_exch = m; }
 
 static inline uint32_t os_num_tiles_t(uint32_t n, int tile) {
   return (uint32_t)(((uint64_t)n + tile - 1) / tile);
@@ -1238,8 +1257,8 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
   }
   size_t lds_soa = (size_t)BLOCK * OS_ITEMS * 8 + (size_t)NW * PD * 4 +
                    PD * 4 * 2 + BLOCK * 4 + 16 + BLOCK * OS_ITEMS;
-  size_t lds_aos = (size_t)aos_tile * 16 + (size_t)(512 / kWave) * PD * 4 +
-                   PD * 4 * 2 + 512 * 4 + 16;
+  size_t lds_aos = (size_t)aos_tile * (g_split_exch ? 8 : 16) +
+                   (size_t)(512 / kWave) * PD * 4 + PD * 4 * 2 + 512 * 4 + 16;
   size_t lds = aos ? lds_aos : lds_soa;
   static bool attr_set = false;
   if (!attr_set) {
@@ -1287,19 +1306,22 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          sb, desc, ticket, key_dst, val_dst, g_pass_stage,
-                         g_timing_buf, pass_pref, sort_word, g_lb_mode);
+                         g_timing_buf, pass_pref, sort_word, g_lb_mode,
+                         g_split_exch);
     } else if (aos && aos_tile == 2048) {
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 4, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          sb, desc, ticket, key_dst, val_dst, g_pass_stage,
-                         g_timing_buf, pass_pref, sort_word, g_lb_mode);
+                         g_timing_buf, pass_pref, sort_word, g_lb_mode,
+                         g_split_exch);
     } else if (aos) {
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 8, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
                          sb, desc, ticket, key_dst, val_dst, g_pass_stage,
-                         g_timing_buf, pass_pref, sort_word, g_lb_mode);
+                         g_timing_buf, pass_pref, sort_word, g_lb_mode,
+                         g_split_exch);
     } else if (vals) {
       hipLaunchKernelGGL(
           (onesweep_pass_kernel<true, OS_ITEMS, false, BLOCK, PBITS>),
