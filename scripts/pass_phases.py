@@ -41,15 +41,18 @@ def main():
     names = ["phaseA(load+rank+scan+pub)", "exchange",
              "lookback walk (+barrier)", "writeout"]
     ref = None
-    configs = ((0, 0, "[nb][ND] (r01)"),
-               (1, 0, "transposed [ND][nb]"),
-               (0, 1, "split exchange (3 blocks/CU)"))
+    # (lb_mode, split_rounds, lean, label)
+    configs = [(0, 0, 0, "baseline (2 blocks/CU)"),
+               (0, 2, 1, "lean + split2 (3 blocks/CU)"),
+               (0, 4, 1, "lean + split4 (4+ blocks/CU)"),
+               (0, 0, 1, "lean only (2 blocks/CU)")]
     import os as _os
-    if _os.environ.get("PHASES_SKIP_TRANSPOSED"):
-        configs = (configs[0], configs[2])
-    for lb_mode, split, label in configs:
+    if _os.environ.get("PHASES_TRANSPOSED"):
+        configs.append((1, 0, 0, "transposed [ND][nb]"))
+    for lb_mode, split, lean, label in configs:
         m.set_lookback_mode(lb_mode)
         m.set_split_exchange(split)
+        m.set_lean_pass(lean)
         pp = pairs.clone()
         sort_pairs_aos(pp, 0, 64, tmp=tmp, ws=ws)  # warm this mode
         torch.cuda.synchronize()
@@ -81,6 +84,7 @@ def main():
         print(f"  {'total in-kernel':<28} {t.sum():7.2f} us/block")
     m.set_lookback_mode(0)
     m.set_split_exchange(0)
+    m.set_lean_pass(0)
 
 
 if __name__ == "__main__":

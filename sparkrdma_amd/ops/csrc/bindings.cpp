@@ -97,6 +97,7 @@ PYBIND11_MODULE(_hipshuffle, m) {
   m.def("set_sort_mode", &hs::set_sort_mode);
   m.def("set_lookback_mode", &hs::set_lookback_mode);
   m.def("set_split_exchange", &hs::set_split_exchange);
+  m.def("set_lean_pass", &hs::set_lean_pass);
   m.def("onesweep_sort_aos7_u64", &hs::onesweep_sort_aos7_u64,
         py::arg("pairs"), py::arg("tmp_pairs"), py::arg("n"),
         py::arg("start_bit"), py::arg("end_bit"), py::arg("ws"),

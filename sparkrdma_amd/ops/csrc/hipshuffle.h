@@ -69,6 +69,7 @@ void set_timing_buf(uintptr_t p);
 void set_sort_mode(int m);
 void set_lookback_mode(int m);
 void set_split_exchange(int m);
+void set_lean_pass(int m);
 int onesweep_sort_aos7_u64(uintptr_t pairs, uintptr_t tmp_pairs, uint32_t n,
                            int start_bit, int end_bit, uintptr_t ws,
                            uintptr_t stream);

@@ -514,8 +514,16 @@ __global__ __launch_bounds__(BLOCK) void onesweep_hist_all_kernel(
 // dwordx4 load per element, a pair LDS exchange, ONE 16-byte store per
 // element, and 2x-longer digit write bursts (measured: scattered-write
 // bandwidth doubles from 128 B to 256 B bursts — profiles/).
-template <bool HAS_VAL, int IT, bool AOS, int BS = BLOCK, int PBITS = 8>
-__global__ __launch_bounds__(BS) void onesweep_pass_kernel(
+// LEAN (AoS only): phase A loads ONLY the sort word and the exchange
+// RE-READS the full pair from global (L2-hot: the tile was just pulled)
+// — drops the val_reg/key_reg liveness so VGPRs fit 6 waves/SIMD =
+// 3 blocks/CU resident, hiding more of the lookback wait while keeping
+// full-tile burst lengths (the 2048-tile route paid burst halving +
+// 2x descriptors for its occupancy; this pays only L1/L2 re-reads).
+template <bool HAS_VAL, int IT, bool AOS, int BS = BLOCK, int PBITS = 8,
+          bool LEAN = false>
+__global__ __launch_bounds__(BS)
+__attribute__((amdgpu_waves_per_eu(LEAN ? 6 : 1))) void onesweep_pass_kernel(
     const uint64_t* __restrict__ keys, const uint64_t* __restrict__ vals,
     uint32_t n, int shift, uint64_t* __restrict__ desc /* [nb][ND] */,
     uint32_t* __restrict__ ticket,
@@ -523,15 +531,16 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
     int stage = 3, uint64_t* __restrict__ timing = nullptr,
     const uint32_t* __restrict__ hist_pref = nullptr, int sort_word = 0,
     int lb_mode = 0 /* 0: [nb][ND] descriptors, 1: transposed [ND][nb] */,
-    int split_exch = 0 /* AoS: exchange+writeout in 2 half-tile rounds so
-                          the LDS buffer halves -> 3 blocks/CU resident
-                          hide more of the lookback wait */) {
+    int split_exch = 0 /* AoS: exchange+writeout in N sub-tile rounds
+                          (0/1 = single round); the LDS exchange buffer
+                          shrinks by N so more blocks stay resident to
+                          hide the lookback wait */) {
   constexpr int ND = 1 << PBITS;
   extern __shared__ char smem_raw[];
   constexpr int TILE_T = BS * IT;
   constexpr int NWT = BS / kWave;
   using u64x2 = __attribute__((ext_vector_type(2))) unsigned long long;
-  const int ecap = (AOS && split_exch) ? TILE_T / 2 : TILE_T;
+  const int ecap = (AOS && split_exch > 1) ? TILE_T / split_exch : TILE_T;
   uint64_t* exch = reinterpret_cast<uint64_t*>(smem_raw);  // [ecap] or
   u64x2* exch2 = reinterpret_cast<u64x2*>(smem_raw);       // [ecap] pairs
   uint32_t* counters = reinterpret_cast<uint32_t*>(
@@ -561,21 +570,23 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
   // flight before the serial per-wave LDS counter chain starts — fusing
   // load+rank serialized loads behind LDS RMWs (phase A 11.4 -> 8.2 us
   // per block, full sort 7.56 -> 7.15 ms / 64M).
-  uint64_t key_reg[IT];
-  uint64_t val_reg[AOS ? IT : 1];
+  uint64_t key_reg[(AOS && LEAN) ? 1 : IT];
+  uint64_t val_reg[(AOS && !LEAN) ? IT : 1];
   uint32_t digrank[IT];
   const uint64_t chunk = tile_start + (uint64_t)wave * (IT * kWave);
+  if (!(AOS && LEAN)) {
 #pragma unroll
-  for (int i = 0; i < IT; ++i) {
-    uint64_t e = chunk + (uint64_t)i * kWave + lane;
-    bool valid = e < n;
-    if (AOS) {
-      u64x2 kv = valid ? reinterpret_cast<const u64x2*>(keys)[e]
-                       : u64x2{0, 0};
-      key_reg[i] = kv.x;
-      val_reg[i] = kv.y;
-    } else {
-      key_reg[i] = valid ? keys[e] : 0;
+    for (int i = 0; i < IT; ++i) {
+      uint64_t e = chunk + (uint64_t)i * kWave + lane;
+      bool valid = e < n;
+      if (AOS) {
+        u64x2 kv = valid ? reinterpret_cast<const u64x2*>(keys)[e]
+                         : u64x2{0, 0};
+        key_reg[i] = kv.x;
+        val_reg[(AOS && !LEAN) ? i : 0] = kv.y;
+      } else {
+        key_reg[i] = valid ? keys[e] : 0;
+      }
     }
   }
   // rank in iteration PAIRS: both iterations' per-bit ballots are held
@@ -584,6 +595,29 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
   // the serial chain was 13% of a pass (stage-9 ablation).
   static_assert(IT % 2 == 0, "pair ranking needs even IT");
   const uint64_t lt = (1ull << lane) - 1;
+  if (AOS && LEAN) {
+    // single-iteration ranking: one ballot set live at a time — the
+    // VGPR budget (not LDS) is what buys the 3rd resident block per CU;
+    // the serial LDS chain it re-lengthens is hidden by that occupancy
+    const int sw = sort_word ? 1 : 0;
+#pragma unroll
+    for (int i = 0; i < IT; ++i) {
+      uint64_t e = chunk + (uint64_t)i * kWave + lane;
+      bool valid = e < n;
+      uint64_t k = valid ? keys[2 * e + sw] : 0;
+      uint32_t d = (uint32_t)(k >> shift) & (ND - 1);
+      uint64_t vm = __ballot(valid);
+      uint32_t r = 0;
+      if (valid) {
+        uint64_t match = match_lanes<PBITS>(d, vm);
+        uint32_t rank_in_iter = (uint32_t)__popcll(match & lt);
+        uint32_t base = my[d];
+        r = base + rank_in_iter;
+        if (rank_in_iter == 0) my[d] = base + (uint32_t)__popcll(match);
+      }
+      digrank[i] = (d << 16) | (valid ? r : 0);
+    }
+  } else
 #pragma unroll
   for (int c = 0; c < IT / 2; ++c) {
     const int i0 = 2 * c, i1 = 2 * c + 1;
@@ -591,9 +625,21 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
     uint64_t e1 = chunk + (uint64_t)i1 * kWave + lane;
     bool val0 = e0 < n, val1 = e1 < n;
     // sort_word selects WHICH u64 of the AoS pair orders this pass
-    // (the 80-bit-key path sorts 16 aux bits before the 64 prefix bits)
-    uint64_t sk0 = (AOS && sort_word) ? val_reg[i0] : key_reg[i0];
-    uint64_t sk1 = (AOS && sort_word) ? val_reg[i1] : key_reg[i1];
+    // (the 80-bit-key path sorts 16 aux bits before the 64 prefix bits).
+    // LEAN keeps NO register arrays: the sort word loads here and the
+    // exchange re-reads the pair from the L2-hot tile — the VGPR budget
+    // is what buys the 3rd resident block per CU.
+    uint64_t sk0, sk1;
+    if (AOS && LEAN) {
+      const int sw = sort_word ? 1 : 0;
+      sk0 = val0 ? keys[2 * e0 + sw] : 0;
+      sk1 = val1 ? keys[2 * e1 + sw] : 0;
+    } else {
+      sk0 = (AOS && !LEAN && sort_word)
+          ? val_reg[(AOS && !LEAN) ? i0 : 0] : key_reg[(AOS && LEAN) ? 0 : i0];
+      sk1 = (AOS && !LEAN && sort_word)
+          ? val_reg[(AOS && !LEAN) ? i1 : 0] : key_reg[(AOS && LEAN) ? 0 : i1];
+    }
     uint32_t d0 = (uint32_t)(sk0 >> shift) & (ND - 1);
     uint32_t d1 = (uint32_t)(sk1 >> shift) & (ND - 1);
     uint64_t vm0 = __ballot(val0), vm1 = __ballot(val1);
@@ -670,7 +716,7 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
   if (AOS) {
     uint64_t t2 = t1, t3 = t1;
     const uint32_t off_mask = stage < 3 ? 1023u : 0xFFFFFFFFu;
-    const int rounds = split_exch ? 2 : 1;
+    const int rounds = split_exch > 1 ? split_exch : 1;
     for (int round = 0; round < rounds; ++round) {
       if (round) __syncthreads();     // previous half's writeout done
       if (stage >= 2) {
@@ -682,8 +728,13 @@ __global__ __launch_bounds__(BS) void onesweep_pass_kernel(
             uint32_t d = digrank[i] >> 16;
             uint32_t j = start[d] + my[d] + (digrank[i] & 0xFFFF);
             uint32_t jl = j - (uint32_t)(round * ecap);
-            if (jl < (uint32_t)ecap)
-              exch2[jl] = u64x2{key_reg[i], val_reg[i]};
+            if (jl < (uint32_t)ecap) {
+              if (LEAN)
+                exch2[jl] = reinterpret_cast<const u64x2*>(keys)[e];
+              else
+                exch2[jl] = u64x2{key_reg[i],
+                                  val_reg[(AOS && !LEAN) ? i : 0]};
+            }
           }
         }
       }
@@ -1209,6 +1260,9 @@ void set_lookback_mode(int m) { g_lb_mode = m; }
 // blocks/CU stay resident and hide more of the lookback wait
 static int g_split_exch = 0;
 void set_split_exchange(int m) { g_split_exch = m; }
+// AoS register-lean variant (re-reads pairs at exchange): 3 blocks/CU
+static int g_lean = 0;
+void set_lean_pass(int m) { g_lean = m; }
 
 static inline uint32_t os_num_tiles_t(uint32_t n, int tile) {
   return (uint32_t)(((uint64_t)n + tile - 1) / tile);
@@ -1257,7 +1311,8 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
   }
   size_t lds_soa = (size_t)BLOCK * OS_ITEMS * 8 + (size_t)NW * PD * 4 +
                    PD * 4 * 2 + BLOCK * 4 + 16 + BLOCK * OS_ITEMS;
-  size_t lds_aos = (size_t)aos_tile * (g_split_exch ? 8 : 16) +
+  const int se = g_split_exch > 1 ? g_split_exch : 1;
+  size_t lds_aos = (size_t)aos_tile * 16 / se +
                    (size_t)(512 / kWave) * PD * 4 + PD * 4 * 2 + 512 * 4 + 16;
   size_t lds = aos ? lds_aos : lds_soa;
   static bool attr_set = false;
@@ -1272,7 +1327,9 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
           reinterpret_cast<const void*>(
               &onesweep_pass_kernel<true, 8, true, 512, PBITS>),
           reinterpret_cast<const void*>(
-              &onesweep_pass_kernel<true, 16, true, 512, PBITS>)})
+              &onesweep_pass_kernel<true, 16, true, 512, PBITS>),
+          reinterpret_cast<const void*>(
+              &onesweep_pass_kernel<true, 8, true, 512, PBITS, true>)})
       (void)hipFuncSetAttribute(f, hipFuncAttributeMaxDynamicSharedMemorySize,
                                 160 * 1024 - 1024);
     attr_set = true;
@@ -1315,6 +1372,13 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
                          sb, desc, ticket, key_dst, val_dst, g_pass_stage,
                          g_timing_buf, pass_pref, sort_word, g_lb_mode,
                          g_split_exch);
+    } else if (aos && g_lean) {
+      hipLaunchKernelGGL(
+          (onesweep_pass_kernel<true, 8, true, 512, PBITS, true>),
+          dim3(nb), dim3(512), lds, s,
+          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
+          sb, desc, ticket, key_dst, val_dst, g_pass_stage,
+          g_timing_buf, pass_pref, sort_word, g_lb_mode, g_split_exch);
     } else if (aos) {
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 8, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
