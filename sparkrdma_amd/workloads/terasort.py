@@ -165,11 +165,12 @@ class TeraSort:
         lo, hi = rank * self.ppe, (rank + 1) * self.ppe - 1
         # pipelined reduce: H chunk readers start fetching concurrently at
         # construction; chunk h sorts while later chunks' one-sided copies
-        # are still in flight (overlaps xGMI fetch with the radix sort —
-        # only pays off when fetches are remote, i.e. multi-GPU)
-        # chunking shrinks each chunk's shared top bits, which can add a
-        # whole radix pass; overlap only pays when fetches are remote
-        H = 1 if eng.world_size == 1 else min(4, self.ppe)
+        # are still in flight — overlapping the fetch (xGMI remote, or
+        # ~30 ms of local D2D at 40 GB) with the radix sort. Chunking
+        # GROWS the shared top bits (per-chunk pid ranges are pow2-
+        # aligned), so it never adds a radix pass.
+        import os as _os
+        H = int(_os.environ.get("TERASORT_H", 0)) or min(4, self.ppe)
         # ppe is pow2, so per = ppe/H is exact
         per = self.ppe // H
         spans = [(lo + h * per, lo + (h + 1) * per - 1) for h in range(H)]
