@@ -251,6 +251,9 @@ class ShuffleWriter:
         func, shift, nparts, nbits_eff = self._gpu_part_params(partitioner)
         has_val = vals is not None
         n = keys.numel()
+        if n == 0:
+            self._commit_empty()
+            return
         dev = keys.device
         stream = torch.cuda.current_stream().cuda_stream
         nd = 1 << nbits_eff
@@ -460,6 +463,9 @@ class ShuffleWriter:
         batches = self._gpu_records
         recs = (batches[0] if len(batches) == 1 else torch.cat(batches))
         n = recs.numel() // W
+        if n == 0:
+            self._commit_empty()
+            return
         dev = recs.device
         stream = torch.cuda.current_stream().cuda_stream
         two_level = (R - 1).bit_length() > 12
@@ -580,6 +586,17 @@ class ShuffleWriter:
                 blocks.append(hblk)
         self.metrics.records_written += n
         mgr.keep_alive(self.handle, self.map_id, blocks)
+        mgr.publish_map_output(self.handle, self.map_id, table_addr)
+
+    def _commit_empty(self) -> None:
+        """Zero-record map task: publish an all-empty location table."""
+        mgr = self.manager
+        R = self.handle.num_partitions
+        table, table_addr = mgr.alloc_table(R)
+        meta_key = make_key(mgr.executor_id, 1)
+        for p in range(R):
+            table.put(p, 0, 0, meta_key)
+        mgr.keep_alive(self.handle, self.map_id, [])
         mgr.publish_map_output(self.handle, self.map_id, table_addr)
 
     def _commit_segments(self, segments: List[bytes]) -> None:
