@@ -165,12 +165,14 @@ class TeraSort:
         lo, hi = rank * self.ppe, (rank + 1) * self.ppe - 1
         # pipelined reduce: H chunk readers start fetching concurrently at
         # construction; chunk h sorts while later chunks' one-sided copies
-        # are still in flight — overlapping the fetch (xGMI remote, or
-        # ~30 ms of local D2D at 40 GB) with the radix sort. Chunking
-        # GROWS the shared top bits (per-chunk pid ranges are pow2-
-        # aligned), so it never adds a radix pass.
+        # are still in flight — overlapping remote xGMI fetches with the
+        # radix sort. Chunking GROWS the shared top bits (per-chunk pid
+        # ranges are pow2-aligned) so it never adds a radix pass; at
+        # world=1 the local D2D fetch is already stream-overlapped and
+        # H=1 measured fastest (253.7 vs 257.8 GB/s — r02 A/B).
         import os as _os
-        H = int(_os.environ.get("TERASORT_H", 0)) or min(4, self.ppe)
+        H = int(_os.environ.get("TERASORT_H", 0)) or \
+            (1 if eng.world_size == 1 else min(4, self.ppe))
         # ppe is pow2, so per = ppe/H is exact
         per = self.ppe // H
         spans = [(lo + h * per, lo + (h + 1) * per - 1) for h in range(H)]
