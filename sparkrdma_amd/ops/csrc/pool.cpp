@@ -117,12 +117,12 @@ void enable_peer_access(int peer_device) {
 //     destroyed per batch — the analog of the reference caching its
 //     serialized verbs objects (RdmaChannel.java:43-44,370-420).
 //   * `peer` is the peer's GPU ordinal (the xGMI link selector), not an
-//     executor id; two streams per peer rotate so two SDMA engines can
-//     drive one link concurrently while different peers' copies never
-//     serialize behind each other.
+//     executor id; four streams per peer rotate so several SDMA engines
+//     can drive one link/local-HBM concurrently while different peers'
+//     copies never serialize behind each other.
 
 struct PeerEngine {
-  static constexpr int kStreams = 2;
+  static constexpr int kStreams = 4;
   hipStream_t streams[kStreams] = {};
   unsigned next = 0;
   std::mutex mu;
