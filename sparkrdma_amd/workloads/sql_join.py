@@ -22,7 +22,6 @@ import numpy as np
 
 from ..engine import Engine
 from ..partitioner import RangePartitioner
-from ..writer import unpack_partition_segment
 
 
 @dataclass
@@ -68,7 +67,11 @@ class SortMergeJoin:
             self.a_vals = ka.view(np.uint8).reshape(-1, 8).copy()
             self.b_vals = kb.view(np.uint8).reshape(-1, 8).copy()
 
-    def _shuffle(self, keys, vals):
+    def _shuffle_sorted(self, keys, vals):
+        """One table's shuffle, consumed through the reader's GENERIC
+        ordering hookup (read_aos(ordering=True) — the shared
+        deserialize->sort path, RdmaShuffleReader.scala:61-114 role):
+        returns (handle, sorted keys, sorted vals, metrics)."""
         eng = self.engine
         handle = eng.register_shuffle(eng.world_size, self.R)
         w = eng.manager.get_writer(handle, eng.rank)
@@ -92,55 +95,38 @@ class SortMergeJoin:
                 arena_hint = torch.empty(cap, dtype=torch.uint8,
                                          device="cuda")
                 self._arena_pool.append(arena_hint)
+            # table A's data must survive table B's shuffle
             arena_hint._in_use = True
         reader = eng.manager.get_reader(handle, lo, hi, arena=arena_hint)
-        parts = reader.collect_partitions()
-        # arena stays marked in-use until run_step() ends — table A's data
-        # must survive table B's shuffle
-        return handle, parts, reader.metrics, \
-            getattr(reader.fetcher, "arena", None)
+        # the owned partition range shares its top log2(W) key bits, so
+        # ONE batched sort over the low bits fully orders each side
+        wbits = (eng.world_size - 1).bit_length()
+        k, v = reader.read_aos(ordering=True,
+                               end_bit=self.key_bits - wbits)
+        return handle, k, v, reader.metrics
 
     def run_step(self) -> JoinResult:
         eng = self.engine
         t0 = time.perf_counter()
-        ha, parts_a, ma, arena_a = self._shuffle(self.a_keys, self.a_vals)
-        hb, parts_b, mb, arena_b = self._shuffle(self.b_keys, self.b_vals)
+        ha, ak, av, ma = self._shuffle_sorted(self.a_keys, self.a_vals)
+        hb, bk, bv, mb = self._shuffle_sorted(self.b_keys, self.b_vals)
         matches = 0
         if self.device == "cuda":
             import torch
             from ..ops.join import merge_join_sorted
-            from ..ops.radix import sort_pairs_aos
-            # batched: the whole owned partition range shares its top
-            # log2(W) key bits, so one AoS sort per side + one merge
-            # replaces ppe per-partition sorts (same trick as TeraSort)
-            wbits = (self.engine.world_size - 1).bit_length()
-            sort_bits = self.key_bits - wbits
-            a_chunks = ([arena_a.view(torch.int64)] if arena_a is not None
-                        else [c for p in parts_a for c in parts_a[p]])
-            b_chunks = ([arena_b.view(torch.int64)] if arena_b is not None
-                        else [c for p in parts_b for c in parts_b[p]])
-            ak, av = self._sort_side(a_chunks, sort_bits)
-            bk, bv = self._sort_side(b_chunks, sort_bits)
-            if ak is not None and bk is not None:
-                jk, ja, jb = merge_join_sorted(ak, av, bk, bv)
+            if ak.numel() and bk.numel():
+                jk, ja, jb = merge_join_sorted(
+                    ak.contiguous(), av.contiguous(),
+                    bk.contiguous(), bv.contiguous())
                 matches += jk.numel()
                 if self.validate and jk.numel():
                     assert torch.equal(jk, ja) and torch.equal(jk, jb), \
                         "joined payloads must equal keys"
             torch.cuda.synchronize()
-        else:
-            for p in parts_a:
-                ak, av = self._concat_cpu(parts_a[p])
-                bk, bv = self._concat_cpu(parts_b[p])
-                if ak is None or bk is None:
-                    continue
-                ao = np.argsort(ak, kind="stable")
-                bo = np.argsort(bk, kind="stable")
-                ak = ak[ao]
-                bk = bk[bo]
-                lo = np.searchsorted(bk, ak, side="left")
-                hi = np.searchsorted(bk, ak, side="right")
-                matches += int((hi - lo).sum())
+        elif len(ak) and len(bk):
+            lo = np.searchsorted(bk, ak, side="left")
+            hi = np.searchsorted(bk, ak, side="right")
+            matches += int((hi - lo).sum())
         eng.unregister_shuffle(ha)
         eng.unregister_shuffle(hb)
         for t in getattr(self, "_arena_pool", []):
@@ -150,29 +136,6 @@ class SortMergeJoin:
             self._validate_counts(matches)
         return JoinResult(dt, self.n, self.n, matches,
                           ma.remote_bytes_read + mb.remote_bytes_read)
-
-    @staticmethod
-    def _sort_side(chunks, sort_bits):
-        import torch
-        from ..ops.radix import sort_pairs_aos
-        from ..utils import as_device_i64
-        ts = [as_device_i64(c) for c in chunks]  # AoS records
-        if not ts:
-            return None, None
-        pairs = torch.cat(ts) if len(ts) > 1 else ts[0].contiguous()
-        pairs = sort_pairs_aos(pairs, 0, sort_bits)
-        return pairs[0::2].contiguous(), pairs[1::2].contiguous()
-
-    @staticmethod
-    def _concat_cpu(chunks):
-        ks = []
-        for c in chunks:
-            k, _ = unpack_partition_segment(c, 8)
-            ks.append(np.array(k))
-        if not ks:
-            return None, None
-        k = np.concatenate(ks)
-        return k, None
 
     def _validate_counts(self, matches: int) -> None:
         """Local-count oracle only works single-process; multi-rank runs
