@@ -76,6 +76,12 @@ PYBIND11_MODULE(_hipshuffle, m) {
         py::arg("start_bit"), py::arg("end_bit"), py::arg("ws"),
         py::arg("stream") = 0, py::arg("sort_word") = 0,
         py::call_guard<py::gil_scoped_release>());
+  m.def("onesweep_sort_aos_fused_u64", &hs::onesweep_sort_aos_fused_u64,
+        py::arg("pairs"), py::arg("tmp_pairs"), py::arg("n"),
+        py::arg("start_bit"), py::arg("end_bit"), py::arg("ws"),
+        py::arg("stream") = 0, py::arg("sort_word") = 0,
+        py::arg("recs") = 0, py::arg("out") = 0, py::arg("rec_bytes") = 0,
+        py::call_guard<py::gil_scoped_release>());
   m.def("sort_workspace_bytes", &hs::sort_workspace_bytes);
   m.def("sort_pairs_u64", &hs::sort_pairs_u64, py::arg("keys"),
         py::arg("vals"), py::arg("tmp_keys"), py::arg("tmp_vals"),

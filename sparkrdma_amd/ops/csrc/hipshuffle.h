@@ -80,6 +80,11 @@ int onesweep_sort_aos_word_u64(uintptr_t pairs, uintptr_t tmp_pairs,
                                uint32_t n, int start_bit, int end_bit,
                                uintptr_t ws, uintptr_t stream,
                                int sort_word);
+int onesweep_sort_aos_fused_u64(uintptr_t pairs, uintptr_t tmp_pairs,
+                                uint32_t n, int start_bit, int end_bit,
+                                uintptr_t ws, uintptr_t stream,
+                                int sort_word, uintptr_t recs,
+                                uintptr_t out, uint32_t rec_bytes);
 int onesweep_sort_pairs_u64(uintptr_t keys, uintptr_t vals,
                             uintptr_t tmp_keys, uintptr_t tmp_vals,
                             uint32_t n, int start_bit, int end_bit,

@@ -521,7 +521,7 @@ __global__ __launch_bounds__(BLOCK) void onesweep_hist_all_kernel(
 // full-tile burst lengths (the 2048-tile route paid burst halving +
 // 2x descriptors for its occupancy; this pays only L1/L2 re-reads).
 template <bool HAS_VAL, int IT, bool AOS, int BS = BLOCK, int PBITS = 8,
-          bool LEAN = false>
+          bool LEAN = false, bool FUSE_GATHER = false>
 __global__ __launch_bounds__(BS)
 __attribute__((amdgpu_waves_per_eu(LEAN ? 6 : 1))) void onesweep_pass_kernel(
     const uint64_t* __restrict__ keys, const uint64_t* __restrict__ vals,
@@ -531,10 +531,15 @@ __attribute__((amdgpu_waves_per_eu(LEAN ? 6 : 1))) void onesweep_pass_kernel(
     int stage = 3, uint64_t* __restrict__ timing = nullptr,
     const uint32_t* __restrict__ hist_pref = nullptr, int sort_word = 0,
     int lb_mode = 0 /* 0: [nb][ND] descriptors, 1: transposed [ND][nb] */,
-    int split_exch = 0 /* AoS: exchange+writeout in N sub-tile rounds
-                          (0/1 = single round); the LDS exchange buffer
-                          shrinks by N so more blocks stay resident to
-                          hide the lookback wait */) {
+    int split_exch = 0, /* AoS: exchange+writeout in N sub-tile rounds
+                           (0/1 = single round); the LDS exchange buffer
+                           shrinks by N so more blocks stay resident to
+                           hide the lookback wait */
+    const uint32_t* __restrict__ rec_words = nullptr, /* FUSE_GATHER */
+    uint32_t rec_w4 = 0 /* record words; key_dst then holds per-digit
+                           RECORD base addresses and the writeout copies
+                           whole records instead of pairs — the final
+                           sort pass IS the gather */) {
   constexpr int ND = 1 << PBITS;
   extern __shared__ char smem_raw[];
   constexpr int TILE_T = BS * IT;
@@ -750,6 +755,23 @@ __attribute__((amdgpu_waves_per_eu(LEAN ? 6 : 1))) void onesweep_pass_kernel(
       }
       __syncthreads();
       if (round == 0) t3 = timing ? __builtin_amdgcn_s_memrealtime() : 0;
+      if (FUSE_GATHER) {
+        // copy each slot's RECORD to its final position: key_dst[d]
+        // holds per-digit record bases; word-mapped like gather_records
+        // (writes coalesce within digit runs, reads are the random
+        // per-record chunks the separate gather paid anyway)
+        const uint32_t total_w = tile_n * rec_w4;
+        for (uint32_t w = tid; w < total_w; w += BS) {
+          uint32_t j = w / rec_w4, o = w - j * rec_w4;
+          u64x2 kv = exch2[j];
+          uint32_t d = (uint32_t)((uint64_t)(sort_word ? kv.y : kv.x)
+                                  >> shift) & (ND - 1);
+          uint32_t off = (pref[d] + (j - start[d])) & off_mask;
+          uint64_t idx = (uint64_t)kv.y & ((1ull << 48) - 1);
+          reinterpret_cast<uint32_t*>(key_dst[d])[(uint64_t)off * rec_w4 + o] =
+              rec_words[idx * rec_w4 + o];
+        }
+      } else {
 #pragma unroll
       for (int i = 0; i < IT; ++i) {
         uint32_t jl = (uint32_t)i * BS + tid;
@@ -762,6 +784,7 @@ __attribute__((amdgpu_waves_per_eu(LEAN ? 6 : 1))) void onesweep_pass_kernel(
           uint32_t off = (pref[d] + (j - start[d])) & off_mask;
           reinterpret_cast<u64x2*>(key_dst[d])[off] = kv;
         }
+      }
       }
     }
     if (timing && tid == 0) {
@@ -1291,7 +1314,9 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
                               uintptr_t tmp_keys, uintptr_t tmp_vals,
                               uint32_t n, int start_bit, int end_bit,
                               uintptr_t ws, hipStream_t s, int aos,
-                              int sort_word = 0) {
+                              int sort_word = 0, uintptr_t fuse_recs = 0,
+                              uintptr_t fuse_out = 0,
+                              uint32_t fuse_rec_bytes = 0) {
   constexpr int PD = 1 << PBITS;
   const int aos_tile = g_aos_tile;
   uint32_t nb = aos ? os_num_tiles_t(n, aos_tile) : os_num_tiles(n);
@@ -1332,7 +1357,10 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
           reinterpret_cast<const void*>(
               &onesweep_pass_kernel<true, 16, true, 512, PBITS>),
           reinterpret_cast<const void*>(
-              &onesweep_pass_kernel<true, 8, true, 512, PBITS, true>)})
+              &onesweep_pass_kernel<true, 8, true, 512, PBITS, true>),
+          reinterpret_cast<const void*>(
+              &onesweep_pass_kernel<true, 8, true, 512, PBITS, false,
+                                    true>)})
       (void)hipFuncSetAttribute(f, hipFuncAttributeMaxDynamicSharedMemorySize,
                                 160 * 1024 - 1024);
     attr_set = true;
@@ -1346,6 +1374,10 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
   for (int p = 0; p < passes; ++p) {
     int sb = start_bit + p * PBITS;
     if (sb > 64 - PBITS) sb = 64 - PBITS;  // same clamp as hist_all
+    // fused final pass: the writeout copies whole records to fuse_out
+    // (the gather folded into the sort — key_dst becomes record bases)
+    const bool fuse = fuse_recs && p == passes - 1 && aos &&
+                      aos_tile == 4096 && g_split_exch <= 1 && !g_lean;
     if (use_hist) {
       hist_launch<PBITS>(reinterpret_cast<const uint64_t*>(src_k), n, sb,
                          hist32, s, 0, 2);
@@ -1353,15 +1385,28 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
                          s);
     }
     hipLaunchKernelGGL((onesweep_digit_bases_kernel<PD>), dim3(1),
-                       dim3(BLOCK), 0, s, totals, p, (uint64_t)dst_k,
-                       (uint64_t)dst_v, key_dst, val_dst, aos ? 16 : 8);
+                       dim3(BLOCK), 0, s, totals, p,
+                       fuse ? (uint64_t)fuse_out : (uint64_t)dst_k,
+                       fuse ? 0 : (uint64_t)dst_v, key_dst, val_dst,
+                       fuse ? (int)fuse_rec_bytes : (aos ? 16 : 8));
     HIP_CHECK(hipGetLastError());
     if (!use_hist) {
       HIP_CHECK(hipMemsetAsync(ticket, 0, 16, s));
       HIP_CHECK(hipMemsetAsync(desc, 0, (size_t)nb * PD * 8, s));
     }
     const uint32_t* pass_pref = use_hist ? hist32 : nullptr;
-    if (aos && aos_tile == 8192) {
+    if (fuse) {   // (ticket/desc were reset in the !use_hist block)
+      hipLaunchKernelGGL(
+          (onesweep_pass_kernel<true, 8, true, 512, PBITS, false, true>),
+          dim3(nb), dim3(512), lds, s,
+          reinterpret_cast<const uint64_t*>(src_k), nullptr, n, sb, desc,
+          ticket, key_dst, val_dst, g_pass_stage, nullptr, nullptr,
+          sort_word, g_lb_mode, 0,
+          reinterpret_cast<const uint32_t*>(fuse_recs),
+          fuse_rec_bytes / 4);
+      HIP_CHECK(hipGetLastError());
+      return cur;  // records land in fuse_out; pair buffers are dead
+    } else if (aos && aos_tile == 8192) {
       hipLaunchKernelGGL((onesweep_pass_kernel<true, 16, true, 512, PBITS>),
                          dim3(nb), dim3(512), lds, s,
                          reinterpret_cast<const uint64_t*>(src_k), nullptr, n,
@@ -1408,6 +1453,9 @@ static int onesweep_sort_tmpl(uintptr_t keys, uintptr_t vals,
     std::swap(src_v, dst_v);
     cur ^= 1;
   }
+  // a requested fusion that no pass applied (non-default tile/ablation
+  // modes) must NOT silently leave `out` unwritten — caller falls back
+  if (fuse_recs) return -1;
   return cur;
 }
 
@@ -1437,6 +1485,19 @@ int onesweep_sort_aos_word_u64(uintptr_t pairs, uintptr_t tmp_pairs,
   return onesweep_sort_tmpl<8>(pairs, 0, tmp_pairs, 0, n, start_bit, end_bit,
                                ws, reinterpret_cast<hipStream_t>(stream), 1,
                                sort_word);
+}
+
+// sort + FUSED final gather: records land sorted in `out` (the last
+// pass's writeout copies whole records; the separate gather kernel and
+// the final pair round trip disappear).
+int onesweep_sort_aos_fused_u64(uintptr_t pairs, uintptr_t tmp_pairs,
+                                uint32_t n, int start_bit, int end_bit,
+                                uintptr_t ws, uintptr_t stream,
+                                int sort_word, uintptr_t recs,
+                                uintptr_t out, uint32_t rec_bytes) {
+  return onesweep_sort_tmpl<8>(pairs, 0, tmp_pairs, 0, n, start_bit, end_bit,
+                               ws, reinterpret_cast<hipStream_t>(stream), 1,
+                               sort_word, recs, out, rec_bytes);
 }
 
 int onesweep_sort_aos7_u64(uintptr_t pairs, uintptr_t tmp_pairs, uint32_t n,
@@ -1472,6 +1533,9 @@ int sort_pairs_u64(uintptr_t keys, uintptr_t vals, uintptr_t tmp_keys,
     std::swap(src_v, dst_v);
     cur ^= 1;
   }
+  // a requested fusion that no pass applied (non-default tile/ablation
+  // modes) must NOT silently leave `out` unwritten — caller falls back
+  if (fuse_recs) return -1;
   return cur;
 }
 

@@ -166,7 +166,8 @@ def sort_records(recs: torch.Tensor, rec_bytes: int, key_bytes: int = 8,
                  end_bit: int = 64, out: Optional[torch.Tensor] = None,
                  pairs: Optional[torch.Tensor] = None,
                  tmp: Optional[torch.Tensor] = None,
-                 ws: Optional[torch.Tensor] = None) -> torch.Tensor:
+                 ws: Optional[torch.Tensor] = None,
+                 fuse: bool = True) -> torch.Tensor:
     """Sort W-byte AoS records by their 80-bit key: u64 LE prefix at
     offset 0 (bits [0, end_bit) significant — callers whose partitions
     share top prefix bits pass end_bit = 64 - shared) then, for
@@ -202,16 +203,27 @@ def sort_records(recs: torch.Tensor, rec_bytes: int, key_bytes: int = 8,
                                          _stream(), 1)
         if r == 1:
             cur, other = other, cur
-    r = m.onesweep_sort_aos_word_u64(cur.data_ptr(), other.data_ptr(), n,
-                                     0, min(end_bit, 64), ws.data_ptr(),
-                                     _stream(), 0)
-    if r == 1:
-        cur, other = other, cur
     if out is None:
         out = torch.empty_like(recs)
     else:
         assert out.numel() >= recs.numel()
         out = out[:recs.numel()]
+    if fuse:
+        # the FINAL prefix pass writes whole records (kernels.hip
+        # FUSE_GATHER): the last pair writeout + the separate gather's
+        # pair re-read disappear. -1 = fusion not applicable under the
+        # current ablation globals -> fall through to the plain path
+        res = m.onesweep_sort_aos_fused_u64(
+            cur.data_ptr(), other.data_ptr(), n, 0, min(end_bit, 64),
+            ws.data_ptr(), _stream(), 0, recs.data_ptr(),
+            out.data_ptr(), rec_bytes)
+        if res != -1:
+            return out
+    r = m.onesweep_sort_aos_word_u64(cur.data_ptr(), other.data_ptr(), n,
+                                     0, min(end_bit, 64), ws.data_ptr(),
+                                     _stream(), 0)
+    if r == 1:
+        cur, other = other, cur
     m.gather_records(recs.data_ptr(), cur.data_ptr(), n, rec_bytes, 0,
                      out.data_ptr(), 0, 0, 0, 0, 0, 0, _stream())
     return out
