@@ -1533,9 +1533,6 @@ int sort_pairs_u64(uintptr_t keys, uintptr_t vals, uintptr_t tmp_keys,
     std::swap(src_v, dst_v);
     cur ^= 1;
   }
-  // a requested fusion that no pass applied (non-default tile/ablation
-  // modes) must NOT silently leave `out` unwritten — caller falls back
-  if (fuse_recs) return -1;
   return cur;
 }
 
