@@ -201,7 +201,7 @@ def main():
             "config": {
                 "model": args.workload,
                 "global_batch": n_rec * world,
-                "seq_len": ts.RECORD_BYTES,
+                "seq_len": getattr(ts, "RECORD_BYTES", 16),
                 "dataset_gb": round(total_bytes / (1 << 30), 1),
                 "partitions": world * ppe,
                 "mode": getattr(ts, "mode", args.mode),

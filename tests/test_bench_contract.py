@@ -33,3 +33,26 @@ def test_bench_json_contract(tmp_path):
     cfg = j["config"]
     for field in ("model", "global_batch", "partitions", "parallelism"):
         assert field in cfg, field
+
+
+import pytest
+
+
+@pytest.mark.parametrize("workload", ["pagerank", "join", "reducebykey",
+                                      "groupby"])
+def test_bench_all_workloads_emit_json(tmp_path, workload):
+    """Every BASELINE workload's bench path runs end-to-end on CPU and
+    emits the JSON line (guards attr regressions like r02's seq_len)."""
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--cpu",
+         "--workload", workload, "--steps", "1", "--warmup", "0",
+         "--gb-per-gpu", "0.002", "--partitions-per-executor", "16"],
+        capture_output=True, text=True, timeout=600, env=env, cwd=REPO)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    j = json.loads(lines[0])
+    assert j["value"] > 0 and j["config"]["model"] == workload
