@@ -675,6 +675,10 @@ class ShuffleReader:
         import io
         import pickle
         for _ref, data in self.fetcher:
+            if not isinstance(data, (bytes, bytearray, memoryview)):
+                # arena mode landed the bytes in the device arena: one
+                # bulk D2H (bytes() on a CUDA tensor would sync per byte)
+                data = data.cpu().numpy().tobytes()
             bio = io.BytesIO(bytes(data))
             end = len(bio.getvalue())
             while bio.tell() < end:

@@ -96,3 +96,17 @@ def test_reduce_by_key_gpu(engine):
                     partitions_per_executor=32, device="cuda", validate=True)
     res = r.run_step()
     assert res.groups == 3000
+
+
+def test_groupby_pickled_lane_on_gpu_plane(engine):
+    """The pickled-record lane with the GPU plane active: host-byte
+    blocks land in the device arena and read_records bulk-copies them
+    back (the r02 per-byte-sync hang regression guard)."""
+    import time
+    from sparkrdma_amd.workloads.groupby import GroupByKey
+    g = GroupByKey(engine, rows_per_executor=20_000)
+    t0 = time.perf_counter()
+    r = g.run_step()
+    dt = time.perf_counter() - t0
+    assert r.rows == 20_000 and r.groups > 0
+    assert dt < 60, f"pickled lane pathologically slow: {dt:.1f}s"
