@@ -152,3 +152,101 @@ def test_multiprocess_shuffle(tmp_path, world_size):
     assert np.array_equal(np.sort(all_keys), want)
     # with >1 executors some bytes must have crossed processes
     assert sum(results[r][1] for r in results) > 0
+
+
+def _crash_worker(rank, world, driver_port, shm_dir, q):
+    """Rank `world-1` publishes, wipes its served segments, and dies
+    (node-loss simulation); survivors must surface FetchFailedError and
+    see the driver prune membership."""
+    try:
+        import glob
+        import sys
+        import time
+        sys.path.insert(0, os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))))
+        from sparkrdma_amd.conf import ShuffleConf
+        from sparkrdma_amd.engine import Engine
+        from sparkrdma_amd.partitioner import RangePartitioner
+        from sparkrdma_amd.reader import FetchFailedError
+
+        conf = ShuffleConf(shm_dir=shm_dir, max_buffer_allocation_size=1 << 30)
+        eng = Engine(conf, rank=rank, world_size=world,
+                     driver_port=driver_port)
+        R = world * 2
+        part = RangePartitioner.uniform(R, key_max=2 ** 32 - 1)
+        handle = eng.register_shuffle(num_maps=world, num_partitions=R)
+        rng = np.random.default_rng(rank)
+        keys = rng.integers(0, 2 ** 32, 5000, dtype=np.uint64)
+        w = eng.manager.get_writer(handle, rank)
+        w.write_batch(keys, keys.view(np.uint8).reshape(-1, 8).copy())
+        w.stop(True, partitioner=part)
+        eng.barrier()
+        crash_rank = world - 1
+        if rank == crash_rank:
+            # simulate abrupt node loss AFTER publish: data segments
+            # vanish, process dies without any teardown
+            for p in glob.glob(os.path.join(
+                    shm_dir, f"sparkrdma_{eng.manager.app_id}_e{rank}_s*")):
+                if not p.endswith("_s1"):   # keep metadata, kill data
+                    os.unlink(p)
+            q.put((rank, "crashed"))
+            q.close()
+            q.join_thread()   # flush before dying (else the msg is lost)
+            os._exit(1)
+        time.sleep(0.5)   # let the crash + segment wipe land
+        try:
+            reader = eng.manager.get_reader(handle, rank * 2, rank * 2 + 1)
+            list(reader)
+            q.put((rank, "NO-ERROR"))
+        except FetchFailedError:
+            q.put((rank, "fetch-failed"))
+        # driver prunes the lost executor and re-announces
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline:
+            if crash_rank not in eng.manager._members:
+                break
+            time.sleep(0.05)
+        q.put((rank, "members", sorted(eng.manager._members)))
+        eng.manager.stop()
+        if eng.driver is not None:
+            eng.driver.stop()
+    except BaseException as e:
+        import traceback
+        q.put((rank, f"ERROR: {e}\n{traceback.format_exc()}"))
+        raise
+
+
+def test_executor_crash_mid_shuffle(tmp_path):
+    """Node-loss semantics parity: a peer dying after publish fails the
+    fetching task (FetchFailedError ≡ Spark FetchFailedException) and the
+    driver prunes + re-announces membership (reference
+    RdmaShuffleManager.scala:155-165, driver.py finally-block)."""
+    import socket
+    world = 3
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    procs = [ctx.Process(target=_crash_worker,
+                         args=(r, world, port, str(tmp_path), q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    msgs = []
+    for _ in range(1 + 2 * (world - 1)):   # crash note + 2 per survivor
+        msgs.append(q.get(timeout=120))
+    for m in msgs:
+        assert not (len(m) == 2 and str(m[1]).startswith("ERROR")), m
+    outcomes = {m[0]: m[1] for m in msgs if len(m) == 2}
+    assert outcomes[world - 1] == "crashed"
+    for r in range(world - 1):
+        assert outcomes[r] == "fetch-failed", outcomes
+    members = {m[0]: m[2] for m in msgs if len(m) == 3}
+    for r in range(world - 1):
+        assert world - 1 not in members[r], \
+            f"rank {r} still sees the dead executor: {members[r]}"
+    for p in procs[:world - 1]:
+        p.join(timeout=60)
+    procs[world - 1].join(timeout=60)
