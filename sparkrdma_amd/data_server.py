@@ -37,6 +37,25 @@ from .map_output import split_key
 
 log = logging.getLogger(__name__)
 
+
+def _native():
+    """The hipshuffle extension's GIL-free socket helpers (None when the
+    extension is unavailable — pure-Python loops then serve)."""
+    global _NATIVE
+    if _NATIVE is False:
+        return None
+    if _NATIVE is None:
+        try:
+            from .ops import load
+            _NATIVE = load(build_if_missing=False)
+        except Exception:
+            _NATIVE = False
+            return None
+    return _NATIVE
+
+
+_NATIVE = None
+
 _REQ = struct.Struct("<IQQ")
 _RSP = struct.Struct("<qI")
 _CHUNK = struct.Struct("<II")
@@ -194,6 +213,14 @@ class DataServer:
 
         def send_slot(idx, nbytes):
             hs.wait_event(evs[idx])
+            if not (self._flags & FLAG_DEFLATE):
+                # GIL-free raw path: header + payload straight from the
+                # pinned buffer
+                sock.sendall(_CHUNK.pack(nbytes, nbytes))
+                if hs.tcp_send_all(sock.fileno(), pinned.ptrs[idx],
+                                   nbytes) < 0:
+                    raise ConnectionError("send failed")
+                return
             view = (ctypes.c_char * nbytes).from_address(pinned.ptrs[idx])
             self._send_chunk(sock, view)
 
@@ -252,8 +279,15 @@ class DataClient:
                 self._cv.wait(1.0)
         try:
             sock = socket.create_connection(ep, timeout=30)
+            # BLOCKING mode with a kernel-level receive timeout: python
+            # timeout-mode sockets are O_NONBLOCK underneath, which the
+            # GIL-free native recv loop cannot use; SO_RCVTIMEO keeps the
+            # hang protection for both the native and python paths
+            sock.settimeout(None)
             sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
             try:
+                sock.setsockopt(socket.SOL_SOCKET, socket.SO_RCVTIMEO,
+                                struct.pack("ll", 30, 0))
                 sock.setsockopt(socket.SOL_SOCKET, socket.SO_RCVBUF, 8 << 20)
             except OSError:
                 pass
@@ -290,6 +324,17 @@ class DataClient:
                 self._release(ep, sock, broken=False)
                 raise IOError(f"remote read failed (status {status})")
             out = bytearray(status)
+            nat = _native()
+            if status and not (flags & FLAG_DEFLATE) and nat is not None:
+                # whole chunked body received by the GIL-free C loop
+                import numpy as np
+                addr = np.frombuffer(out, dtype=np.uint8).ctypes.data
+                got = nat.tcp_recv_chunks(sock.fileno(), addr, status)
+                if got != status:
+                    raise ConnectionError(
+                        f"native chunk receive failed ({got}/{status})")
+                self._release(ep, sock, broken=False)
+                return bytes(out)
             mv = memoryview(out)
             off = 0
             while off < status:

@@ -39,6 +39,10 @@ PYBIND11_MODULE(_hipshuffle, m) {
         py::call_guard<py::gil_scoped_release>());
   m.def("wait_event", &hs::wait_event,
         py::call_guard<py::gil_scoped_release>());
+  m.def("tcp_recv_chunks", &hs::tcp_recv_chunks,
+        py::call_guard<py::gil_scoped_release>());
+  m.def("tcp_send_all", &hs::tcp_send_all,
+        py::call_guard<py::gil_scoped_release>());
   m.def("host_alloc_pinned", &hs::host_alloc_pinned);
   m.def("host_free_pinned", &hs::host_free_pinned);
   m.def("memcpy_h2d", &hs::memcpy_h2d,

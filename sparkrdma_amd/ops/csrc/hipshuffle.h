@@ -21,6 +21,8 @@ uint64_t read_batch_ids(int peer, const std::vector<uintptr_t>& dsts,
                         const std::vector<size_t>& sizes);
 bool poll_event(uint64_t id);
 void wait_event(uint64_t id);
+int64_t tcp_recv_chunks(int fd, uintptr_t out, uint64_t total);
+int64_t tcp_send_all(int fd, uintptr_t buf, uint64_t n);
 uintptr_t host_alloc_pinned(size_t n);
 void host_free_pinned(uintptr_t p);
 void memcpy_h2d(uintptr_t dst, uintptr_t src, size_t n);

@@ -14,6 +14,8 @@
 
 #include "common.h"
 
+#include <sys/socket.h>
+
 #include <atomic>
 #include <cstring>
 #include <mutex>
@@ -245,6 +247,53 @@ uintptr_t host_alloc_pinned(size_t n) {
 
 void host_free_pinned(uintptr_t p) {
   HIP_CHECK(hipHostFree(reinterpret_cast<void*>(p)));
+}
+
+// ---------------------------------------------------------------------------
+// GIL-free socket helpers for the cross-host TCP lane: the chunk-framed
+// receive loop and raw sends run entirely in C (plain POSIX fds), so the
+// per-process lane bandwidth is not bound by Python byte handling.
+
+int64_t tcp_recv_chunks(int fd, uintptr_t out, uint64_t total) {
+  // reads [raw_len u32][wire_len u32][payload] frames until `total` raw
+  // bytes landed; only UNCOMPRESSED frames (wire==raw) are handled here
+  // — the deflate path stays in Python. Returns bytes received, or
+  // -1 on EOF/framing error, -2 if a compressed frame appears (caller
+  // must not use this helper when the codec is on).
+  uint8_t* dst = reinterpret_cast<uint8_t*>(out);
+  uint64_t off = 0;
+  while (off < total) {
+    uint32_t hdr[2];
+    size_t got = 0;
+    while (got < sizeof(hdr)) {
+      ssize_t r = ::recv(fd, reinterpret_cast<char*>(hdr) + got,
+                         sizeof(hdr) - got, 0);
+      if (r <= 0) return -1;
+      got += (size_t)r;
+    }
+    uint32_t raw_len = hdr[0], wire_len = hdr[1];
+    if (wire_len != raw_len) return -2;
+    if (off + raw_len > total) return -1;
+    uint64_t done = 0;
+    while (done < raw_len) {
+      ssize_t r = ::recv(fd, dst + off + done, raw_len - done, 0);
+      if (r <= 0) return -1;
+      done += (uint64_t)r;
+    }
+    off += raw_len;
+  }
+  return (int64_t)off;
+}
+
+int64_t tcp_send_all(int fd, uintptr_t buf, uint64_t n) {
+  const uint8_t* p = reinterpret_cast<const uint8_t*>(buf);
+  uint64_t off = 0;
+  while (off < n) {
+    ssize_t r = ::send(fd, p + off, n - off, MSG_NOSIGNAL);
+    if (r <= 0) return -1;
+    off += (uint64_t)r;
+  }
+  return (int64_t)off;
 }
 
 // host<->device staging helpers (bytes path / spill)
