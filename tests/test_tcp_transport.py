@@ -46,6 +46,8 @@ def test_tcp_shuffle_roundtrip(tcp_cluster):
     for i, mgr in enumerate(managers):
         reader = mgr.get_reader(handle, i * 4, i * 4 + 3)
         for ref, data in reader:
+            if not isinstance(data, (bytes, bytearray, memoryview)):
+                data = data.cpu().numpy().tobytes()   # GPU-box arena mode
             k, v = unpack_partition_segment(data, 8)
             assert np.array_equal(np.asarray(k).view(np.uint8).reshape(-1),
                                   np.asarray(v).reshape(-1))
