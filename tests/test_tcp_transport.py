@@ -109,6 +109,8 @@ def test_tcp_compressed_chunked_roundtrip(tcp_cluster_compressed):
     for i, mgr in enumerate(managers):
         reader = mgr.get_reader(handle, i * 2, i * 2 + 1)
         for ref, data in reader:
+            if not isinstance(data, (bytes, bytearray, memoryview)):
+                data = data.cpu().numpy().tobytes()   # GPU-box arena mode
             k, v = unpack_partition_segment(data, 8)
             assert np.array_equal(
                 np.asarray(v)[:, 0],
