@@ -213,8 +213,10 @@ def main():
                 "shuffle_read_gb_per_s": _fetch_rate(results, ts),
                 # honesty label: at N=1 every one-sided read is a local
                 # D2D copy; xGMI rates only appear at N>1
-                "read_locality": ("local_d2d" if world == 1 or remote_gb == 0
-                                  else "xgmi_remote"),
+                "read_locality": (
+                    ("local_d2d" if use_cuda else "local_shm")
+                    if world == 1 or remote_gb == 0
+                    else ("xgmi_remote" if use_cuda else "shm_remote")),
             },
         }))
     eng.shutdown()
