@@ -140,6 +140,10 @@ class GpuDataPlane:
         optimization, never a correctness gate. Slab handles themselves
         are imported continuously by ``importer_tick`` (slabs appear as
         writers commit, long after Announce)."""
+        # background thread: pin OUR device first — peer-access enables
+        # and IPC opens bind to the CURRENT device context, and threads
+        # default to device 0 (wrong context on executors using GPU 1-7)
+        self.hs.set_device(self.device)
         todo = {m.executor_id: m for m in members
                 if m.executor_id != self.executor_id}
         deadline = time.monotonic() + deadline_s
@@ -166,6 +170,7 @@ class GpuDataPlane:
         newly published handles — keeps hop-3 issue latency independent of
         first-touch (the reference needs no analog: ibverbs rkeys are
         usable without a per-MR open; hipIpc handles are not)."""
+        self.hs.set_device(self.device)   # importer thread: our context
         for m in members:
             if m.executor_id == self.executor_id:
                 continue
@@ -218,6 +223,9 @@ class GpuDataPlane:
             base = self._peer_bases.get(key)
             if base is not None:
                 return base
+            # hop-2 worker threads reach here: the hipIpcOpenMemHandle
+            # below must run in OUR device context, not thread-default 0
+            self.hs.set_device(self.device)
             exec_id, seg_id = split_key(key)
             slot = seg_id & 0x7FFF
             if exec_id == self.executor_id:
