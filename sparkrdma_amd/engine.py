@@ -106,3 +106,11 @@ class Engine:
         self.manager.stop()
         if self.driver is not None:
             self.driver.stop()
+
+    # context-manager form: `with Engine(conf) as eng: ...` tears the
+    # control plane down on any exit path
+    def __enter__(self) -> "Engine":
+        return self
+
+    def __exit__(self, exc_type, exc, tb) -> None:
+        self.shutdown()
