@@ -90,6 +90,8 @@ class ShuffleManager:
         self._shuffle_outputs: Dict[int, Dict[int, list]] = {}  # sid -> map_id -> blocks
         self.reader_stats = (ShuffleReaderStats(conf)
                              if conf.collect_shuffle_reader_stats else None)
+        from .stats import TaskMetrics
+        self.lifetime_metrics = TaskMetrics()   # executor-lifetime rollup
         self.gpu = None
         self._data_server = None
         self._data_client = None
@@ -317,6 +319,8 @@ class ShuffleManager:
         self._stopped = True
         if self.reader_stats is not None:
             self.reader_stats.print_histograms(log)
+        log.info("executor %d lifetime: %s", self.executor_id,
+                 self.lifetime_metrics.format())
         if self._pool is not None:
             log.info("%s", self._pool.format_stats())
         if self._conn is not None:

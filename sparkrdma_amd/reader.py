@@ -511,6 +511,9 @@ class FetcherIterator:
         finally:
             self._done.set()
             self._pool.shutdown(wait=False)
+            lm = getattr(mgr, "lifetime_metrics", None)
+            if lm is not None:   # test fakes may omit the rollup
+                lm.merge(self.metrics)
 
 
 class ShuffleReader:

@@ -73,3 +73,25 @@ class TaskMetrics:
     records_written: int = 0
     write_ns: int = 0
     extra: Dict[str, float] = field(default_factory=dict)
+
+    def merge(self, other: "TaskMetrics") -> None:
+        """Accumulate another task's counters (the executor-lifetime
+        rollup the reference feeds into Spark's task metrics)."""
+        self.remote_blocks_fetched += other.remote_blocks_fetched
+        self.local_blocks_fetched += other.local_blocks_fetched
+        self.remote_bytes_read += other.remote_bytes_read
+        self.local_bytes_read += other.local_bytes_read
+        self.fetch_wait_ns += other.fetch_wait_ns
+        self.records_read += other.records_read
+        self.bytes_written += other.bytes_written
+        self.records_written += other.records_written
+        self.write_ns += other.write_ns
+
+    def format(self) -> str:
+        return (f"read: {self.remote_bytes_read >> 20} MiB remote / "
+                f"{self.local_bytes_read >> 20} MiB local "
+                f"({self.remote_blocks_fetched}+{self.local_blocks_fetched} "
+                f"blocks, fetch-wait {self.fetch_wait_ns / 1e9:.2f}s) | "
+                f"write: {self.bytes_written >> 20} MiB, "
+                f"{self.records_written} records, "
+                f"{self.write_ns / 1e9:.2f}s")

@@ -167,6 +167,9 @@ class ShuffleWriter:
                 segments = self._partition_fixed(partitioner)
             self._commit_segments(segments)
         self.metrics.write_ns += time.perf_counter_ns() - t0
+        lm = getattr(self.manager, "lifetime_metrics", None)
+        if lm is not None:
+            lm.merge(self.metrics)
 
     def _partition_fixed(self, partitioner) -> List[bytes]:
         R = self.handle.num_partitions
