@@ -64,3 +64,35 @@ def test_connect_retry_exhaustion(tmp_path):
                        rdma_cm_event_timeout_ms=500)
     with pytest.raises(ConnectionError):
         ShuffleManager(conf, executor_id=0, driver_port=1)  # nothing listens
+
+
+def test_read_after_unregister_fails_cleanly(tmp_path):
+    """A reader constructed after unregister_shuffle must surface a fetch
+    failure (the driver table is gone on both the mmap and RPC lanes),
+    never hang or return stale data — the reference's liveness contract
+    (blocks live only until unregisterShuffle,
+    RdmaShuffleManager.scala:293-299)."""
+    import numpy as np
+    from sparkrdma_amd.conf import ShuffleConf
+    from sparkrdma_amd.driver import Driver
+    from sparkrdma_amd.manager import ShuffleManager
+    from sparkrdma_amd.partitioner import HashPartitioner
+    from sparkrdma_amd.reader import FetchFailedError
+
+    conf = ShuffleConf(shm_dir=str(tmp_path),
+                       partition_location_fetch_timeout_ms=500)
+    driver = Driver(conf)
+    mgr = ShuffleManager(conf, executor_id=0, driver_port=driver.port)
+    try:
+        part = HashPartitioner(4)
+        handle = mgr.register_shuffle(num_maps=1, num_partitions=4)
+        w = mgr.get_writer(handle, 0)
+        w.write_batch(np.arange(100, dtype=np.uint64))
+        w.stop(True, partitioner=part)
+        mgr.unregister_shuffle(handle.shuffle_id)
+        reader = mgr.get_reader(handle, 0, 3)
+        with pytest.raises(FetchFailedError):
+            list(reader)
+    finally:
+        mgr.stop()
+        driver.stop()
