@@ -152,7 +152,10 @@ def test_wide_writer_nonpow2_partitions(tmp_path):
         arr = _mk_records(rng, n, W)
         recs = torch.from_numpy(arr.reshape(-1)).cuda()
         w = mgr.get_writer(handle, 0)
-        w.write_device_records(recs, W, key_bytes=10)
+        # exercise MULTI-batch accumulation (torch.cat path)
+        half = (n // 2) * W
+        w.write_device_records(recs[:half], W, key_bytes=10)
+        w.write_device_records(recs[half:], W, key_bytes=10)
         w.stop(True, partitioner=part)
         prefix = arr[:, :8].copy().view("<u8").ravel()
         want_pids = part.partition_ids(prefix)
@@ -193,7 +196,8 @@ def _wide_ipc_worker(rank, world, driver_port, shm_dir, q):
                       partitions_per_executor=64, device="cuda",
                       validate=True, record_bytes=100)
         res = ts.run_step()
-        q.put((rank, res.records, res.remote_bytes))
+        res2 = ts.run_step()   # cross-step stability (pool/cache reuse)
+        q.put((rank, res.records, res.remote_bytes + res2.remote_bytes))
         eng.barrier()
         eng.shutdown()
     except BaseException as e:
