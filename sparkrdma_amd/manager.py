@@ -80,6 +80,8 @@ class ShuffleManager:
         self._meta_segment: Optional[HostSegment] = None
         self._meta_bump = META_TABLE_REGION_OFF
         self._meta_lock = threading.Lock()
+        self._meta_free: Dict[int, List[int]] = {}   # size -> free addrs
+        self._shuffle_tables: Dict[int, List[tuple]] = {}  # sid -> (addr, size)
         self._pool: Optional[BlockPool] = None
         self._data_segments: Dict[int, HostSegment] = {}
         self._next_segment_id = FIRST_DATA_SEGMENT_ID
@@ -299,6 +301,9 @@ class ShuffleManager:
         for blocks in outputs.values():
             for b in blocks:
                 b.release()
+        with self._meta_lock:
+            for addr, size in self._shuffle_tables.pop(shuffle_id, []):
+                self._meta_free.setdefault(size, []).append(addr)
         mm = self._driver_tables.pop(shuffle_id, None)
         if mm is not None:
             mm.close()
@@ -351,15 +356,27 @@ class ShuffleManager:
     # ------------------------------------------------------------------
     # map-output plumbing (used by writer/reader)
 
-    def alloc_table(self, num_partitions: int) -> tuple:
+    def alloc_table(self, num_partitions: int,
+                    shuffle_id: Optional[int] = None) -> tuple:
         """Allocate a MapTaskOutput table in the metadata segment; returns
-        (MapTaskOutput, table_addr)."""
+        (MapTaskOutput, table_addr). Regions recycle on
+        unregister_shuffle (the reference returns its table buffers to
+        the registered pool the same way, RdmaShuffleManager.scala:296) —
+        a long-running executor never exhausts the segment."""
         nbytes = num_partitions * 16
+        size = (nbytes + 63) & ~63
         with self._meta_lock:
-            addr = self._meta_bump
-            self._meta_bump += (nbytes + 63) & ~63
-            if self._meta_bump > META_SEGMENT_SIZE:
-                raise MemoryError("metadata segment exhausted")
+            free = self._meta_free.get(size)
+            if free:
+                addr = free.pop()
+            else:
+                addr = self._meta_bump
+                self._meta_bump += size
+                if self._meta_bump > META_SEGMENT_SIZE:
+                    raise MemoryError("metadata segment exhausted")
+            if shuffle_id is not None:
+                self._shuffle_tables.setdefault(shuffle_id, []).append(
+                    (addr, size))
         table = MapTaskOutput(num_partitions,
                               self._meta_segment.view(addr, nbytes))
         return table, addr

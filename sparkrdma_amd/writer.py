@@ -274,7 +274,7 @@ class ShuffleWriter:
         seg_bytes = counts * rec_w
 
         # greedy chunking of partitions into HBM blocks (same policy as host)
-        table, table_addr = mgr.alloc_table(R)
+        table, table_addr = mgr.alloc_table(R, self.handle.shuffle_id)
         write_block = mgr.conf.shuffle_write_block_size
         pool = mgr.gpu.pool
         blocks = []
@@ -499,7 +499,7 @@ class ShuffleWriter:
 
         # greedy chunking of partitions into HBM blocks (same policy as
         # the 16-byte path / RdmaMappedFile.java:113-157)
-        table, table_addr = mgr.alloc_table(R)
+        table, table_addr = mgr.alloc_table(R, self.handle.shuffle_id)
         write_block = mgr.conf.shuffle_write_block_size
         pool = mgr.gpu.pool
         blocks = []
@@ -595,7 +595,7 @@ class ShuffleWriter:
         """Zero-record map task: publish an all-empty location table."""
         mgr = self.manager
         R = self.handle.num_partitions
-        table, table_addr = mgr.alloc_table(R)
+        table, table_addr = mgr.alloc_table(R, self.handle.shuffle_id)
         meta_key = make_key(mgr.executor_id, 1)
         for p in range(R):
             table.put(p, 0, 0, meta_key)
@@ -606,7 +606,8 @@ class ShuffleWriter:
         """Greedy chunking at partition boundaries + table fill + publish."""
         mgr = self.manager
         write_block = mgr.conf.shuffle_write_block_size
-        table, table_addr = mgr.alloc_table(self.handle.num_partitions)
+        table, table_addr = mgr.alloc_table(self.handle.num_partitions,
+                                    self.handle.shuffle_id)
         blocks = []
         # greedy grouping of partitions into blocks
         group: List[int] = []

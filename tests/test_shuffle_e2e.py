@@ -202,3 +202,27 @@ def test_concurrent_shuffles(cluster):
     total = sum(len(unpack_partition_segment(c, 0)[0]) for _, c in reader)
     assert total == 9000
     managers[0].unregister_shuffle(h2.shuffle_id)
+
+
+def test_metadata_tables_recycle(cluster):
+    """Table regions return to a free list at unregister — a long-running
+    executor's metadata segment does not grow per shuffle (the reference
+    returns table buffers to its pool, RdmaShuffleManager.scala:296)."""
+    _, managers = cluster
+    m = managers[0]
+    part = HashPartitioner(4)
+
+    def one_round():
+        h = m.register_shuffle(num_maps=1, num_partitions=4)
+        w = m.get_writer(h, 0)
+        w.write_batch(np.arange(100, dtype=np.uint64))
+        w.stop(True, partitioner=part)
+        assert sum(1 for _ in m.get_reader(h, 0, 3)) >= 1
+        m.unregister_shuffle(h.shuffle_id)
+
+    one_round()
+    bump_after_first = m._meta_bump
+    for _ in range(10):
+        one_round()
+    assert m._meta_bump == bump_after_first, \
+        "metadata tables were not recycled"
