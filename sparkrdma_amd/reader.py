@@ -480,7 +480,7 @@ class FetcherIterator:
                         "fetch failed" + (
                             f" ({len(res.fetch.blocks)} blocks at "
                             f"key={res.fetch.key:#x})" if res.fetch else "")
-                        ) from res.error
+                        + f": {res.error!r}") from res.error
                 if res.fetch is None:      # sentinel: recheck termination
                     continue
                 with self._lock:
