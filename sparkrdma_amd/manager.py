@@ -85,6 +85,7 @@ class ShuffleManager:
         self._pool: Optional[BlockPool] = None
         self._data_segments: Dict[int, HostSegment] = {}
         self._next_segment_id = FIRST_DATA_SEGMENT_ID
+        self._free_segment_ids: List[int] = []   # recycled host slab ids
         self._registry: Optional[SegmentRegistry] = None
         self._driver_tables: Dict[int, mmap.mmap] = {}   # shuffle_id -> rw mmap
         self._remote_tables: set = set()   # shuffle ids served via RPC lane
@@ -263,8 +264,15 @@ class ShuffleManager:
         return min(1 << 30, self.conf.max_buffer_allocation_size)
 
     def _alloc_host_slab(self, size: int) -> int:
-        seg_id = self._next_segment_id
-        self._next_segment_id += 1
+        # ids recycle (15-bit key space): readers revalidate the path's
+        # inode per pread, so a reused id never serves the old file
+        if self._free_segment_ids:
+            seg_id = self._free_segment_ids.pop()
+        else:
+            seg_id = self._next_segment_id
+            if seg_id > 0x7FFF:
+                raise MemoryError("host segment id space exhausted")
+            self._next_segment_id += 1
         path = segment_path(self.conf.shm_dir, self.app_id, self.executor_id, seg_id)
         self._data_segments[seg_id] = HostSegment(path, size)
         return seg_id
@@ -273,6 +281,7 @@ class ShuffleManager:
         seg = self._data_segments.pop(seg_id)
         seg.close()
         seg.unlink()
+        self._free_segment_ids.append(seg_id)
 
     # ------------------------------------------------------------------
     # ShuffleManager public surface

@@ -88,20 +88,55 @@ class HostSegment:
 
 class HostSegmentReader:
     """Fetcher-side one-sided reader of a peer's segment (pread, no mmap —
-    avoids faulting the whole file into this process)."""
+    avoids faulting the whole file into this process).
+
+    Segment IDS RECYCLE on long-running executors (the owner unlinks the
+    old file and creates a new one under the same path), so ``revalidate``
+    re-opens when the path's inode no longer matches the cached fd —
+    the host-plane analog of the GPU slab generation check."""
 
     def __init__(self, path: str):
         self.path = path
         self.fd = os.open(path, os.O_RDONLY)
+        self._ino = os.fstat(self.fd).st_ino
+        self._swap_lock = threading.Lock()
+        self._old_fds: list = []
+
+    def revalidate(self) -> None:
+        try:
+            ino = os.stat(self.path).st_ino
+        except FileNotFoundError:
+            return   # owner gone: keep the fd; reads surface short/stale
+        if ino == self._ino:
+            return
+        with self._swap_lock:
+            if ino == self._ino:
+                return
+            new_fd = os.open(self.path, os.O_RDONLY)
+            # DEFER closing the old fd: a concurrent pread may hold it;
+            # stale fds only ever map to segments with no live blocks
+            # (liveness discipline), so the laggard read is of a dead
+            # range, never of current data
+            self._old_fds.append(self.fd)
+            self.fd = new_fd
+            self._ino = os.fstat(new_fd).st_ino
 
     def read(self, addr: int, length: int) -> bytes:
+        self.revalidate()
         return os.pread(self.fd, length, addr)
 
     def read_into(self, buf, addr: int) -> int:
+        self.revalidate()
         return os.preadv(self.fd, [buf], addr)
 
     def close(self) -> None:
         os.close(self.fd)
+        for fd in self._old_fds:
+            try:
+                os.close(fd)
+            except OSError:
+                pass
+        self._old_fds.clear()
 
 
 class SegmentRegistry:

@@ -14,6 +14,7 @@ from sparkrdma_amd.driver import Driver
 from sparkrdma_amd.manager import ShuffleManager
 from sparkrdma_amd.partitioner import HashPartitioner
 from sparkrdma_amd.reader import FetchFailedError
+from sparkrdma_amd.segments import FIRST_DATA_SEGMENT_ID
 from sparkrdma_amd.writer import unpack_partition_segment
 
 
@@ -226,3 +227,41 @@ def test_metadata_tables_recycle(cluster):
         one_round()
     assert m._meta_bump == bump_after_first, \
         "metadata tables were not recycled"
+
+
+def test_host_segment_ids_recycle_with_revalidation(tmp_path):
+    """Host slab ids RECYCLE (15-bit key space); a fetcher's cached fd
+    revalidates the path's inode per read, so a reused id never serves
+    the unlinked old file — the host-plane analog of the GPU slab
+    generation check."""
+    conf = ShuffleConf(shm_dir=str(tmp_path),
+                       max_buffer_allocation_size=1 << 26)  # trims eagerly
+    driver = Driver(conf)
+    m0 = ShuffleManager(conf, executor_id=0, driver_port=driver.port)
+    m1 = ShuffleManager(conf, executor_id=1, driver_port=driver.port)
+    try:
+        part = HashPartitioner(4)
+        for round_ in range(8):
+            h = m0.register_shuffle(num_maps=1, num_partitions=4)
+            keys = np.arange(round_ * 1000, round_ * 1000 + 500,
+                             dtype=np.uint64)
+            w = m0.get_writer(h, 0)
+            w.write_batch(keys, keys.view(np.uint8).reshape(-1, 8).copy())
+            w.stop(True, partitioner=part)
+            reader = m1.get_reader(h, 0, 3)
+            got = []
+            for _ref, data in reader:
+                k, v = unpack_partition_segment(data, 8)
+                assert np.array_equal(np.asarray(k).view(np.uint8),
+                                      np.asarray(v).reshape(-1))
+                got.append(np.array(k))
+            assert np.array_equal(np.sort(np.concatenate(got)), keys), \
+                f"round {round_}: stale host segment data"
+            m0.unregister_shuffle(h.shuffle_id)
+        # ids stayed bounded (recycled), not monotonically growing
+        assert m0._next_segment_id <= FIRST_DATA_SEGMENT_ID + 4, \
+            m0._next_segment_id
+    finally:
+        m0.stop()
+        m1.stop()
+        driver.stop()
