@@ -16,13 +16,17 @@ Layering:
                            analog: Spark itself played this role)
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from .conf import ShuffleConf
+from .engine import Engine
 from .manager import ShuffleHandle, ShuffleManager
 from .partitioner import HashPartitioner, RangePartitioner
+from .reader import FetchFailedError, ShuffleReader
+from .writer import ShuffleWriter
 
 __all__ = [
-    "ShuffleConf", "ShuffleManager", "ShuffleHandle",
+    "ShuffleConf", "ShuffleManager", "ShuffleHandle", "Engine",
+    "ShuffleReader", "ShuffleWriter", "FetchFailedError",
     "HashPartitioner", "RangePartitioner",
 ]
