@@ -471,3 +471,38 @@ def test_fuzz_shuffle_configs(seed, tmp_path):
     finally:
         mgr.stop()
         driver.stop()
+
+
+def test_wide_four_process_mesh(tmp_path):
+    """World-4 wide-record mesh on one GPU: 4 executor processes, every
+    rank fetches from 3 peers — the closest single-box approximation of
+    the 8-GPU all-to-all the driver runs at round end."""
+    import multiprocessing as mp
+    import socket
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    procs = [ctx.Process(target=_wide_ipc_worker,
+                         args=(r, 4, port, str(tmp_path), q))
+             for r in range(4)]
+    try:
+        for p in procs:
+            p.start()
+        remote_total = 0
+        for _ in range(4):
+            rank, records, remote = q.get(timeout=300)
+            assert not isinstance(records, str), f"rank {rank}: {records}"
+            assert records == 300_000
+            remote_total += remote
+        for p in procs:
+            p.join(timeout=120)
+            assert p.exitcode == 0
+        # at world 4, ~3/4 of each rank's input is remote (x2 steps)
+        assert remote_total > 4 * 300_000 * 100, remote_total
+    finally:
+        for p in procs:
+            if p.is_alive():
+                p.terminate()
