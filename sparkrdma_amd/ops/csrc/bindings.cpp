@@ -69,7 +69,7 @@ PYBIND11_MODULE(_hipshuffle, m) {
         py::arg("rec_bytes"), py::arg("key_bytes"), py::arg("pairs"),
         py::arg("stream") = 0, py::arg("pid_func") = -1,
         py::arg("pid_shift") = 0, py::arg("pid_mask") = 0,
-        py::arg("pid_nparts") = 0);
+        py::arg("pid_nparts") = 0, py::arg("idx_base") = 0);
   m.def("gather_records", &hs::gather_records, py::arg("recs"),
         py::arg("pairs"), py::arg("n"), py::arg("rec_bytes"),
         py::arg("dst_mode"), py::arg("out_base"), py::arg("dst_addr") = 0,

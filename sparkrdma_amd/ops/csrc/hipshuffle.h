@@ -43,7 +43,8 @@ void radix_scatter(uintptr_t keys, uintptr_t vals, uint32_t n, int shift,
 void extract_pairs(uintptr_t recs, uint64_t n, uint32_t rec_bytes,
                    uint32_t key_bytes, uintptr_t pairs, uintptr_t stream,
                    int pid_func = -1, int pid_shift = 0,
-                   uint32_t pid_mask = 0, uint32_t pid_nparts = 0);
+                   uint32_t pid_mask = 0, uint32_t pid_nparts = 0,
+                   uint64_t idx_base = 0);
 void gather_records(uintptr_t recs, uintptr_t pairs, uint64_t n,
                     uint32_t rec_bytes, int dst_mode, uint64_t out_base,
                     uintptr_t dst_addr, uintptr_t dstart, int shift,

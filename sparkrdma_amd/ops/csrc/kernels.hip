@@ -892,7 +892,9 @@ __global__ void onesweep_digit_bases_kernel(
 __global__ __launch_bounds__(BLOCK) void extract_pairs_kernel(
     const uint8_t* __restrict__ recs, uint64_t n, uint32_t rec_bytes,
     uint32_t key_bytes, uint64_t* __restrict__ pairs, int pid_func,
-    int pid_shift, uint32_t pid_mask, uint32_t pid_nparts) {
+    int pid_shift, uint32_t pid_mask, uint32_t pid_nparts,
+    uint64_t idx_base /* aux indices are idx_base + e: per-chunk extracts
+                         reference records in a larger arena */) {
   const uint64_t stride = (uint64_t)gridDim.x * BLOCK;
   for (uint64_t e = (uint64_t)blockIdx.x * BLOCK + threadIdx.x; e < n;
        e += stride) {
@@ -901,7 +903,7 @@ __global__ __launch_bounds__(BLOCK) void extract_pairs_kernel(
     uint32_t lo = *reinterpret_cast<const uint32_t*>(p);
     uint32_t hi = *reinterpret_cast<const uint32_t*>(p + 4);
     uint64_t prefix = ((uint64_t)hi << 32) | lo;
-    uint64_t aux = e;
+    uint64_t aux = idx_base + e;
     if (key_bytes > 8) {
       uint16_t klo = (uint16_t)(*reinterpret_cast<const uint32_t*>(p + 8));
       aux |= (uint64_t)klo << 48;
@@ -955,7 +957,7 @@ __global__ __launch_bounds__(BLOCK) void gather_records_kernel(
 void extract_pairs(uintptr_t recs, uint64_t n, uint32_t rec_bytes,
                    uint32_t key_bytes, uintptr_t pairs, uintptr_t stream,
                    int pid_func, int pid_shift, uint32_t pid_mask,
-                   uint32_t pid_nparts) {
+                   uint32_t pid_nparts, uint64_t idx_base) {
   if (rec_bytes % 4 || rec_bytes < 8)
     throw std::runtime_error("rec_bytes must be a multiple of 4, >= 8");
   if (key_bytes != 8 && key_bytes != 10)
@@ -964,11 +966,13 @@ void extract_pairs(uintptr_t recs, uint64_t n, uint32_t rec_bytes,
     throw std::runtime_error("record count exceeds 2^48");
   auto s = reinterpret_cast<hipStream_t>(stream);
   uint32_t grid = (uint32_t)min((n + BLOCK - 1) / BLOCK, (uint64_t)4096);
+  if ((idx_base + n) >> 48)
+    throw std::runtime_error("idx_base + n exceeds 2^48");
   hipLaunchKernelGGL(extract_pairs_kernel, dim3(grid ? grid : 1),
                      dim3(BLOCK), 0, s,
                      reinterpret_cast<const uint8_t*>(recs), n, rec_bytes,
                      key_bytes, reinterpret_cast<uint64_t*>(pairs),
-                     pid_func, pid_shift, pid_mask, pid_nparts);
+                     pid_func, pid_shift, pid_mask, pid_nparts, idx_base);
   HIP_CHECK(hipGetLastError());
 }
 

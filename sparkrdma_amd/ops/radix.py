@@ -167,7 +167,8 @@ def sort_records(recs: torch.Tensor, rec_bytes: int, key_bytes: int = 8,
                  pairs: Optional[torch.Tensor] = None,
                  tmp: Optional[torch.Tensor] = None,
                  ws: Optional[torch.Tensor] = None,
-                 fuse: bool = True) -> torch.Tensor:
+                 fuse: bool = True,
+                 pairs_filled: bool = False) -> torch.Tensor:
     """Sort W-byte AoS records by their 80-bit key: u64 LE prefix at
     offset 0 (bits [0, end_bit) significant — callers whose partitions
     share top prefix bits pass end_bit = 64 - shared) then, for
@@ -183,7 +184,13 @@ def sort_records(recs: torch.Tensor, rec_bytes: int, key_bytes: int = 8,
     if n == 0:
         return recs
     dev = recs.device
-    prs = extract_pairs(recs, rec_bytes, key_bytes, pairs)
+    if pairs_filled:
+        # caller already extracted (e.g. incrementally per fetched chunk,
+        # overlapping the fetch phase)
+        assert pairs is not None and pairs.numel() >= 2 * n
+        prs = pairs[:2 * n]
+    else:
+        prs = extract_pairs(recs, rec_bytes, key_bytes, pairs)
     if tmp is None:
         tmp = torch.empty_like(prs)
     else:

@@ -186,7 +186,15 @@ class TeraSort:
         chunk_shared_bits = (self.R // per - 1).bit_length()
         for (chunk_lo, _chunk_hi), reader in zip(spans, readers):
             tf = time.perf_counter()
-            parts = reader.collect_partitions()
+            if self.wide:
+                # drive the iterator OURSELVES: each fetched chunk's pair
+                # extraction launches as the chunk lands, overlapping the
+                # remaining fetches (collect_partitions is not needed —
+                # the arena holds everything)
+                pre_extracted = self._extract_while_fetching(reader)
+                parts = None
+            else:
+                parts = reader.collect_partitions()
             ts_ = time.perf_counter()
             arena = getattr(reader.fetcher, "arena", None)
             if self.wide and arena is not None:
@@ -199,7 +207,8 @@ class TeraSort:
                     out=self._rec_out(arena.numel()),
                     pairs=self._sort_tmp(2 * nrec, "_pairs_cache"),
                     tmp=self._sort_tmp(2 * nrec, "_tmp_cache"),
-                    ws=self._sort_ws())
+                    ws=self._sort_ws(),
+                    pairs_filled=pre_extracted)
                 torch.cuda.synchronize()
                 outs.append(out)
             elif self.device == "cuda" and arena is not None:
@@ -243,6 +252,62 @@ class TeraSort:
                                              device="cuda")
                                  for _ in range(H)]
         return self._arena_cache
+
+    def _extract_while_fetching(self, reader) -> bool:
+        """Iterate the fetcher, launching the (prefix, aux) pair
+        extraction for each CONTIGUOUS landed run — the extract overlaps
+        the in-flight fetches instead of re-reading the full arena
+        afterwards. Returns True when every byte was covered (else the
+        caller extracts in one pass — growth/non-tensor fallback)."""
+        import torch
+        from ..ops import load
+        hs = load()
+        W = self.RECORD_BYTES
+        f = reader.fetcher
+        buf = f._arena_buf
+        if buf is None:
+            for _ in f:
+                pass
+            return False
+        base = buf.data_ptr()
+        cap = buf.numel()
+        pairs_buf = self._sort_tmp(2 * (cap // W) + 16, "_pairs_cache")
+        stream = torch.cuda.current_stream().cuda_stream
+        ok = True
+        covered = 0
+        run_off = run_len = 0
+
+        def flush():
+            nonlocal ok, covered
+            if not run_len:
+                return
+            if run_off % W or run_len % W:
+                ok = False
+                return
+            hs.extract_pairs(base + run_off, run_len // W, W, 10,
+                             pairs_buf.data_ptr() + (run_off // W) * 16,
+                             stream, idx_base=run_off // W)
+            covered += run_len
+
+        for _ref, view in f:
+            if (not isinstance(view, torch.Tensor)
+                    or f._arena_buf is not buf):
+                ok = False       # grew / non-arena chunk: full re-extract
+                continue
+            boff = view.data_ptr() - base
+            nb = view.numel()
+            if not (0 <= boff and boff + nb <= cap):
+                ok = False
+                continue
+            if boff == run_off + run_len:
+                run_len += nb
+            else:
+                flush()
+                run_off, run_len = boff, nb
+        flush()
+        arena = f.arena
+        return bool(ok and arena is not None
+                    and covered == arena.numel())
 
     def _sort_tmp(self, n_i64: int, attr: str = "_tmp_cache"):
         import torch
