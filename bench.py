@@ -100,6 +100,10 @@ def main():
         conf.shuffle_read_block_size = 512 << 20
         conf.max_bytes_in_flight = 8 << 30
     eng = Engine(conf, rank=rank, world_size=world)
+    if rank == 0:
+        print(f"[bench] engine up: world={world} transport={conf.transport} "
+              f"workload={args.workload} rec_bytes={rec_bytes}",
+              file=sys.stderr, flush=True)
 
     device = "cuda" if use_cuda else "cpu"
     if args.workload == "terasort":
@@ -145,6 +149,9 @@ def main():
                   f"fetch={r.fetch_s*1e3:.1f} sort={r.sort_s*1e3:.1f}",
                   file=sys.stderr)
     barrier_sync()
+    if rank == 0:
+        print(f"[bench] warmup done ({args.warmup}); timing {args.steps} "
+              "steps", file=sys.stderr, flush=True)
     t0 = time.perf_counter()
     results = [ts.run_step() for _ in range(args.steps)]
     barrier_sync()
