@@ -107,8 +107,12 @@ def main():
 
     device = "cuda" if use_cuda else "cpu"
     if args.workload == "terasort":
-        mode = args.mode if (use_cuda or args.mode != "rccl") \
-            else "framework"
+        mode = args.mode
+        if mode == "framework" and use_cuda and (
+                conf.use_rccl or conf.transport == "rccl"):
+            mode = "rccl"   # spark.shuffle.rdma.useRccl / transport=rccl
+        if mode == "rccl" and not use_cuda:
+            mode = "framework"
         ts = TeraSort(eng, n_rec, partitions_per_executor=ppe,
                       device=device, mode=mode, validate=args.validate,
                       record_bytes=rec_bytes)
