@@ -24,7 +24,9 @@
 //     (u64 flag|count agent-scope descriptors), AoS records, deferred
 //     walk; selectable hist+scan mode kept for A/B.
 //
-// Limits: n < 2^32 elements per call; buckets = pow2, 2^4..2^12.
+// Limits: n < 2^32 elements per call; ONE pass covers 2^4..2^12 LDS
+// digits — arbitrary partition counts come from the partition funcs
+// (PartFunc 2/3) and, above 4096, the writer's two-level pid radix.
 
 #include "common.h"
 
