@@ -1,7 +1,9 @@
 """TeraSort — the reference's headline workload (README.md:7-17).
 
-Synthetic records: u64 key (uniform random) + u64 payload. One *step* is a
-complete sort job over the fixed per-executor dataset:
+Synthetic records: canonical 100-byte records (10 B key = u64 LE prefix +
+u16 LE low, 90 B payload) on the wide path, or (u64 key, u64 payload)
+16-byte pairs. One *step* is a complete sort job over the fixed
+per-executor dataset:
 
   map:    radix-partition the local shard by the top log2(R) key bits into
           R global partitions, serialized straight into HBM blocks
