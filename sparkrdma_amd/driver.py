@@ -169,7 +169,7 @@ class Driver:
             sid, map_id, addr, key = rpc.unpack_table_write(body)
             with self._lock:
                 meta = self._shuffles.get(sid)
-            if meta is None:
+            if meta is None or not 0 <= map_id < meta.num_maps:
                 conn.send(rpc.MSG_ERROR, rpc.pack_unregister(sid))
             else:
                 import struct as _struct

@@ -121,3 +121,14 @@ def test_write_records_partition_ids_interface(pair):
     parts = reader.collect_partitions()
     sizes = {p: sum(len(c) for c in chunks) for p, chunks in parts.items()}
     assert all(sizes[p] > 0 for p in range(4))  # one record per quarter
+
+
+def test_table_write_bounds_checked(pair):
+    """A publish with an out-of-range map_id is rejected cleanly (ERROR
+    reply), never corrupting adjacent metadata."""
+    driver, mgr = pair
+    h = mgr.register_shuffle(num_maps=2, num_partitions=4)
+    from sparkrdma_amd import rpc
+    mtype, _ = mgr._rpc_call(
+        rpc.MSG_TABLE_WRITE, rpc.pack_table_write(h.shuffle_id, 99, 0, 1))
+    assert mtype == rpc.MSG_ERROR
