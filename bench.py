@@ -118,9 +118,11 @@ def main():
                       record_bytes=rec_bytes)
     elif args.workload == "pagerank":
         from sparkrdma_amd.workloads.pagerank import PageRank
-        # 19 GB edge list analog: 16 B records
+        # 19 GB edge list analog: 16 B records; vertex count scales with
+        # the dataset (pow2) so tiny CPU contract runs stay tiny
         edges = n_rec
-        ts = PageRank(eng, num_vertices=1 << 26, edges_per_executor=edges,
+        vbits = min(26, max(16, edges.bit_length() - 2))
+        ts = PageRank(eng, num_vertices=1 << vbits, edges_per_executor=edges,
                       partitions_per_executor=ppe, device=device,
                       iterations=3)
     elif args.workload == "join":
