@@ -154,3 +154,43 @@ def test_fetcher_randomization_deterministic_by_seed():
         list(it)
         runs.append([r[1] for r in mgr.reads])
     assert runs[0] == runs[1]
+
+
+def test_range_partitioner_mulhi_equivalence():
+    """The GPU kernel's func-2 partition (pid = floor(key * R / 2^64),
+    multiply-high) must equal the CPU RangePartitioner's searchsorted
+    over CEILING bounds for every key INCLUDING exact boundaries — the
+    bit-for-bit contract the non-pow2 GPU path relies on."""
+    rng = np.random.default_rng(77)
+    for R in (2, 3, 7, 12, 100, 1000, 4096, 5000, 12000, 65521):
+        p = RangePartitioner.uniform(R)
+        keys = rng.integers(0, 2 ** 64, 5000, dtype=np.uint64)
+        # inject exact boundary keys and their neighbours
+        idx = rng.integers(0, R - 1, size=min(R - 1, 64))
+        b = p.bounds[idx]
+        keys = np.concatenate([keys, b, b - 1, b + 1,
+                               np.array([0, 2**64 - 1], dtype=np.uint64)])
+        want = p.partition_ids(keys)
+        # mulhi in python ints (the kernel's (u128 k * R) >> 64)
+        got = np.array([(int(k) * R) >> 64 for k in keys], dtype=np.int64)
+        assert np.array_equal(got, want.astype(np.int64)), R
+
+
+def test_hash_partitioner_mod_equivalence():
+    """func-3 (splitmix mix then % R) equals HashPartitioner.partition_ids
+    for non-pow2 R (the GPU hash-mod contract)."""
+    rng = np.random.default_rng(78)
+    MIX = 0xFF51AFD7ED558CCD
+    for R in (3, 7, 100, 1000, 5000):
+        p = HashPartitioner(R)
+        keys = rng.integers(0, 2 ** 64, 3000, dtype=np.uint64)
+        want = p.partition_ids(keys)
+        got = []
+        for k in keys:
+            k = int(k)
+            k ^= k >> 33
+            k = (k * MIX) & ((1 << 64) - 1)
+            k ^= k >> 33
+            got.append(k % R)
+        assert np.array_equal(np.array(got, dtype=np.int64),
+                              want.astype(np.int64)), R
