@@ -86,6 +86,8 @@ class TaskMetrics:
         self.bytes_written += other.bytes_written
         self.records_written += other.records_written
         self.write_ns += other.write_ns
+        for k, v in other.extra.items():
+            self.extra[k] = self.extra.get(k, 0.0) + v
 
     def format(self) -> str:
         return (f"read: {self.remote_bytes_read >> 20} MiB remote / "
