@@ -254,29 +254,26 @@ class MsgConnection:
             self.sock.sendall(b"".join(frames))
 
     def recv(self, timeout: Optional[float] = None):
-        """Receive one complete message, or None on EOF."""
+        """Receive one complete message, or None on EOF. Messages decoded
+        beyond the first in one TCP read are queued and returned in order
+        by later calls."""
+        pending = getattr(self, "_pending", None)
+        if pending:
+            return pending.pop(0)
         self.sock.settimeout(timeout)
         while True:
-            msgs = self._decoder.feed(b"")
-            if msgs:
-                return msgs[0]
             data = self.sock.recv(65536)
             if not data:
                 return None
             msgs = self._decoder.feed(data)
             if msgs:
-                # buffer any extras back is unnecessary: feed() retains state,
-                # but we may have decoded >1; stash them.
                 if len(msgs) > 1:
                     self._pending = getattr(self, "_pending", [])
                     self._pending.extend(msgs[1:])
                 return msgs[0]
 
-    def recv_any(self, timeout: Optional[float] = None):
-        pending = getattr(self, "_pending", None)
-        if pending:
-            return pending.pop(0)
-        return self.recv(timeout)
+    # kept as an alias for call-site clarity (drains queued messages first)
+    recv_any = recv
 
     def close(self) -> None:
         try:
