@@ -65,3 +65,27 @@ def test_empty_body():
 def test_max_frame_too_small():
     with pytest.raises(ValueError):
         rpc.encode_frames(0, b"x", max_frame=8)
+
+
+def test_decoder_fuzz_random_splits():
+    """A stream of many segmented messages fed to the decoder in random-
+    sized slices (1..7000 B) must reassemble every message, in order,
+    regardless of where TCP read boundaries fall."""
+    import random
+    rng = random.Random(1234)
+    msgs = []
+    stream = bytearray()
+    for i in range(60):
+        body = bytes(rng.getrandbits(8) for _ in range(rng.randrange(0, 5000)))
+        mtype = rng.randrange(0, 13)
+        msgs.append((mtype, body))
+        for f in rpc.encode_frames(mtype, body, rng.choice((64, 256, 4096))):
+            stream.extend(f)
+    dec = rpc.FrameDecoder()
+    got = []
+    off = 0
+    while off < len(stream):
+        n = rng.randrange(1, 7000)
+        got.extend(dec.feed(bytes(stream[off:off + n])))
+        off += n
+    assert got == msgs
