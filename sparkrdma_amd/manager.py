@@ -77,6 +77,7 @@ class ShuffleManager:
         self._rpc_serial = threading.Lock()   # one in-flight driver RPC
         self._replies: List[tuple] = []
         self._reply_evt = threading.Event()
+        self._driver_lost = threading.Event()
         self._meta_segment: Optional[HostSegment] = None
         self._meta_bump = META_TABLE_REGION_OFF
         self._meta_lock = threading.Lock()
@@ -160,11 +161,18 @@ class ShuffleManager:
                         self._reply_evt.set()
         except (OSError, ValueError):
             return
+        finally:
+            # driver connection gone (crash or our own stop): fail pending
+            # and future RPCs promptly instead of burning the full timeout
+            self._driver_lost.set()
+            self._reply_evt.set()
 
     def _rpc_call(self, mtype: int, body: bytes, timeout: float = 30.0) -> tuple:
         # serialize request/response pairs: the reply FIFO has no request
         # ids, so two concurrent callers could swap replies
         with self._rpc_serial:
+            if self._driver_lost.is_set():
+                raise ConnectionError("driver connection lost")
             self._conn.send(mtype, body)
             deadline = time.monotonic() + timeout
             while True:
@@ -176,6 +184,8 @@ class ShuffleManager:
                         if not self._replies:
                             self._reply_evt.clear()
                         return reply
+                    if self._driver_lost.is_set():
+                        raise ConnectionError("driver connection lost")
                     self._reply_evt.clear()
 
     def _init_segments(self) -> None:

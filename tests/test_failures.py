@@ -96,3 +96,31 @@ def test_read_after_unregister_fails_cleanly(tmp_path):
     finally:
         mgr.stop()
         driver.stop()
+
+
+def test_driver_death_fails_rpcs_fast(tmp_path):
+    """Driver crash mid-run: manager RPCs must raise ConnectionError
+    promptly (seconds, not the 30 s RPC timeout) once the connection
+    drops — the analog of the reference failing tasks when the CM
+    connection dies."""
+    conf = ShuffleConf(shm_dir=str(tmp_path))
+    driver = Driver(conf)
+    mgr = ShuffleManager(conf, executor_id=0, driver_port=driver.port)
+    try:
+        h = mgr.register_shuffle(num_maps=1, num_partitions=2)
+        driver.stop()
+        t0 = time.monotonic()
+        with pytest.raises((ConnectionError, TimeoutError)) as exc_info:
+            for _ in range(10):  # first call may race the EOF notice
+                mgr.lookup_shuffle(h.shuffle_id)
+                time.sleep(0.05)
+        assert time.monotonic() - t0 < 10.0, "must fail fast, not burn timeouts"
+        assert isinstance(exc_info.value, ConnectionError)
+        # and every later call fails instantly at entry
+        t0 = time.monotonic()
+        with pytest.raises(ConnectionError):
+            mgr.lookup_shuffle(h.shuffle_id)
+        assert time.monotonic() - t0 < 1.0
+    finally:
+        mgr.stop()
+        driver.stop()
