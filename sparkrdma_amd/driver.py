@@ -99,6 +99,7 @@ class Driver:
             # of SparkListenerBlockManagerRemoved pruning (reference
             # RdmaShuffleManager.scala:155-165)
             reannounce = None
+            release = []
             with self._lock:
                 if conn in self._conns:
                     self._conns.remove(conn)
@@ -112,6 +113,14 @@ class Driver:
                     reannounce = (rpc.pack_announce(
                         self.app_id, list(self._members.values())),
                         list(self._conns))
+                # the loss may have satisfied a pending barrier (everyone
+                # still alive is already waiting) — re-evaluate, else the
+                # survivors would hang forever
+                if len(self._barrier_waiters) >= len(self._members) > 0:
+                    gen = self._barrier_gen
+                    self._barrier_gen += 1
+                    release = [(c, gen) for c in self._barrier_waiters]
+                    self._barrier_waiters = []
             if reannounce is not None:
                 payload, conns = reannounce
                 for c in conns:
@@ -119,6 +128,11 @@ class Driver:
                         c.send(rpc.MSG_ANNOUNCE, payload)
                     except OSError:
                         pass
+            for c, gen in release:
+                try:
+                    c.send(rpc.MSG_BARRIER_OK, rpc.pack_unregister(gen))
+                except OSError:
+                    pass
 
     def _dispatch(self, conn: rpc.MsgConnection, mtype: int, body: bytes) -> None:
         if mtype == rpc.MSG_HELLO:

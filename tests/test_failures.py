@@ -124,3 +124,44 @@ def test_driver_death_fails_rpcs_fast(tmp_path):
     finally:
         mgr.stop()
         driver.stop()
+
+
+def test_barrier_released_on_member_loss(tmp_path):
+    """If an executor dies while the others wait at the stage barrier, the
+    driver must re-evaluate the barrier against the pruned membership and
+    release the survivors — not leave them hanging."""
+    conf = ShuffleConf(shm_dir=str(tmp_path))
+    driver = Driver(conf)
+    m0 = ShuffleManager(conf, executor_id=0, driver_port=driver.port)
+    m1 = ShuffleManager(conf, executor_id=1, driver_port=driver.port)
+    m2 = ShuffleManager(conf, executor_id=2, driver_port=driver.port)
+    import threading
+    try:
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline and len(m0._members) < 3:
+            time.sleep(0.01)
+        assert len(m0._members) == 3
+        done = []
+        errs = []
+
+        def wait_barrier(m):
+            try:
+                m.barrier(timeout=15)
+                done.append(m.executor_id)
+            except Exception as e:  # pragma: no cover
+                errs.append(e)
+
+        t0 = threading.Thread(target=wait_barrier, args=(m0,))
+        t1 = threading.Thread(target=wait_barrier, args=(m1,))
+        t0.start()
+        t1.start()
+        time.sleep(0.3)   # both waiting at the barrier
+        m2.stop()         # third member dies instead of arriving
+        t0.join(timeout=10)
+        t1.join(timeout=10)
+        assert not errs
+        assert sorted(done) == [0, 1], "survivors must be released"
+    finally:
+        m0.stop()
+        m1.stop()
+        driver.stop()
