@@ -44,13 +44,18 @@ class Block:
     refcount: int = 1
 
     def retain(self) -> "Block":
-        self.refcount += 1
+        with self.pool._lock:
+            self.refcount += 1
         return self
 
     def release(self) -> None:
-        self.refcount -= 1
-        if self.refcount == 0:
-            self.pool.put(self)
+        # refcount ops share the pool lock (the reference uses an atomic,
+        # RdmaRegisteredBuffer.java:45-62); put() re-takes it, so drop first
+        with self.pool._lock:
+            self.refcount -= 1
+            if self.refcount > 0:
+                return
+        self.pool.put(self)
 
 
 class _BuddySlab:
