@@ -114,3 +114,44 @@ def test_preallocate_warms_pool():
     assert pool.stats.slab_count == before
     for b in blocks:
         b.release()
+
+
+def test_buddy_fuzz_against_model():
+    """Random alloc/free sequence vs an interval model: no live block may
+    overlap another, every block stays inside its slab, and draining all
+    blocks returns every slab to fully-free (perfect coalescing)."""
+    import random
+    rng = random.Random(9)
+    SLAB = 1 << 22          # 4 MiB
+    next_id = [0]
+
+    def alloc_slab(size):
+        next_id[0] += 1
+        return next_id[0]
+
+    pool = BlockPool(SLAB, 8 * SLAB, alloc_slab)
+    live = []               # (block, (seg, lo, hi))
+    for step in range(3000):
+        if live and (rng.random() < 0.45 or len(live) > 400):
+            i = rng.randrange(len(live))
+            b, _ = live.pop(i)
+            b.release()
+        else:
+            size = rng.choice((1, 100, 16 << 10, 40 << 10, 64 << 10,
+                               1 << 20, (1 << 22) - 7))
+            try:
+                b = pool.get(size)
+            except MemoryError:
+                continue
+            assert b.capacity >= size
+            lo, hi = b.offset, b.offset + b.capacity
+            assert 0 <= lo and hi <= SLAB
+            for _ob, (seg, olo, ohi) in live:
+                if seg == b.segment_id:
+                    assert hi <= olo or lo >= ohi, "overlapping live blocks"
+            live.append((b, (b.segment_id, lo, hi)))
+    for b, _ in live:
+        b.release()
+    assert pool.stats.used_bytes == 0
+    for slab in pool._slabs.values():
+        assert slab.fully_free, "coalescing must restore full slabs"
