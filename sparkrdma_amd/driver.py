@@ -97,42 +97,50 @@ class Driver:
         finally:
             # executor loss: prune membership and re-announce, the analog
             # of SparkListenerBlockManagerRemoved pruning (reference
-            # RdmaShuffleManager.scala:155-165)
-            reannounce = None
-            release = []
-            with self._lock:
-                if conn in self._conns:
-                    self._conns.remove(conn)
-                if conn in self._barrier_waiters:
-                    self._barrier_waiters.remove(conn)
-                exec_id = self._conn_exec.pop(conn, None)
-                if exec_id is not None and not self._stopped.is_set():
-                    self._members.pop(exec_id, None)
+            # RdmaShuffleManager.scala:155-165). A clean MSG_BYE departure
+            # already pruned; this then finds nothing to do.
+            self._prune_executor(conn, clean=False)
+
+    def _prune_executor(self, conn: rpc.MsgConnection, clean: bool) -> None:
+        """Drop an executor (clean BYE or lost connection): prune
+        membership, re-announce, and release any barrier the departure
+        now satisfies (everyone still alive is already waiting)."""
+        reannounce = None
+        release = []
+        with self._lock:
+            if not clean and conn in self._conns:
+                self._conns.remove(conn)
+            if conn in self._barrier_waiters:
+                self._barrier_waiters.remove(conn)
+            exec_id = self._conn_exec.pop(conn, None)
+            if exec_id is not None and not self._stopped.is_set():
+                self._members.pop(exec_id, None)
+                if clean:
+                    log.info("executor %d detached; %d members remain",
+                             exec_id, len(self._members))
+                else:
                     log.warning("executor %d lost; %d members remain",
                                 exec_id, len(self._members))
-                    reannounce = (rpc.pack_announce(
-                        self.app_id, list(self._members.values())),
-                        list(self._conns))
-                # the loss may have satisfied a pending barrier (everyone
-                # still alive is already waiting) — re-evaluate, else the
-                # survivors would hang forever
-                if len(self._barrier_waiters) >= len(self._members) > 0:
-                    gen = self._barrier_gen
-                    self._barrier_gen += 1
-                    release = [(c, gen) for c in self._barrier_waiters]
-                    self._barrier_waiters = []
-            if reannounce is not None:
-                payload, conns = reannounce
-                for c in conns:
-                    try:
-                        c.send(rpc.MSG_ANNOUNCE, payload)
-                    except OSError:
-                        pass
-            for c, gen in release:
+                reannounce = (rpc.pack_announce(
+                    self.app_id, list(self._members.values())),
+                    [c for c in self._conns if c is not conn])
+            if len(self._barrier_waiters) >= len(self._members) > 0:
+                gen = self._barrier_gen
+                self._barrier_gen += 1
+                release = [(c, gen) for c in self._barrier_waiters]
+                self._barrier_waiters = []
+        if reannounce is not None:
+            payload, conns = reannounce
+            for c in conns:
                 try:
-                    c.send(rpc.MSG_BARRIER_OK, rpc.pack_unregister(gen))
+                    c.send(rpc.MSG_ANNOUNCE, payload)
                 except OSError:
                     pass
+        for c, gen in release:
+            try:
+                c.send(rpc.MSG_BARRIER_OK, rpc.pack_unregister(gen))
+            except OSError:
+                pass
 
     def _dispatch(self, conn: rpc.MsgConnection, mtype: int, body: bytes) -> None:
         if mtype == rpc.MSG_HELLO:
@@ -191,7 +199,7 @@ class Driver:
                                  _struct.pack("<QI", addr, key))
                 conn.send(rpc.MSG_HANDLE, rpc.pack_handle(sid, 0, 0, "", 0))
         elif mtype == rpc.MSG_BYE:
-            pass
+            self._prune_executor(conn, clean=True)
         elif mtype == rpc.MSG_BARRIER:
             with self._lock:
                 self._barrier_waiters.append(conn)
