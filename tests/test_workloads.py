@@ -67,3 +67,15 @@ def test_reduce_by_key_cpu(engine):
                     partitions_per_executor=16, device="cpu", validate=True)
     res = r.run_step()
     assert res.groups == 500
+
+
+def test_api_tour_example_runs():
+    """examples/api_tour.py is living documentation — keep it green."""
+    import subprocess
+    import sys
+    import os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run([sys.executable, os.path.join(repo, "examples", "api_tour.py")],
+                       capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "range-partitioned" in r.stdout
