@@ -314,6 +314,7 @@ class DataClient:
              length: int) -> bytes:
         ep = (host, port)
         sock = self._acquire(ep)
+        released = False
         try:
             sock.sendall(_REQ.pack(key, addr, length))
             hdr = DataServer._recv_exact(sock, _RSP.size)
@@ -321,7 +322,9 @@ class DataClient:
                 raise ConnectionError("data server closed connection")
             status, flags = _RSP.unpack(hdr)
             if status < 0:
+                # protocol-level rejection: the connection itself is fine
                 self._release(ep, sock, broken=False)
+                released = True
                 raise IOError(f"remote read failed (status {status})")
             out = bytearray(status)
             nat = _native()
@@ -334,6 +337,7 @@ class DataClient:
                     raise ConnectionError(
                         f"native chunk receive failed ({got}/{status})")
                 self._release(ep, sock, broken=False)
+                released = True
                 return bytes(out)
             mv = memoryview(out)
             off = 0
@@ -355,9 +359,14 @@ class DataClient:
                     mv[off:off + raw_len] = dec
                 off += raw_len
             self._release(ep, sock, broken=False)
+            released = True
             return bytes(out)
-        except (OSError, ConnectionError):
-            self._release(ep, sock, broken=True)
+        except BaseException:
+            # ANY failure after acquire (socket error, zlib error, ...)
+            # must return the pool slot exactly once — a leak here would
+            # eventually deadlock _acquire at MAX_CONNS_PER_PEER
+            if not released:
+                self._release(ep, sock, broken=True)
             raise
 
     @staticmethod

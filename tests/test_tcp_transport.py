@@ -67,6 +67,25 @@ def test_tcp_server_rejects_foreign_key(tcp_cluster):
                              make_key(0, 2), 0, 64)
 
 
+def test_tcp_client_pool_survives_rejections(tcp_cluster):
+    """A protocol-level rejection (status<0) must release the pooled
+    connection exactly once: repeated rejections past MAX_CONNS_PER_PEER
+    must neither deadlock _acquire nor corrupt the pool counts, and the
+    same client must still serve valid reads afterwards."""
+    m0, m1 = tcp_cluster
+    from sparkrdma_amd.data_server import DataClient
+    from sparkrdma_amd.map_output import make_key
+    client = m0._data_client
+    ep_port = m1._data_server.port
+    for _ in range(DataClient.MAX_CONNS_PER_PEER * 3):
+        with pytest.raises(IOError):
+            client.read("127.0.0.1", ep_port, make_key(0, 2), 0, 64)
+    ep = ("127.0.0.1", ep_port)
+    assert client._counts.get(ep, 0) == len(client._free.get(ep, [])), \
+        "every rejected read must return its socket to the free list"
+    assert 0 <= client._counts.get(ep, 0) <= DataClient.MAX_CONNS_PER_PEER
+
+
 @pytest.fixture
 def tcp_cluster_compressed(tmp_path):
     conf = ShuffleConf(shm_dir=str(tmp_path), transport="tcp",
