@@ -210,6 +210,10 @@ class CopyEngine {
 
   void release_event(uint64_t id, hipEvent_t ev) {
     std::lock_guard<std::mutex> g(ev_mu_);
+    // erase() guards concurrent poll()/wait() reaps of the same id: only
+    // the first releaser recycles the event; a loser that then queries a
+    // re-acquired event can at worst see NotReady once more, never a
+    // false completion (its id is gone from the map -> "already reaped").
     if (events_.erase(id)) free_events_.push_back(ev);
   }
 
