@@ -10,7 +10,6 @@ import time
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
-import numpy as np
 import torch
 
 from sparkrdma_amd.conf import ShuffleConf

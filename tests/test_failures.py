@@ -49,7 +49,6 @@ def test_fetch_failure_surfaces(tmp_path):
         # simulate executor-1 crash: its data segments vanish
         for seg in list(m1._data_segments.values()):
             seg.unlink()
-        import sparkrdma_amd.segments as seg_mod
         with pytest.raises((FetchFailedError, FileNotFoundError)):
             reader = m0.get_reader(handle, 0, 3)
             list(reader)

@@ -6,7 +6,6 @@ announce-time mesh pre-build (RdmaShuffleManager.scala:121-126)."""
 import threading
 import time
 
-import numpy as np
 import pytest
 
 from sparkrdma_amd.conf import ShuffleConf

@@ -8,7 +8,6 @@ import socket
 import sys
 import time
 
-import numpy as np
 import pytest
 
 pytestmark = pytest.mark.gpu

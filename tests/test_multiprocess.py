@@ -4,7 +4,6 @@ Engine wrapper the way torchrun does (RANK/WORLD_SIZE env)."""
 
 import multiprocessing as mp
 import os
-import pickle
 
 import numpy as np
 import pytest

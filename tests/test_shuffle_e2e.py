@@ -112,7 +112,6 @@ def test_bytes_record_shuffle_groupby(cluster):
 
 
 def _unpickle_one(buf, off):
-    import pickletools
     import pickle
     import io
     bio = io.BytesIO(buf[off:])
