@@ -187,3 +187,18 @@ def test_tcp_client_pool_parallel_reads(tcp_cluster):
     assert not errs
     assert m0._data_client._counts[("127.0.0.1", m1._data_server.port)] > 1
     blk.release()
+
+
+def test_tcp_server_rejects_oversize_read(tcp_cluster):
+    """length > MAX_READ gets a clean -1 status before any streaming
+    starts, and the connection stays usable for the next request."""
+    m0, m1 = tcp_cluster
+    from sparkrdma_amd import data_server as ds
+    from sparkrdma_amd.map_output import make_key
+    with pytest.raises(IOError):
+        m0._data_client.read("127.0.0.1", m1._data_server.port,
+                             make_key(1, 2), 0, ds.MAX_READ + 1)
+    # pool still consistent, later rejection also clean
+    with pytest.raises(IOError):
+        m0._data_client.read("127.0.0.1", m1._data_server.port,
+                             make_key(0, 2), 0, 64)
