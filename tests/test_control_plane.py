@@ -132,3 +132,23 @@ def test_table_write_bounds_checked(pair):
     mtype, _ = mgr._rpc_call(
         rpc.MSG_TABLE_WRITE, rpc.pack_table_write(h.shuffle_id, 99, 0, 1))
     assert mtype == rpc.MSG_ERROR
+
+
+def test_late_joining_executor_announced_to_all(pair, tmp_path):
+    """An executor that joins after the app is running must appear in
+    every member's view (the announce fan-out re-sends full membership),
+    and the newcomer sees the existing members too."""
+    import time
+    driver, mgr = pair
+    late = ShuffleManager(ShuffleConf(shm_dir=mgr.conf.shm_dir,
+                                      max_buffer_allocation_size=1 << 30),
+                          executor_id=7, driver_port=driver.port)
+    try:
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline and (
+                len(mgr._members) < 2 or len(late._members) < 2):
+            time.sleep(0.01)
+        assert set(mgr._members) == {0, 7}
+        assert set(late._members) == {0, 7}
+    finally:
+        late.stop()
