@@ -152,3 +152,20 @@ def test_late_joining_executor_announced_to_all(pair, tmp_path):
         assert set(late._members) == {0, 7}
     finally:
         late.stop()
+
+
+def test_double_stop_idempotent(tmp_path):
+    """stop() twice on every component is a no-op, not an error — teardown
+    runs from finally blocks and signal handlers, so it must be safe to
+    repeat (and Engine.__exit__ may race an explicit shutdown())."""
+    from sparkrdma_amd.engine import Engine
+    conf = ShuffleConf(shm_dir=str(tmp_path), max_buffer_allocation_size=1 << 30)
+    eng = Engine(conf, rank=0, world_size=1, driver_port=0)
+    h = eng.register_shuffle(1, 2)
+    w = eng.manager.get_writer(h, 0)
+    w.write_records([(1, 2)], None)
+    w.stop(True)
+    eng.shutdown()
+    eng.shutdown()           # second shutdown: no-op
+    eng.manager.stop()       # direct repeats too
+    eng.driver.stop()
