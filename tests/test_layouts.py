@@ -79,3 +79,30 @@ def test_driver_table():
     assert parsed[3] == (192, make_key(3, 1))
     with pytest.raises(IndexError):
         d.publish(4, 0, 1)
+
+
+def test_partition_segment_codec_roundtrip():
+    """pack/unpack_partition_segment: keys-only and keys+values, incl.
+    empty — the CPU fixed-width wire format readers rely on."""
+    import numpy as np
+    from sparkrdma_amd.writer import pack_partition_segment, \
+        unpack_partition_segment
+
+    rng = np.random.default_rng(3)
+    # keys only
+    keys = rng.integers(0, 2 ** 64, 257, dtype=np.uint64)
+    buf = pack_partition_segment(keys, None)
+    k, v = unpack_partition_segment(buf, 0)
+    assert np.array_equal(np.asarray(k), keys)
+    assert v is None or len(v) == 0
+    # keys + 12-byte values
+    vals = rng.integers(0, 256, (257, 12), dtype=np.uint8)
+    buf = pack_partition_segment(keys, vals)
+    k, v = unpack_partition_segment(buf, 12)
+    assert np.array_equal(np.asarray(k), keys)
+    assert np.array_equal(np.asarray(v).reshape(-1, 12), vals)
+    # empty
+    empty = np.empty(0, dtype=np.uint64)
+    buf = pack_partition_segment(empty, None)
+    k, v = unpack_partition_segment(buf, 0)
+    assert len(k) == 0
