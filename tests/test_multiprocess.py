@@ -114,7 +114,7 @@ def test_multiprocess_pagerank_matches_dense(tmp_path):
     np.testing.assert_allclose(got, want, rtol=1e-12)
 
 
-@pytest.mark.parametrize("world_size", [2, 4])
+@pytest.mark.parametrize("world_size", [2, 4, 8])
 def test_multiprocess_shuffle(tmp_path, world_size):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
