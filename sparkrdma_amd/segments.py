@@ -77,7 +77,13 @@ class HostSegment:
         return memoryview(self.mm)[addr:addr + length]
 
     def close(self) -> None:
-        self.mm.close()
+        try:
+            self.mm.close()
+        except BufferError:
+            # a caller still holds an exported view (numpy table over the
+            # segment, etc.) — leave the mapping to die with the process
+            # rather than crash teardown
+            pass
 
     def unlink(self) -> None:
         try:
@@ -116,8 +122,17 @@ class HostSegmentReader:
             # DEFER closing the old fd: a concurrent pread may hold it;
             # stale fds only ever map to segments with no live blocks
             # (liveness discipline), so the laggard read is of a dead
-            # range, never of current data
+            # range, never of current data. Bound the deferral: an fd
+            # 8 generations stale predates 8 full segment-recycle cycles
+            # — any pread that grabbed it is long finished — so closing
+            # it keeps a long-running executor's fd count constant
+            # (the soak found 1 leaked fd per recycle without this).
             self._old_fds.append(self.fd)
+            while len(self._old_fds) > 8:
+                try:
+                    os.close(self._old_fds.pop(0))
+                except OSError:
+                    pass
             self.fd = new_fd
             self._ino = os.fstat(new_fd).st_ino
 

@@ -264,3 +264,32 @@ def test_host_segment_ids_recycle_with_revalidation(tmp_path):
         m0.stop()
         m1.stop()
         driver.stop()
+
+
+def test_fd_count_bounded_over_many_shuffles(tmp_path):
+    """A long-running executor's fd count must stay CONSTANT across many
+    register/write/read/unregister cycles: segment-id recycling triggers
+    reader revalidation, whose deferred-close list is bounded (the 45-min
+    soak found 1 leaked fd per recycle before the bound existed)."""
+    import os
+    from sparkrdma_amd.engine import Engine
+
+    def nfds():
+        return len(os.listdir("/proc/self/fd"))
+
+    conf = ShuffleConf(shm_dir=str(tmp_path), max_buffer_allocation_size=1 << 30)
+    with Engine(conf, rank=0, world_size=1, driver_port=0) as eng:
+        baseline = None
+        for i in range(120):
+            h = eng.register_shuffle(1, 4)
+            w = eng.manager.get_writer(h, 0)
+            w.write_records([(k, k) for k in range(50)], None)
+            w.stop(True)
+            r = eng.manager.get_reader(h, 0, 3)
+            assert sum(1 for _ in r.read_records()) == 50
+            eng.unregister_shuffle(h)
+            if i == 30:
+                baseline = nfds()   # after caches/deferral window warm
+        assert baseline is not None
+        assert nfds() <= baseline + 4, \
+            f"fd leak: {nfds()} vs baseline {baseline}"
