@@ -274,6 +274,8 @@ def test_fd_count_bounded_over_many_shuffles(tmp_path):
     import os
     from sparkrdma_amd.engine import Engine
 
+    import threading
+
     def nfds():
         return len(os.listdir("/proc/self/fd"))
 
@@ -290,6 +292,9 @@ def test_fd_count_bounded_over_many_shuffles(tmp_path):
             eng.unregister_shuffle(h)
             if i == 30:
                 baseline = nfds()   # after caches/deferral window warm
+                threads30 = threading.active_count()
         assert baseline is not None
         assert nfds() <= baseline + 4, \
             f"fd leak: {nfds()} vs baseline {baseline}"
+        assert threading.active_count() <= threads30 + 2, \
+            "thread leak across shuffle lifecycles"
