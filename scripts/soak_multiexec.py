@@ -30,33 +30,45 @@ def worker(rank, world, driver_port, shm_dir, minutes, seed, q):
         with Engine(conf, rank=rank, world_size=world,
                     driver_port=driver_port) as eng:
             deadline = time.monotonic() + minutes * 60
-            while time.monotonic() < deadline:
+            # termination: rank 0 owns the clock; other ranks follow until
+            # rank 0 departs (its shutdown releases the barrier, then the
+            # next lookup fails LOUDLY with KeyError/ConnectionError — the
+            # framework's desync detection doubling as the stop signal)
+            while rank != 0 or time.monotonic() < deadline:
                 R = rng.choice((2, 4, 8, 16))     # same on both ranks
                 n = rng.randrange(100, 20_000)
                 part = HashPartitioner(R)
-                h = eng.register_shuffle(world, R)
-                w = eng.manager.get_writer(h, rank)
-                keys = np.random.default_rng(
-                    seed + rounds * world + rank).integers(
-                    0, 2 ** 64, n, dtype=np.uint64)
-                w.write_batch(keys)
-                w.stop(True, partitioner=part)
-                eng.barrier()
-                lo = rank * (R // world)
-                hi = (rank + 1) * (R // world) - 1
-                reader = eng.manager.get_reader(h, lo, hi)
-                got = sum(len(c) // 8 for _ref, c in reader)
-                # expected: both ranks' keys falling into [lo, hi]
-                want = 0
-                for r2 in range(world):
-                    k2 = np.random.default_rng(
-                        seed + rounds * world + r2).integers(
+                try:
+                    h = eng.register_shuffle(world, R)
+                    w = eng.manager.get_writer(h, rank)
+                    keys = np.random.default_rng(
+                        seed + rounds * world + rank).integers(
                         0, 2 ** 64, n, dtype=np.uint64)
-                    pid = part.partition_ids(k2)
-                    want += int(np.sum((pid >= lo) & (pid <= hi)))
-                assert got == want, f"rank {rank} round {rounds}: {got} != {want}"
-                assert reader.metrics.remote_blocks_fetched > 0 or world == 1
-                eng.unregister_shuffle(h)
+                    w.write_batch(keys)
+                    w.stop(True, partitioner=part)
+                    eng.barrier()
+                    lo = rank * (R // world)
+                    hi = (rank + 1) * (R // world) - 1
+                    reader = eng.manager.get_reader(h, lo, hi)
+                    got = sum(len(c) // 8 for _ref, c in reader)
+                    # expected: both ranks' keys falling into [lo, hi]
+                    want = 0
+                    for r2 in range(world):
+                        k2 = np.random.default_rng(
+                            seed + rounds * world + r2).integers(
+                            0, 2 ** 64, n, dtype=np.uint64)
+                        pid = part.partition_ids(k2)
+                        want += int(np.sum((pid >= lo) & (pid <= hi)))
+                    assert got == want, \
+                        f"rank {rank} round {rounds}: {got} != {want}"
+                    assert reader.metrics.remote_blocks_fetched > 0 or world == 1
+                    eng.unregister_shuffle(h)
+                except AssertionError:
+                    raise       # correctness failures are NEVER clean stops
+                except Exception:
+                    if rank != 0 and time.monotonic() > deadline - 30:
+                        break   # rank 0 exited mid-round: clean stop
+                    raise
                 rounds += 1
                 recs += n
         q.put((rank, rounds, recs, None))
