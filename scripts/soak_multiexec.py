@@ -4,7 +4,11 @@ map output and each reads the OTHER's partitions too (cross-executor
 one-sided reads), verifying counts — endurance for the announce/prebuild/
 remote-read lane that the single-executor soak cannot exercise.
 
-  python scripts/soak_multiexec.py [minutes] [seed]
+  python scripts/soak_multiexec.py [minutes] [seed] [transport] [world]
+
+transport "tcp" forces every cross-executor read through the data
+servers (the multi-node lane) — endurance for the streaming protocol
+and the client connection pool.
 """
 
 import multiprocessing as mp
@@ -16,7 +20,8 @@ import time
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
-def worker(rank, world, driver_port, shm_dir, minutes, seed, q):
+def worker(rank, world, driver_port, shm_dir, minutes, seed, q,
+           transport="auto"):
     import random
     import numpy as np
     from sparkrdma_amd.conf import ShuffleConf
@@ -24,7 +29,8 @@ def worker(rank, world, driver_port, shm_dir, minutes, seed, q):
     from sparkrdma_amd.partitioner import HashPartitioner
 
     rng = random.Random(seed * 100 + 7)   # SAME stream on both ranks
-    conf = ShuffleConf(shm_dir=shm_dir, max_buffer_allocation_size=1 << 30)
+    conf = ShuffleConf(shm_dir=shm_dir, max_buffer_allocation_size=1 << 30,
+                       transport=transport)
     rounds = recs = 0
     try:
         with Engine(conf, rank=rank, world_size=world,
@@ -80,17 +86,19 @@ def worker(rank, world, driver_port, shm_dir, minutes, seed, q):
 def main():
     minutes = float(sys.argv[1]) if len(sys.argv) > 1 else 10
     seed = int(sys.argv[2]) if len(sys.argv) > 2 else 0
+    transport = sys.argv[3] if len(sys.argv) > 3 else "auto"
     tmp = tempfile.mkdtemp(prefix="sparkrdma_soak2_")
     import socket
     s = socket.socket()
     s.bind(("127.0.0.1", 0))
     port = s.getsockname()[1]
     s.close()
-    world = 2
+    world = int(sys.argv[4]) if len(sys.argv) > 4 else 2
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=worker,
-                         args=(r, world, port, tmp, minutes, seed, q))
+                         args=(r, world, port, tmp, minutes, seed, q,
+                               transport))
              for r in range(world)]
     for p in procs:
         p.start()
@@ -107,7 +115,7 @@ def main():
         print("SOAK FAILED:", fails)
         sys.exit(1)
     print(f"multi-executor soak ok ({minutes:.0f} min, world={world}, "
-          "cross-executor reads verified every cycle)")
+          f"transport={transport}, cross-executor reads verified every cycle)")
 
 
 if __name__ == "__main__":
