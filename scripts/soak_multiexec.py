@@ -41,7 +41,7 @@ def worker(rank, world, driver_port, shm_dir, minutes, seed, q,
             # next lookup fails LOUDLY with KeyError/ConnectionError — the
             # framework's desync detection doubling as the stop signal)
             while rank != 0 or time.monotonic() < deadline:
-                R = rng.choice((2, 4, 8, 16))     # same on both ranks
+                R = world * rng.choice((1, 2, 4, 8))  # same on all ranks
                 n = rng.randrange(100, 20_000)
                 part = HashPartitioner(R)
                 try:
