@@ -79,3 +79,15 @@ def test_api_tour_example_runs():
                        capture_output=True, text=True, timeout=120)
     assert r.returncode == 0, r.stderr[-2000:]
     assert "range-partitioned" in r.stdout
+
+
+def test_soak_scripts_smoke():
+    """The endurance soak scripts stay runnable (a few seconds each)."""
+    import subprocess
+    import sys
+    import os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "scripts", "soak_control_plane.py"),
+         "0.05"], capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0 and "soak ok" in r.stdout, r.stderr[-1500:]
